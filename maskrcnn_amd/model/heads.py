@@ -1,0 +1,82 @@
+"""FPN classifier head and mask head.
+
+Replicates reference fpn_classifier_graph / build_fpn_mask_graph
+(mrcnn/model.py:~905-1015):
+
+Classifier: PyramidROIAlign(7x7) -> conv 7x7 -> 1024 (acts as fc1) + BN ->
+conv 1x1 -> 1024 + BN -> class logits / softmax + per-class bbox deltas.
+
+Mask: PyramidROIAlign(14x14) -> 4x (3x3 conv 256 + BN + ReLU) ->
+deconv 2x2/2 -> 1x1 conv num_classes + sigmoid -> [B, R, C, 28, 28].
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .resnet import make_norm
+from .roialign import PyramidROIAlign
+
+
+class FPNClassifierHead(nn.Module):
+    def __init__(self, config, in_channels=256):
+        super().__init__()
+        pool = config.POOL_SIZE
+        fc = config.FPN_CLASSIF_FC_LAYERS_SIZE
+        nc = config.NUM_CLASSES
+        train_bn = bool(config.TRAIN_BN)
+        self.roi_align = PyramidROIAlign(pool, config.IMAGE_SHAPE,
+                                         mode="crop_and_resize")
+        # 7x7 conv with VALID padding == fully-connected over the pooled tile.
+        self.mrcnn_class_conv1 = nn.Conv2d(in_channels, fc, pool)
+        self.mrcnn_class_bn1 = make_norm(fc, train_bn)
+        self.mrcnn_class_conv2 = nn.Conv2d(fc, fc, 1)
+        self.mrcnn_class_bn2 = make_norm(fc, train_bn)
+        self.mrcnn_class_logits = nn.Linear(fc, nc)
+        self.mrcnn_bbox_fc = nn.Linear(fc, nc * 4)
+        self.num_classes = nc
+
+    def forward(self, rois, feature_maps):
+        """rois [B,N,4] -> (logits [B,N,C], probs [B,N,C], bbox [B,N,C,4])."""
+        B, N, _ = rois.shape
+        x = self.roi_align(rois, feature_maps)           # [B,N,256,7,7]
+        x = x.reshape(B * N, *x.shape[2:])
+        x = F.relu(self.mrcnn_class_bn1(self.mrcnn_class_conv1(x)))
+        x = F.relu(self.mrcnn_class_bn2(self.mrcnn_class_conv2(x)))
+        x = x.flatten(1)                                 # [B*N, fc]
+        logits = self.mrcnn_class_logits(x)
+        probs = F.softmax(logits, dim=-1)
+        bbox = self.mrcnn_bbox_fc(x).reshape(B * N, self.num_classes, 4)
+        return (logits.reshape(B, N, -1), probs.reshape(B, N, -1),
+                bbox.reshape(B, N, self.num_classes, 4))
+
+
+class FPNMaskHead(nn.Module):
+    def __init__(self, config, in_channels=256):
+        super().__init__()
+        nc = config.NUM_CLASSES
+        train_bn = bool(config.TRAIN_BN)
+        self.roi_align = PyramidROIAlign(config.MASK_POOL_SIZE,
+                                         config.IMAGE_SHAPE,
+                                         mode="crop_and_resize")
+        layers = []
+        for i in range(4):
+            layers += [nn.Conv2d(in_channels if i == 0 else 256, 256, 3,
+                                 padding=1),
+                       make_norm(256, train_bn),
+                       nn.ReLU(inplace=True)]
+        self.convs = nn.Sequential(*layers)
+        self.mrcnn_mask_deconv = nn.ConvTranspose2d(256, 256, 2, stride=2)
+        self.mrcnn_mask = nn.Conv2d(256, nc, 1)
+
+    def forward(self, rois, feature_maps):
+        """rois [B,N,4] -> masks [B,N,C,28,28] (sigmoid)."""
+        B, N, _ = rois.shape
+        x = self.roi_align(rois, feature_maps)           # [B,N,256,14,14]
+        x = x.reshape(B * N, *x.shape[2:])
+        x = self.convs(x)
+        x = F.relu(self.mrcnn_mask_deconv(x))
+        x = torch.sigmoid(self.mrcnn_mask(x))            # [B*N, C, 28, 28]
+        return x.reshape(B, N, *x.shape[1:])

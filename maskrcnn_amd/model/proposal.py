@@ -1,0 +1,62 @@
+"""Proposal generation (RPN outputs -> RoIs).
+
+Replicates reference ProposalLayer (mrcnn/model.py:~250-335): per image,
+top-k PRE_NMS_LIMIT fg scores, apply deltas (x RPN_BBOX_STD_DEV) to
+anchors, clip to [0,1], NMS at RPN_NMS_THRESHOLD, pad to proposal_count.
+
+Batched torch implementation: top-k and delta application run batched on
+the full [B, A] tensors (A=261,888 @1024^2); only the NMS loops over the
+(small) batch, using the HIP bitmask NMS on GPU.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from ..ops import nms
+from ..utils.boxes import apply_box_deltas_t, clip_boxes_t
+
+
+class ProposalLayer(nn.Module):
+    def __init__(self, config):
+        super().__init__()
+        self.config = config
+        self.register_buffer(
+            "rpn_bbox_std_dev",
+            torch.tensor(config.RPN_BBOX_STD_DEV, dtype=torch.float32),
+            persistent=False)
+
+    def forward(self, rpn_probs, rpn_bbox, anchors, proposal_count):
+        """
+        rpn_probs: [B, A, 2] (bg, fg)
+        rpn_bbox:  [B, A, 4] deltas
+        anchors:   [A, 4] normalized
+        Returns rois [B, proposal_count, 4] normalized, zero-padded.
+        """
+        cfg = self.config
+        scores = rpn_probs[:, :, 1].float()
+        deltas = rpn_bbox.float() * self.rpn_bbox_std_dev.view(1, 1, 4)
+        B, A = scores.shape
+
+        pre_nms_limit = min(cfg.PRE_NMS_LIMIT, A)
+        top_scores, ix = torch.topk(scores, pre_nms_limit, dim=1,
+                                    sorted=True)
+        # Gather deltas and anchors per image.
+        deltas = torch.gather(deltas, 1,
+                              ix.unsqueeze(-1).expand(-1, -1, 4))
+        anchors_b = anchors.unsqueeze(0).expand(B, A, 4).float()
+        anchors_g = torch.gather(anchors_b, 1,
+                                 ix.unsqueeze(-1).expand(-1, -1, 4))
+
+        boxes = apply_box_deltas_t(anchors_g, deltas)
+        window = torch.tensor([0., 0., 1., 1.], device=boxes.device)
+        boxes = clip_boxes_t(boxes, window)
+
+        rois = boxes.new_zeros((B, proposal_count, 4))
+        for b in range(B):
+            keep = nms(boxes[b], top_scores[b], cfg.RPN_NMS_THRESHOLD,
+                       max_out=proposal_count)
+            n = keep.shape[0]
+            rois[b, :n] = boxes[b, keep]
+        return rois

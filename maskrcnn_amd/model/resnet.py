@@ -1,0 +1,130 @@
+"""ResNet-50/101 backbone (bottleneck-v1, frozen-BN).
+
+Replicates the capability of reference mrcnn/model.py resnet_graph
+(~37-200): stem 7x7/2 + maxpool 3x3/2, bottleneck stages with counts
+resnet50=(3,4,6,3) / resnet101=(3,4,23,3), returning [C1..C5].
+PyTorch-native NCHW; BN is frozen by default (reference TRAIN_BN=False —
+BatchNorm(training=False) wrapper, model.py:~40), which on MI355X lets the
+serving path fold BN into conv weights (csrc conv epilogue, N5).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+
+class FrozenBatchNorm2d(nn.Module):
+    """BatchNorm with statistics and affine params frozen.
+
+    Equivalent to the reference's BatchNorm(training=False): always uses
+    running stats; affine params still exist (loadable) but see no batch
+    statistics update. Cheaper than nn.BatchNorm2d(eval) and safe under DDP.
+    """
+
+    def __init__(self, num_features, eps=1e-5):
+        super().__init__()
+        self.num_features = num_features
+        self.eps = eps
+        self.register_buffer("weight", torch.ones(num_features))
+        self.register_buffer("bias", torch.zeros(num_features))
+        self.register_buffer("running_mean", torch.zeros(num_features))
+        self.register_buffer("running_var", torch.ones(num_features))
+
+    def forward(self, x):
+        scale = self.weight * (self.running_var + self.eps).rsqrt()
+        shift = self.bias - self.running_mean * scale
+        scale = scale.to(x.dtype).view(1, -1, 1, 1)
+        shift = shift.to(x.dtype).view(1, -1, 1, 1)
+        return x * scale + shift
+
+    def extra_repr(self):
+        return f"{self.num_features}, eps={self.eps}"
+
+
+def make_norm(num_features, train_bn: bool):
+    if train_bn:
+        return nn.BatchNorm2d(num_features, eps=1e-5)
+    return FrozenBatchNorm2d(num_features)
+
+
+class Bottleneck(nn.Module):
+    """1x1 -> 3x3 -> 1x1 bottleneck with identity or projection shortcut.
+
+    Mirrors reference conv_block / identity_block (model.py:~60-150).
+    """
+    expansion = 4
+
+    def __init__(self, in_ch, planes, stride=1, downsample=None,
+                 train_bn=False):
+        super().__init__()
+        out_ch = planes * self.expansion
+        self.conv1 = nn.Conv2d(in_ch, planes, 1, stride=stride, bias=True)
+        self.bn1 = make_norm(planes, train_bn)
+        self.conv2 = nn.Conv2d(planes, planes, 3, padding=1, bias=True)
+        self.bn2 = make_norm(planes, train_bn)
+        self.conv3 = nn.Conv2d(planes, out_ch, 1, bias=True)
+        self.bn3 = make_norm(out_ch, train_bn)
+        self.relu = nn.ReLU(inplace=True)
+        self.downsample = downsample
+
+    def forward(self, x):
+        identity = x
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        return self.relu(out + identity)
+
+
+class ResNet(nn.Module):
+    """Returns feature maps [C1, C2, C3, C4, C5] (strides 2,4,8,16,32)."""
+
+    STAGE_BLOCKS = {"resnet50": (3, 4, 6, 3), "resnet101": (3, 4, 23, 3)}
+
+    def __init__(self, architecture="resnet101", stage5=True, train_bn=False):
+        super().__init__()
+        assert architecture in self.STAGE_BLOCKS, architecture
+        blocks = self.STAGE_BLOCKS[architecture]
+        self.architecture = architecture
+        self.stage5 = stage5
+
+        # Stem (reference: 7x7/2 conv + BN + ReLU + 3x3/2 maxpool).
+        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=True)
+        self.bn_conv1 = make_norm(64, train_bn)
+        self.relu = nn.ReLU(inplace=True)
+        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+
+        self.in_ch = 64
+        self.layer2 = self._make_stage(64, blocks[0], stride=1,
+                                       train_bn=train_bn)
+        self.layer3 = self._make_stage(128, blocks[1], stride=2,
+                                       train_bn=train_bn)
+        self.layer4 = self._make_stage(256, blocks[2], stride=2,
+                                       train_bn=train_bn)
+        if stage5:
+            self.layer5 = self._make_stage(512, blocks[3], stride=2,
+                                           train_bn=train_bn)
+        else:
+            self.layer5 = None
+
+    def _make_stage(self, planes, num_blocks, stride, train_bn):
+        out_ch = planes * Bottleneck.expansion
+        downsample = nn.Sequential(
+            nn.Conv2d(self.in_ch, out_ch, 1, stride=stride, bias=True),
+            make_norm(out_ch, train_bn))
+        layers = [Bottleneck(self.in_ch, planes, stride, downsample,
+                             train_bn)]
+        self.in_ch = out_ch
+        for _ in range(1, num_blocks):
+            layers.append(Bottleneck(self.in_ch, planes, train_bn=train_bn))
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        c1 = self.maxpool(self.relu(self.bn_conv1(self.conv1(x))))
+        c2 = self.layer2(c1)
+        c3 = self.layer3(c2)
+        c4 = self.layer4(c3)
+        c5 = self.layer5(c4) if self.layer5 is not None else None
+        return [c1, c2, c3, c4, c5]
