@@ -1,0 +1,109 @@
+"""Data-plane tests: dataset, resize modes, RPN targets, loader."""
+
+import numpy as np
+import pytest
+
+from maskrcnn_amd.data.loader import make_data_loader
+from maskrcnn_amd.data.targets import build_rpn_targets, load_image_gt
+from maskrcnn_amd.utils.image import resize_image
+from maskrcnn_amd.utils.masks import (expand_mask, minimize_mask, rle_decode,
+                                      rle_encode, rle_iou, unmold_mask)
+
+
+def test_resize_square_mode():
+    img = np.random.randint(0, 255, (300, 500, 3), dtype=np.uint8)
+    out, window, scale, padding, crop = resize_image(
+        img, min_dim=256, max_dim=256, mode="square")
+    assert out.shape == (256, 256, 3)
+    y1, x1, y2, x2 = window
+    # Aspect preserved: 300x500 -> scale 256/500 -> 154x256 content.
+    assert x2 - x1 == 256
+    assert abs((y2 - y1) - round(300 * 256 / 500)) <= 1
+
+
+def test_resize_pad64_mode():
+    img = np.random.randint(0, 255, (100, 200, 3), dtype=np.uint8)
+    out, window, scale, padding, crop = resize_image(
+        img, min_dim=128, mode="pad64")
+    assert out.shape[0] % 64 == 0 and out.shape[1] % 64 == 0
+
+
+def test_resize_crop_mode():
+    img = np.random.randint(0, 255, (300, 400, 3), dtype=np.uint8)
+    out, window, scale, padding, crop = resize_image(
+        img, min_dim=128, mode="crop")
+    assert out.shape == (128, 128, 3)
+    assert crop is not None
+
+
+def test_mini_mask_roundtrip():
+    mask = np.zeros((64, 64, 1), dtype=bool)
+    mask[10:40, 20:50, 0] = True
+    from maskrcnn_amd.utils.boxes import extract_bboxes
+    bbox = extract_bboxes(mask)
+    mini = minimize_mask(bbox, mask, (28, 28))
+    assert mini.shape == (28, 28, 1)
+    back = expand_mask(bbox, mini, (64, 64))
+    inter = np.logical_and(back[..., 0], mask[..., 0]).sum()
+    union = np.logical_or(back[..., 0], mask[..., 0]).sum()
+    assert inter / union > 0.9
+
+
+def test_unmold_mask():
+    small = np.random.rand(28, 28)
+    full = unmold_mask(small, np.array([10, 10, 30, 40]), (100, 100, 3))
+    assert full.shape == (100, 100)
+    assert not full[:10].any() and not full[:, :10].any()
+
+
+def test_rle_roundtrip():
+    rng = np.random.RandomState(0)
+    mask = rng.rand(37, 23) > 0.5
+    rle = rle_encode(mask)
+    back = rle_decode(rle)
+    np.testing.assert_array_equal(back, mask)
+    assert rle_iou(rle, rle) == pytest.approx(1.0)
+
+
+def test_build_rpn_targets(shapes_config, shapes_dataset):
+    cfg = shapes_config
+    from maskrcnn_amd.utils.anchors import AnchorCache
+    anchors = AnchorCache(cfg).get_pixel(cfg.IMAGE_SHAPE)
+    image, meta, class_ids, bbox, mask = load_image_gt(
+        shapes_dataset, cfg, 0)
+    rpn_match, rpn_bbox = build_rpn_targets(
+        image.shape, anchors, class_ids, bbox, cfg)
+    assert rpn_match.shape == (anchors.shape[0],)
+    n_pos = (rpn_match == 1).sum()
+    n_neg = (rpn_match == -1).sum()
+    assert n_pos > 0
+    assert n_pos + n_neg <= cfg.RPN_TRAIN_ANCHORS_PER_IMAGE
+    # Deltas for positives only, finite.
+    assert np.isfinite(rpn_bbox[:n_pos]).all()
+
+
+def test_loader_batch_shapes(shapes_config, shapes_dataset):
+    cfg = shapes_config
+    loader = make_data_loader(shapes_dataset, cfg, batch_size=2)
+    batch = next(iter(loader))
+    H, W, _ = cfg.IMAGE_SHAPE
+    A = 0
+    from maskrcnn_amd.config import compute_backbone_shapes
+    for s, fs in zip(cfg.RPN_ANCHOR_SCALES,
+                     compute_backbone_shapes(cfg, cfg.IMAGE_SHAPE)):
+        A += fs[0] * fs[1] * len(cfg.RPN_ANCHOR_RATIOS)
+    assert batch["image"].shape == (2, 3, H, W)
+    assert batch["image_meta"].shape == (2, cfg.IMAGE_META_SIZE)
+    assert batch["rpn_match"].shape == (2, A, 1)
+    assert batch["rpn_bbox"].shape == (2, cfg.RPN_TRAIN_ANCHORS_PER_IMAGE, 4)
+    assert batch["gt_class_ids"].shape == (2, cfg.MAX_GT_INSTANCES)
+    assert batch["gt_boxes"].shape == (2, cfg.MAX_GT_INSTANCES, 4)
+    mh, mw = cfg.MINI_MASK_SHAPE
+    assert batch["gt_masks"].shape == (2, mh, mw, cfg.MAX_GT_INSTANCES)
+
+
+def test_dataset_prepare_class_mapping(shapes_dataset):
+    ds = shapes_dataset
+    assert ds.num_classes == 4
+    assert ds.class_names[0] == "BG"
+    assert ds.map_source_class_id("shapes.1") == 1

@@ -1,0 +1,113 @@
+"""End-to-end CPU model tests (BASELINE config #1 class: plumbing)."""
+
+import numpy as np
+import pytest
+import torch
+
+from maskrcnn_amd.model import MaskRCNN
+from maskrcnn_amd.data.loader import make_data_loader
+from maskrcnn_amd.engine.trainer import Trainer, build_optimizer
+
+
+def test_inference_forward_shapes(shapes_config):
+    cfg = shapes_config
+    m = MaskRCNN("inference", cfg, model_dir="/tmp/mrcnn_test_logs")
+    img = (np.random.rand(100, 150, 3) * 255).astype(np.uint8)
+    results = m.detect([img])
+    r = results[0]
+    assert r["rois"].ndim == 2 and r["rois"].shape[1] == 4
+    assert r["class_ids"].shape[0] == r["rois"].shape[0]
+    assert r["scores"].shape[0] == r["rois"].shape[0]
+    assert r["masks"].shape[:2] == img.shape[:2]
+
+
+def test_inference_output_signature(shapes_config):
+    """The 7 named outputs of the serving signature exist with right shapes."""
+    cfg = shapes_config
+    m = MaskRCNN("inference", cfg, model_dir="/tmp/mrcnn_test_logs")
+    img = (np.random.rand(128, 128, 3) * 255).astype(np.uint8)
+    molded, metas, _ = m.mold_inputs([img])
+    out = m.detect_molded(molded, metas)
+    D = cfg.DETECTION_MAX_INSTANCES
+    C = cfg.NUM_CLASSES
+    N = cfg.POST_NMS_ROIS_INFERENCE
+    assert out["mrcnn_detection"].shape == (1, D, 6)
+    assert out["mrcnn_mask"].shape == (1, D, C, 28, 28)
+    assert out["rois"].shape == (1, N, 4)
+    assert out["mrcnn_class"].shape == (1, N, C)
+    assert out["mrcnn_bbox"].shape == (1, N, C, 4)
+    assert out["rpn_class"].shape[0] == 1
+    assert out["rpn_bbox"].shape[2] == 4
+
+
+def test_train_step_runs_and_losses_finite(shapes_config, shapes_dataset):
+    cfg = shapes_config
+    m = MaskRCNN("training", cfg, model_dir="/tmp/mrcnn_test_logs")
+    tr = Trainer(m, shapes_dataset)
+    opt = build_optimizer(m.model, cfg, cfg.LEARNING_RATE)
+    loader = make_data_loader(shapes_dataset, cfg)
+    batch = next(iter(loader))
+    losses = tr.train_step(batch, opt)
+    for k, v in losses.items():
+        assert np.isfinite(v), (k, v)
+    assert set(losses) == {"rpn_class_loss", "rpn_bbox_loss",
+                           "mrcnn_class_loss", "mrcnn_bbox_loss",
+                           "mrcnn_mask_loss", "total_loss"}
+
+
+def test_loss_decreases_overfit_single_batch(shapes_config, shapes_dataset):
+    """Overfit one batch: total loss must drop measurably in 15 steps."""
+    cfg = shapes_config
+    torch.manual_seed(0)
+    m = MaskRCNN("training", cfg, model_dir="/tmp/mrcnn_test_logs")
+    tr = Trainer(m, shapes_dataset)
+    opt = build_optimizer(m.model, cfg, 0.02)
+    loader = make_data_loader(shapes_dataset, cfg, shuffle=False)
+    batch = next(iter(loader))
+    first = tr.train_step(batch, opt)["total_loss"]
+    last = first
+    for _ in range(15):
+        last = tr.train_step(batch, opt)["total_loss"]
+    assert last < first * 0.9, (first, last)
+
+
+def test_set_trainable_freezes_backbone(shapes_config):
+    m = MaskRCNN("training", shapes_config, model_dir="/tmp/mrcnn_test_logs")
+    m.set_trainable("heads")
+    for name, p in m.model.named_parameters():
+        if name.startswith("backbone."):
+            assert not p.requires_grad, name
+        if name.startswith("rpn.") or name.startswith("fpn."):
+            assert p.requires_grad, name
+    n_all = m.set_trainable("all")
+    assert n_all == sum(1 for _ in m.model.parameters())
+
+
+def test_checkpoint_save_load_roundtrip(shapes_config, tmp_path):
+    m = MaskRCNN("training", shapes_config, model_dir=str(tmp_path))
+    p = str(tmp_path / "w.pth")
+    m.save_weights(p)
+    m2 = MaskRCNN("inference", shapes_config, model_dir=str(tmp_path))
+    m2.load_weights(p)
+    for (k1, v1), (k2, v2) in zip(m.model.state_dict().items(),
+                                  m2.model.state_dict().items()):
+        assert k1 == k2
+        assert torch.equal(v1, v2), k1
+
+
+def test_load_weights_exclude_head_surgery(shapes_config, tmp_path):
+    m = MaskRCNN("training", shapes_config, model_dir=str(tmp_path))
+    p = str(tmp_path / "w.pth")
+    m.save_weights(p)
+
+    class MoreClasses(type(shapes_config)):
+        NUM_CLASSES = 7
+    m2 = MaskRCNN("training", MoreClasses(), model_dir=str(tmp_path))
+    m2.load_weights(p, by_name=True,
+                    exclude=[r"classifier_head\.mrcnn_class_logits",
+                             r"classifier_head\.mrcnn_bbox_fc",
+                             r"mask_head\.mrcnn_mask\."])
+    # Backbone weights transferred.
+    assert torch.equal(
+        m2.model.backbone.conv1.weight,
+        m.model.backbone.conv1.weight)
