@@ -1,0 +1,238 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: R101-FPN Mask R-CNN bf16 training step throughput.
+
+Measures the BASELINE.json headline metric — images/sec of ResNet-101-FPN
+Mask R-CNN training (per-GPU batch 2, 1024x1024, synthetic COCO-shaped
+data, random-init weights) on 1..8 MI355X GPUs (weak scaling, DP over
+RCCL/xGMI).
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for
+N>1 launched under torch.distributed.run with one rank per GPU. Rank 0
+prints ONE JSON line with the whole-job aggregate.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+
+def build_config(args):
+    from maskrcnn_amd.config import Config
+
+    class BenchConfig(Config):
+        NAME = "coco_bench"
+        BACKBONE = "resnet101"
+        NUM_CLASSES = 81
+        IMAGES_PER_GPU = args.batch
+        GPU_COUNT = 1  # one process per GPU; DP handled by DDP wrapper
+        IMAGE_MIN_DIM = args.image_size
+        IMAGE_MAX_DIM = args.image_size
+        STEPS_PER_EPOCH = args.steps
+        COMPUTE_DTYPE = "bf16" if args.dtype == "bf16" else "fp32"
+
+    return BenchConfig()
+
+
+def make_synthetic_batch(config, device, seed=0):
+    """One COCO-shaped synthetic training batch, fully on device.
+
+    Random molded images + random GT; RPN targets computed once (untimed)
+    with the real anchor-matching code path.
+    """
+    from maskrcnn_amd.data.targets import build_rpn_targets
+    from maskrcnn_amd.utils.anchors import AnchorCache
+    from maskrcnn_amd.utils.meta import compose_image_meta
+
+    rng = np.random.RandomState(seed)
+    B = config.BATCH_SIZE
+    H, W, _ = config.IMAGE_SHAPE
+    cache = AnchorCache(config)
+    anchors_px = cache.get_pixel(config.IMAGE_SHAPE)
+
+    images = rng.randn(B, 3, H, W).astype(np.float32) * 40.0
+    metas, rpn_matches, rpn_bboxes = [], [], []
+    gt_cls = np.zeros((B, config.MAX_GT_INSTANCES), dtype=np.int64)
+    gt_boxes = np.zeros((B, config.MAX_GT_INSTANCES, 4), dtype=np.float32)
+    mh, mw = config.MINI_MASK_SHAPE
+    gt_masks = np.zeros((B, mh, mw, config.MAX_GT_INSTANCES),
+                        dtype=np.float32)
+
+    for b in range(B):
+        n = rng.randint(8, 30)  # typical COCO instance count range
+        ys = rng.uniform(0, H - 64, n)
+        xs = rng.uniform(0, W - 64, n)
+        hs = rng.uniform(32, H / 3, n)
+        ws = rng.uniform(32, W / 3, n)
+        boxes_px = np.stack([ys, xs, np.minimum(ys + hs, H - 1),
+                             np.minimum(xs + ws, W - 1)], axis=1)
+        cls = rng.randint(1, config.NUM_CLASSES, n)
+        rpn_match, rpn_bbox = build_rpn_targets(
+            (H, W, 3), anchors_px, cls, boxes_px, config, rng=rng)
+        rpn_matches.append(rpn_match[:, None].astype(np.int64))
+        rpn_bboxes.append(rpn_bbox.astype(np.float32))
+        gt_cls[b, :n] = cls
+        gt_boxes[b, :n] = boxes_px / np.array([H - 1, W - 1, H - 1, W - 1],
+                                              dtype=np.float32)
+        gt_masks[b, :, :, :n] = (rng.rand(mh, mw, n) > 0.5)
+        metas.append(compose_image_meta(
+            b, (H, W, 3), (H, W, 3), (0, 0, H, W), 1.0,
+            np.ones(config.NUM_CLASSES, np.int32)))
+
+    t = lambda x: torch.from_numpy(np.ascontiguousarray(x)).to(device)
+    return {
+        "image": t(images),
+        "image_meta": t(np.stack(metas).astype(np.float32)),
+        "rpn_match": t(np.stack(rpn_matches)),
+        "rpn_bbox": t(np.stack(rpn_bboxes)),
+        "gt_class_ids": t(gt_cls),
+        "gt_boxes": t(gt_boxes),
+        "gt_masks": t(gt_masks),
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--batch", type=int, default=2)
+    ap.add_argument("--image-size", type=int, default=1024)
+    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--backend", default=None,
+                    help="override dist backend (nccl|gloo)")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world > 1
+
+    use_cuda = torch.cuda.is_available()
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+
+    if distributed:
+        import torch.distributed as dist
+        backend = args.backend or ("nccl" if use_cuda else "gloo")
+        dist.init_process_group(backend=backend)
+
+    config = build_config(args)
+
+    from maskrcnn_amd.model.mask_rcnn import MaskRCNNModel
+    from maskrcnn_amd.model.losses import compute_losses
+    from maskrcnn_amd.engine.trainer import build_optimizer
+    from maskrcnn_amd.utils.anchors import AnchorCache
+    from maskrcnn_amd.utils.meta import parse_image_meta_t
+
+    torch.manual_seed(1234 + rank)
+    model = MaskRCNNModel(config).to(device)
+    if config.CHANNELS_LAST and use_cuda:
+        model = model.to(memory_format=torch.channels_last)
+    model.train()
+
+    if distributed:
+        from maskrcnn_amd.distributed import DistributedTrainer
+        ddp = DistributedTrainer(model, config)
+    else:
+        ddp = None
+
+    optimizer = build_optimizer(model, config, config.LEARNING_RATE)
+    anchors = AnchorCache(config).get(config.IMAGE_SHAPE, device=device)
+    batch = make_synthetic_batch(config, device, seed=100 + rank)
+    if config.CHANNELS_LAST and use_cuda:
+        batch["image"] = batch["image"].to(
+            memory_format=torch.channels_last)
+    meta = parse_image_meta_t(batch["image_meta"])
+
+    use_bf16 = (config.COMPUTE_DTYPE == "bf16" and use_cuda)
+
+    def one_step():
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                            enabled=use_bf16):
+            outputs = model(batch["image"], batch["image_meta"], anchors,
+                            gt_class_ids=batch["gt_class_ids"],
+                            gt_boxes=batch["gt_boxes"],
+                            gt_masks=batch["gt_masks"], mode="training")
+            losses = compute_losses(
+                config,
+                {"rpn_match": batch["rpn_match"],
+                 "rpn_bbox": batch["rpn_bbox"],
+                 "active_class_ids": meta["active_class_ids"]},
+                outputs)
+        optimizer.zero_grad(set_to_none=True)
+        losses["total_loss"].backward()
+        if ddp is not None:
+            ddp.finalize_backward()
+        torch.nn.utils.clip_grad_norm_(
+            (p for p in model.parameters() if p.requires_grad),
+            config.GRADIENT_CLIP_NORM)
+        optimizer.step()
+        return losses
+
+    for _ in range(args.warmup):
+        one_step()
+
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # Max over ranks = whole-job time.
+    if distributed:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], device=device if use_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    n_gpus = world if distributed else 1
+    global_batch = config.BATCH_SIZE * n_gpus
+    images_per_sec = global_batch * args.steps / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "train_images_per_sec",
+            "value": round(images_per_sec, 3),
+            "unit": "images/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype if use_cuda else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"{config.BACKBONE}-FPN maskrcnn",
+                "global_batch": global_batch,
+                "image_size": args.image_size,
+                "num_classes": config.NUM_CLASSES,
+                "parallelism": f"dp{n_gpus}",
+            },
+        }))
+
+    if distributed:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,125 @@
+// PyTorch bindings for the maskrcnn_amd CDNA4 HIP kernels.
+//
+// Compiled together with csrc/hip/*.hip by hipcc into the in-tree
+// maskrcnn_amd/_C extension (see build_ext.py). Tensors arrive as torch
+// "cuda" tensors (ROCm masquerades as CUDA); we pull the raw HIP stream
+// from c10::hip.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void roi_align_fwd_launch(const void*, const float*, const int*, void*, int,
+                          int, int, int, int, int, long, long, long, long,
+                          long, long, long, long, int, bool, bool,
+                          hipStream_t);
+void roi_align_bwd_launch(const void*, const float*, const int*, float*, int,
+                          int, int, int, int, int, long, long, long, long,
+                          long, long, long, long, int, bool, bool,
+                          hipStream_t);
+long nms_workspace_bytes(int N);
+void nms_launch(const float*, int, float, int, void*, int*, int*,
+                hipStream_t);
+}
+
+namespace {
+
+bool is_channels_last(const torch::Tensor& t) {
+  return t.is_contiguous(at::MemoryFormat::ChannelsLast);
+}
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+torch::Tensor roi_align_forward(torch::Tensor feats, torch::Tensor boxes,
+                                torch::Tensor box_ind, int64_t pool_h,
+                                int64_t pool_w, int64_t sampling,
+                                bool aligned) {
+  TORCH_CHECK(feats.is_cuda() && boxes.is_cuda() && box_ind.is_cuda(),
+              "roi_align: tensors must be on GPU");
+  TORCH_CHECK(feats.dim() == 4, "feats must be [B,C,H,W]");
+  TORCH_CHECK(boxes.dim() == 2 && boxes.size(1) == 4, "boxes must be [N,4]");
+  TORCH_CHECK(feats.scalar_type() == torch::kFloat ||
+                  feats.scalar_type() == torch::kBFloat16,
+              "roi_align: float/bf16 only");
+  const bool cl = is_channels_last(feats);
+  if (!cl) feats = feats.contiguous();
+  auto boxes_f = boxes.to(torch::kFloat).contiguous();
+  auto ind = box_ind.to(torch::kInt).contiguous();
+
+  const int N = boxes.size(0), C = feats.size(1), H = feats.size(2),
+            W = feats.size(3);
+  auto out = torch::empty(
+      {N, C, pool_h, pool_w},
+      feats.options().memory_format(cl ? at::MemoryFormat::ChannelsLast
+                                       : at::MemoryFormat::Contiguous));
+  const bool bf16 = feats.scalar_type() == torch::kBFloat16;
+  const int samp = aligned ? std::max<int>(1, sampling) : 0;
+  roi_align_fwd_launch(
+      feats.data_ptr(), boxes_f.data_ptr<float>(), ind.data_ptr<int>(),
+      out.data_ptr(), N, C, H, W, (int)pool_h, (int)pool_w,
+      feats.stride(0), feats.stride(1), feats.stride(2), feats.stride(3),
+      out.stride(0), out.stride(1), out.stride(2), out.stride(3), samp, cl,
+      bf16, cur_stream());
+  return out;
+}
+
+torch::Tensor roi_align_backward(torch::Tensor grad_out, torch::Tensor boxes,
+                                 torch::Tensor box_ind, int64_t B, int64_t C,
+                                 int64_t H, int64_t W, int64_t pool_h,
+                                 int64_t pool_w, int64_t sampling,
+                                 bool aligned) {
+  TORCH_CHECK(grad_out.is_cuda(), "grad_out must be on GPU");
+  const bool cl = is_channels_last(grad_out);
+  if (!cl) grad_out = grad_out.contiguous();
+  auto boxes_f = boxes.to(torch::kFloat).contiguous();
+  auto ind = box_ind.to(torch::kInt).contiguous();
+  const int N = boxes.size(0);
+  auto grad_in = torch::zeros(
+      {B, C, H, W},
+      grad_out.options().dtype(torch::kFloat)
+          .memory_format(cl ? at::MemoryFormat::ChannelsLast
+                            : at::MemoryFormat::Contiguous));
+  const bool bf16 = grad_out.scalar_type() == torch::kBFloat16;
+  const int samp = aligned ? std::max<int>(1, sampling) : 0;
+  roi_align_bwd_launch(
+      grad_out.data_ptr(), boxes_f.data_ptr<float>(), ind.data_ptr<int>(),
+      grad_in.data_ptr<float>(), N, (int)C, (int)H, (int)W, (int)pool_h,
+      (int)pool_w, grad_in.stride(0), grad_in.stride(1), grad_in.stride(2),
+      grad_in.stride(3), grad_out.stride(0), grad_out.stride(1),
+      grad_out.stride(2), grad_out.stride(3), samp, cl, bf16, cur_stream());
+  return grad_in;
+}
+
+torch::Tensor nms(torch::Tensor boxes, torch::Tensor scores,
+                  double iou_thresh) {
+  TORCH_CHECK(boxes.is_cuda() && scores.is_cuda(), "nms: GPU tensors only");
+  TORCH_CHECK(boxes.dim() == 2 && boxes.size(1) == 4, "boxes [N,4]");
+  const int N = boxes.size(0);
+  if (N == 0)
+    return torch::empty({0}, boxes.options().dtype(torch::kLong));
+  auto order = scores.argsort(0, /*descending=*/true);
+  auto sorted = boxes.index_select(0, order).contiguous().to(torch::kFloat);
+
+  auto ws = torch::empty({nms_workspace_bytes(N)},
+                         boxes.options().dtype(torch::kByte));
+  auto keep = torch::empty({N}, boxes.options().dtype(torch::kInt));
+  auto count = torch::zeros({1}, boxes.options().dtype(torch::kInt));
+  nms_launch(sorted.data_ptr<float>(), N, (float)iou_thresh, N,
+             (void*)ws.data_ptr(), keep.data_ptr<int>(),
+             count.data_ptr<int>(), cur_stream());
+  const int n = count.item<int>();  // one DtoH sync
+  return order.index_select(0, keep.slice(0, 0, n).to(torch::kLong));
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("roi_align_forward", &roi_align_forward,
+        "RoIAlign / crop_and_resize forward (CDNA4)");
+  m.def("roi_align_backward", &roi_align_backward,
+        "RoIAlign backward (CDNA4, atomic scatter)");
+  m.def("nms", &nms, "bitmask greedy NMS (CDNA4)");
+}
