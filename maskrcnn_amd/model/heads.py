@@ -72,11 +72,16 @@ class FPNMaskHead(nn.Module):
         self.mrcnn_mask = nn.Conv2d(256, nc, 1)
 
     def forward(self, rois, feature_maps):
-        """rois [B,N,4] -> masks [B,N,C,28,28] (sigmoid)."""
+        """rois [B,N,4] -> mask LOGITS [B,N,C,28,28].
+
+        (The reference applies sigmoid inside the graph; we keep logits for
+        an autocast-safe BCE-with-logits loss and apply sigmoid at the
+        inference output — numerically identical end to end.)
+        """
         B, N, _ = rois.shape
         x = self.roi_align(rois, feature_maps)           # [B,N,256,14,14]
         x = x.reshape(B * N, *x.shape[2:])
         x = self.convs(x)
         x = F.relu(self.mrcnn_mask_deconv(x))
-        x = torch.sigmoid(self.mrcnn_mask(x))            # [B*N, C, 28, 28]
+        x = self.mrcnn_mask(x)                           # [B*N, C, 28, 28]
         return x.reshape(B, N, *x.shape[1:])

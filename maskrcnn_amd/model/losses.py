@@ -89,19 +89,19 @@ def mrcnn_bbox_loss(target_bbox, target_class_ids, pred_bbox):
     return smooth_l1_loss(tgt, pred).mean()
 
 
-def mrcnn_mask_loss(target_masks, target_class_ids, pred_masks):
-    """target_masks: [B,T,mh,mw]; ids: [B,T]; pred: [B,T,C,mh,mw] (sigmoid)."""
-    B, T, C, mh, mw = pred_masks.shape
+def mrcnn_mask_loss(target_masks, target_class_ids, pred_mask_logits):
+    """target_masks: [B,T,mh,mw]; ids: [B,T]; pred: [B,T,C,mh,mw] LOGITS."""
+    B, T, C, mh, mw = pred_mask_logits.shape
     target_class_ids = target_class_ids.reshape(-1)
     target_masks = target_masks.reshape(-1, mh, mw).float()
-    pred_masks = pred_masks.reshape(-1, C, mh, mw).float()
+    pred = pred_mask_logits.reshape(-1, C, mh, mw).float()
     pos = (target_class_ids > 0).nonzero(as_tuple=True)[0]
     if pos.numel() == 0:
-        return pred_masks.sum() * 0.0
+        return pred.sum() * 0.0
     cls = target_class_ids[pos]
-    y_pred = pred_masks[pos, cls]
+    y_pred = pred[pos, cls]
     y_true = target_masks[pos]
-    return F.binary_cross_entropy(y_pred.clamp(1e-7, 1 - 1e-7), y_true)
+    return F.binary_cross_entropy_with_logits(y_pred, y_true)
 
 
 def compute_losses(config, inputs, outputs):

@@ -113,8 +113,8 @@ class MaskRCNNModel(nn.Module):
                                               mrcnn_bbox, image_meta)
             detection_boxes = detections[..., :4]
             # Mask head runs on the FINAL detections (reference :~2000).
-            mrcnn_mask = self.mask_head(detection_boxes,
-                                        mrcnn_feature_maps)
+            mrcnn_mask = torch.sigmoid(
+                self.mask_head(detection_boxes, mrcnn_feature_maps))
             return {
                 "mrcnn_detection": detections,
                 "mrcnn_class": mrcnn_class,
