@@ -117,6 +117,8 @@ def main():
     device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(device)
+        from maskrcnn_amd.utils.rocm_setup import setup_rocm_perf
+        setup_rocm_perf(local_rank)
 
     if distributed:
         import torch.distributed as dist

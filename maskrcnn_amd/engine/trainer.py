@@ -52,6 +52,8 @@ class Trainer:
         self.augmentation = augmentation
         self.num_workers = num_workers
         self.history = []
+        from ..utils.rocm_setup import setup_rocm_perf
+        setup_rocm_perf()
 
     def _autocast(self):
         use_bf16 = (self.config.COMPUTE_DTYPE == "bf16" and
