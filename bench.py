@@ -16,10 +16,18 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
-import numpy as np
-import torch
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+# MIOPEN_USER_DB_PATH must be exported before libMIOpen loads with torch
+# (profiles/r01_naive_conv.md) — stage the in-repo find-db first.
+from maskrcnn_amd._rocm_env import stage_miopen_db  # noqa: E402
+
+stage_miopen_db(int(os.environ.get("LOCAL_RANK", "0")))
+
+import numpy as np  # noqa: E402
+import torch  # noqa: E402
 
 
 def build_config(args):
@@ -117,7 +125,7 @@ def main():
     device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(device)
-        from maskrcnn_amd.utils.rocm_setup import setup_rocm_perf
+        from maskrcnn_amd._rocm_env import setup_rocm_perf
         setup_rocm_perf(local_rank)
 
     if distributed:

@@ -52,7 +52,7 @@ class Trainer:
         self.augmentation = augmentation
         self.num_workers = num_workers
         self.history = []
-        from ..utils.rocm_setup import setup_rocm_perf
+        from .._rocm_env import setup_rocm_perf
         setup_rocm_perf()
 
     def _autocast(self):
