@@ -41,6 +41,12 @@ def stage_miopen_db(local_rank: int = 0) -> None:
         if not os.path.isdir(stage):
             shutil.copytree(db, stage)
         os.environ["MIOPEN_USER_DB_PATH"] = stage
+        # FAST find: consult the shipped find-db and skip the exhaustive
+        # per-shape search torch's benchmark mode forces (measured: same
+        # 47 ms/step steady state, warmup 139 s -> 41 s). A shape missing
+        # from the db falls back to immediate mode — regenerate the db
+        # (tools/tune_miopen.py) when adding new model configs.
+        os.environ.setdefault("MIOPEN_FIND_MODE", "5")
     except OSError:
         pass
 
