@@ -31,8 +31,7 @@ def stage_miopen_db(local_rank: int = 0) -> None:
         _staged = True
         return
     _staged = True
-    repo_root = os.path.dirname(os.path.dirname(
-        os.path.dirname(os.path.abspath(__file__))))
+    repo_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     db = os.path.join(repo_root, "miopen_db")
     if not os.path.isdir(db):
         return
