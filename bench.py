@@ -20,11 +20,11 @@ import sys
 import time
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
-# MIOPEN_USER_DB_PATH must be exported before libMIOpen loads with torch
-# (profiles/r01_naive_conv.md) — stage the in-repo find-db first.
-from maskrcnn_amd._rocm_env import stage_miopen_db  # noqa: E402
+# MIOPEN_FIND_MODE must be in the environment libMIOpen starts with
+# (profiles/r01_naive_conv.md); re-exec once if the shell didn't set it.
+from maskrcnn_amd._rocm_env import ensure_env_or_reexec  # noqa: E402
 
-stage_miopen_db(int(os.environ.get("LOCAL_RANK", "0")))
+ensure_env_or_reexec()
 
 import numpy as np  # noqa: E402
 import torch  # noqa: E402

@@ -23,9 +23,12 @@ heuristic fallback.
 from __future__ import annotations
 
 import os
+import sys
 
 _staged = False
 _bench_set = False
+
+_ENV = {"MIOPEN_FIND_MODE": "5"}
 
 
 def stage_miopen_env(local_rank: int = 0) -> None:
@@ -34,7 +37,33 @@ def stage_miopen_env(local_rank: int = 0) -> None:
     if _staged:
         return
     _staged = True
-    os.environ.setdefault("MIOPEN_FIND_MODE", "5")
+    for k, v in _ENV.items():
+        os.environ.setdefault(k, v)
+
+
+def ensure_env_or_reexec() -> None:
+    """Guarantee the MIOpen env is visible to libMIOpen by re-exec'ing the
+    interpreter once if needed.
+
+    Empirically (A/B on MI355X), MIOPEN_FIND_MODE set via os.environ before
+    `import torch` is NOT honored (exhaustive Find still runs, +100 s
+    warmup), while the same variable exported by the parent shell is. So
+    entry-point scripts (bench.py, __graft_entry__) call this first: if the
+    variable is missing from the inherited environment, set it and execv
+    the same command line. Library imports never re-exec.
+    """
+    missing = [k for k in _ENV if k not in os.environ]
+    if not missing:  # respect any user-set values
+        return
+    if os.environ.get("MASKRCNN_AMD_NO_REEXEC"):
+        return
+    for k in missing:
+        os.environ[k] = _ENV[k]
+    os.environ["MASKRCNN_AMD_NO_REEXEC"] = "1"  # loop guard
+    try:
+        os.execv(sys.executable, [sys.executable] + sys.argv)
+    except OSError:
+        pass
 
 
 # Backwards-compatible alias (older call sites).
