@@ -1,0 +1,105 @@
+"""Dynamic request batcher.
+
+Replicates TF-Serving's SharedBatchScheduler/BatchingSession behavior
+(SURVEY.md N11, S2 batching_parameters): concurrent Predict calls are
+merged along the batch axis up to max_batch_size, or flushed after
+batch_timeout_micros. Requests are bucketed by molded-image shape (with
+IMAGE_RESIZE_MODE='square' every image is e.g. 1024x1024, so one bucket;
+mixed-size configs get per-shape buckets, SURVEY.md §7 hard-part 5).
+"""
+
+from __future__ import annotations
+
+import queue
+import threading
+from concurrent.futures import Future
+from typing import Dict, Optional
+
+import numpy as np
+
+
+class DynamicBatcher:
+    def __init__(self, engine, max_batch_size: int = 32,
+                 batch_timeout_micros: int = 2000,
+                 max_enqueued_batches: int = 128,
+                 num_batch_threads: int = 1):
+        self.engine = engine
+        self.max_batch_size = max_batch_size
+        self.batch_timeout_s = batch_timeout_micros / 1e6
+        self.queue: "queue.Queue" = queue.Queue(
+            maxsize=max_enqueued_batches * max(1, max_batch_size))
+        self._stop = False
+        self.threads = [
+            threading.Thread(target=self._worker, daemon=True,
+                             name=f"batcher-{i}")
+            for i in range(num_batch_threads)]
+        for t in self.threads:
+            t.start()
+        self.stats = {"batches": 0, "requests": 0, "batch_size_sum": 0}
+
+    def submit(self, input_image: np.ndarray, input_image_meta: np.ndarray,
+               input_anchors: Optional[np.ndarray] = None) -> Future:
+        """Enqueue one request (batch dim must be 1..max_batch_size).
+        Returns a Future resolving to the outputs dict (numpy)."""
+        fut: Future = Future()
+        self.queue.put((input_image, input_image_meta, input_anchors, fut))
+        return fut
+
+    def close(self):
+        self._stop = True
+        for t in self.threads:
+            t.join(timeout=2.0)
+
+    # ------------------------------------------------------------ worker
+    def _worker(self):
+        while not self._stop:
+            try:
+                first = self.queue.get(timeout=0.1)
+            except queue.Empty:
+                continue
+            batch = [first]
+            n = first[0].shape[0]
+            deadline = self.batch_timeout_s
+            import time
+            t0 = time.monotonic()
+            while n < self.max_batch_size:
+                remaining = deadline - (time.monotonic() - t0)
+                if remaining <= 0:
+                    break
+                try:
+                    item = self.queue.get(timeout=remaining)
+                except queue.Empty:
+                    break
+                if item[0].shape[1:] != first[0].shape[1:]:
+                    # Different molded shape: flush current, start new.
+                    self._run(batch)
+                    batch = [item]
+                    n = item[0].shape[0]
+                    t0 = time.monotonic()
+                    continue
+                batch.append(item)
+                n += item[0].shape[0]
+            self._run(batch)
+
+    def _run(self, batch):
+        if not batch:
+            return
+        try:
+            images = np.concatenate([b[0] for b in batch], axis=0)
+            metas = np.concatenate([b[1] for b in batch], axis=0)
+            anchors = batch[0][2]
+            out = self.engine.predict(images, metas, anchors)
+            self.stats["batches"] += 1
+            self.stats["requests"] += len(batch)
+            self.stats["batch_size_sum"] += images.shape[0]
+            off = 0
+            for img, _, _, fut in batch:
+                k = img.shape[0]
+                sliced = {name: arr[off:off + k]
+                          for name, arr in out.items()}
+                fut.set_result(sliced)
+                off += k
+        except Exception as e:  # propagate to every waiter
+            for *_rest, fut in batch:
+                if not fut.done():
+                    fut.set_exception(e)

@@ -1,0 +1,112 @@
+"""Per-GPU inference engine.
+
+The C++ ModelServer's SavedModelBundle+session maps to:
+one engine per GPU running the MaskRCNNModel in inference mode (bf16,
+channels-last) fed by the dynamic batcher. Anchors are computed
+server-side per image shape (fixing the reference's ~4 MB-per-request
+anchor payload wart) but client-provided `input_anchors`
+are honored for signature compatibility.
+"""
+
+from __future__ import annotations
+
+import threading
+from typing import Dict, Optional
+
+import numpy as np
+
+from .._rocm_env import setup_rocm_perf
+from ..config import Config
+from ..utils.anchors import AnchorCache
+
+
+class InferenceEngine:
+    def __init__(self, config: Config, state_dict=None, device=None,
+                 use_bf16: Optional[bool] = None):
+        import torch
+        from ..model.mask_rcnn import MaskRCNNModel
+
+        setup_rocm_perf()
+        self.config = config
+        self.device = torch.device(
+            device or ("cuda" if torch.cuda.is_available() else "cpu"))
+        self.model = MaskRCNNModel(config)
+        if state_dict is not None:
+            self.model.load_state_dict(state_dict)
+        self.model.to(self.device).eval()
+        if config.CHANNELS_LAST and self.device.type == "cuda":
+            self.model = self.model.to(memory_format=torch.channels_last)
+        self.use_bf16 = (use_bf16 if use_bf16 is not None
+                         else (config.COMPUTE_DTYPE == "bf16" and
+                               self.device.type == "cuda"))
+        self.anchor_cache = AnchorCache(config)
+        self._lock = threading.Lock()  # one inference stream per engine
+
+    def predict(self, input_image: np.ndarray,
+                input_image_meta: np.ndarray,
+                input_anchors: Optional[np.ndarray] = None
+                ) -> Dict[str, np.ndarray]:
+        """Run the Predict signature.
+
+        input_image: [B, H, W, 3] molded float32 (NHWC, TF convention).
+        Returns the 7 named outputs as numpy (TF output layout:
+        mrcnn_mask as [B, D, 28, 28, C]).
+        """
+        import torch
+
+        B, H, W, _ = input_image.shape
+        with self._lock, torch.no_grad():
+            images = torch.from_numpy(
+                np.ascontiguousarray(input_image)).float() \
+                .permute(0, 3, 1, 2).to(self.device)
+            if self.config.CHANNELS_LAST and self.device.type == "cuda":
+                images = images.to(memory_format=torch.channels_last)
+            metas = torch.from_numpy(
+                np.ascontiguousarray(input_image_meta)).float() \
+                .to(self.device)
+            if input_anchors is not None:
+                anchors = torch.from_numpy(
+                    np.ascontiguousarray(
+                        input_anchors[0] if input_anchors.ndim == 3
+                        else input_anchors)).float().to(self.device)
+            else:
+                anchors = self.anchor_cache.get((H, W, 3),
+                                                device=self.device)
+            with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                                enabled=self.use_bf16):
+                out = self.model(images, metas, anchors, mode="inference")
+        # Torch [B,D,C,mh,mw] -> TF signature layout [B,D,mh,mw,C].
+        mask = out["mrcnn_mask"].float().permute(0, 1, 3, 4, 2)
+        return {
+            "mrcnn_detection": out["mrcnn_detection"].float().cpu().numpy(),
+            "mrcnn_class": out["mrcnn_class"].float().cpu().numpy(),
+            "mrcnn_bbox": out["mrcnn_bbox"].float().cpu().numpy(),
+            "mrcnn_mask": mask.cpu().numpy(),
+            "rois": out["rois"].float().cpu().numpy(),
+            "rpn_class": out["rpn_class"].float().cpu().numpy(),
+            "rpn_bbox": out["rpn_bbox"].float().cpu().numpy(),
+        }
+
+    @classmethod
+    def from_saved_model(cls, export_dir: str, device=None,
+                         config: Optional[Config] = None):
+        """Build an engine from a SavedModel-layout directory."""
+        import torch
+        from .savedmodel import load_saved_model
+
+        state, config_dict, sigs = load_saved_model(export_dir)
+        if config is None:
+            class LoadedConfig(Config):
+                NAME = str(config_dict.get("NAME", "serving"))
+            config = LoadedConfig()
+            for k, v in config_dict.items():
+                if hasattr(config, k) and not isinstance(
+                        getattr(Config, k, None), property):
+                    try:
+                        setattr(config, k, v)
+                    except Exception:
+                        pass
+            config.__init__()
+        state = {k: (v if isinstance(v, torch.Tensor) else torch.tensor(v))
+                 for k, v in state.items()}
+        return cls(config, state_dict=state, device=device)

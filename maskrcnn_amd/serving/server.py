@@ -1,0 +1,327 @@
+"""ModelServer: gRPC (:8500) + REST (:8501) Predict with version hot-reload.
+
+The MI355X-native replacement for TF-Serving's C++ ModelServer
+(SURVEY.md N11): PredictionService/Predict served via grpc generic
+handlers over the hand-vendored wire protos (wireproto.py), TF-Serving
+REST JSON schema via FastAPI/uvicorn, a FileSystemStoragePathSource-style
+poller that hot-loads the highest numeric version directory, per-GPU
+engines behind the dynamic batcher, and Prometheus metrics.
+"""
+
+from __future__ import annotations
+
+import json
+import logging
+import os
+import threading
+import time
+from typing import Dict, Optional
+
+import numpy as np
+
+from .batcher import DynamicBatcher
+from .engine import InferenceEngine
+from .savedmodel import find_latest_version
+from .wireproto import (ModelSpec, PredictRequest, PredictResponse,
+                        numpy_to_tensor, tensor_to_numpy)
+
+log = logging.getLogger(__name__)
+
+try:
+    from prometheus_client import (Counter, Histogram, Gauge,
+                                   start_http_server)
+    _PROM = True
+    REQS = Counter("predict_requests_total", "Predict requests",
+                   ["proto"])
+    ERRS = Counter("predict_errors_total", "Predict errors", ["proto"])
+    LAT = Histogram("predict_latency_seconds", "Predict latency",
+                    buckets=(.005, .01, .025, .05, .1, .25, .5, 1, 2.5, 5))
+    BATCH = Histogram("predict_batch_size", "executed batch size",
+                      buckets=(1, 2, 4, 8, 16, 32, 64))
+    VERSION = Gauge("model_version", "loaded model version")
+except ImportError:  # pragma: no cover
+    _PROM = False
+
+
+class ModelState:
+    """One served model: engine + batcher + version, hot-swappable."""
+
+    def __init__(self, name: str, base_path: str, device=None,
+                 max_batch_size: int = 32, batch_timeout_micros: int = 2000,
+                 config=None):
+        self.name = name
+        self.base_path = base_path
+        self.device = device
+        self.max_batch_size = max_batch_size
+        self.batch_timeout_micros = batch_timeout_micros
+        self.config_override = config
+        self.version: Optional[int] = None
+        self.engine: Optional[InferenceEngine] = None
+        self.batcher: Optional[DynamicBatcher] = None
+        self._lock = threading.Lock()
+
+    def maybe_reload(self) -> bool:
+        latest = find_latest_version(self.base_path)
+        if latest is None or latest == self.version:
+            return False
+        export_dir = os.path.join(self.base_path, str(latest))
+        log.info("loading %s version %d from %s", self.name, latest,
+                 export_dir)
+        engine = InferenceEngine.from_saved_model(
+            export_dir, device=self.device, config=self.config_override)
+        batcher = DynamicBatcher(
+            engine, max_batch_size=self.max_batch_size,
+            batch_timeout_micros=self.batch_timeout_micros)
+        with self._lock:
+            old = self.batcher
+            self.engine, self.batcher, self.version = \
+                engine, batcher, latest
+        if old is not None:
+            old.close()
+        if _PROM:
+            VERSION.set(latest)
+        return True
+
+    def predict(self, images, metas, anchors=None):
+        with self._lock:
+            batcher = self.batcher
+            version = self.version
+        if batcher is None:
+            raise RuntimeError(f"model {self.name} has no loaded version")
+        fut = batcher.submit(images, metas, anchors)
+        return fut.result(timeout=60.0), version
+
+
+class ModelServer:
+    def __init__(self, model_name: str, model_base_path: str,
+                 grpc_port: int = 8500, rest_port: int = 8501,
+                 metrics_port: Optional[int] = None, device=None,
+                 max_batch_size: int = 32, batch_timeout_micros: int = 2000,
+                 poll_interval_s: float = 5.0, config=None):
+        self.state = ModelState(model_name, model_base_path, device=device,
+                                max_batch_size=max_batch_size,
+                                batch_timeout_micros=batch_timeout_micros,
+                                config=config)
+        self.grpc_port = grpc_port
+        self.rest_port = rest_port
+        self.metrics_port = metrics_port
+        self.poll_interval_s = poll_interval_s
+        self._stop = threading.Event()
+        self._grpc_server = None
+        self._rest_thread = None
+        self._uvicorn_server = None
+
+    # ----------------------------------------------------------- predict
+    def _handle_predict(self, request: PredictRequest) -> PredictResponse:
+        t0 = time.monotonic()
+        if _PROM:
+            REQS.labels("grpc").inc()
+        name = request.model_spec.name or self.state.name
+        if name != self.state.name:
+            raise KeyError(f"unknown model {name!r}")
+        sig = request.model_spec.signature_name or "serving_default"
+        if sig != "serving_default":
+            raise KeyError(f"unknown signature {sig!r}")
+        try:
+            images = tensor_to_numpy(request.inputs["input_image"]) \
+                .astype(np.float32)
+            metas = tensor_to_numpy(request.inputs["input_image_meta"]) \
+                .astype(np.float32)
+        except KeyError as e:
+            raise KeyError(f"missing input {e}")
+        anchors = None
+        if "input_anchors" in request.inputs:
+            anchors = tensor_to_numpy(
+                request.inputs["input_anchors"]).astype(np.float32)
+        outputs, version = self.state.predict(images, metas, anchors)
+        if request.output_filter:
+            outputs = {k: v for k, v in outputs.items()
+                       if k in request.output_filter}
+        resp = PredictResponse(
+            outputs={k: numpy_to_tensor(v) for k, v in outputs.items()},
+            model_spec=ModelSpec(name=self.state.name, version=version,
+                                 signature_name="serving_default"))
+        if _PROM:
+            LAT.observe(time.monotonic() - t0)
+            BATCH.observe(images.shape[0])
+        return resp
+
+    # -------------------------------------------------------------- grpc
+    def _build_grpc(self):
+        import grpc
+        from concurrent import futures as cf
+
+        outer = self
+
+        def predict(request: PredictRequest, context):
+            try:
+                return outer._handle_predict(request)
+            except KeyError as e:
+                if _PROM:
+                    ERRS.labels("grpc").inc()
+                context.abort(grpc.StatusCode.NOT_FOUND, str(e))
+            except Exception as e:  # noqa: BLE001
+                if _PROM:
+                    ERRS.labels("grpc").inc()
+                log.exception("Predict failed")
+                context.abort(grpc.StatusCode.INTERNAL, str(e))
+
+        handler = grpc.method_handlers_generic_handler(
+            "tensorflow.serving.PredictionService",
+            {"Predict": grpc.unary_unary_rpc_method_handler(
+                predict,
+                request_deserializer=PredictRequest.decode,
+                response_serializer=lambda r: r.encode())})
+        server = grpc.server(
+            cf.ThreadPoolExecutor(max_workers=16),
+            options=[("grpc.max_receive_message_length", 256 * 1024 * 1024),
+                     ("grpc.max_send_message_length", 256 * 1024 * 1024)])
+        server.add_generic_rpc_handlers((handler,))
+        bound = server.add_insecure_port(f"[::]:{self.grpc_port}")
+        self.grpc_port = bound
+        return server
+
+    # -------------------------------------------------------------- rest
+    def _build_rest_app(self):
+        from fastapi import FastAPI, Request
+        from fastapi.responses import JSONResponse
+
+        app = FastAPI(title="maskrcnn_amd model server")
+        outer = self
+
+        @app.get("/v1/models/{model_name}")
+        def model_status(model_name: str):
+            ok = (model_name == outer.state.name and
+                  outer.state.version is not None)
+            return {"model_version_status": [{
+                "version": str(outer.state.version or 0),
+                "state": "AVAILABLE" if ok else "UNAVAILABLE",
+                "status": {"error_code": "OK" if ok else "NOT_FOUND",
+                           "error_message": ""},
+            }]}
+
+        @app.post("/v1/models/{model_and_verb:path}")
+        async def predict(model_and_verb: str, request: Request):
+            if _PROM:
+                REQS.labels("rest").inc()
+            if not model_and_verb.endswith(":predict"):
+                return JSONResponse({"error": "unsupported verb"},
+                                    status_code=404)
+            name = model_and_verb[:-len(":predict")]
+            if name != outer.state.name:
+                return JSONResponse(
+                    {"error": f"model {name} not found"}, status_code=404)
+            body = await request.json()
+            try:
+                if "inputs" in body:       # columnar format
+                    ins = body["inputs"]
+                    images = np.asarray(ins["input_image"],
+                                        dtype=np.float32)
+                    metas = np.asarray(ins["input_image_meta"],
+                                       dtype=np.float32)
+                    anchors = (np.asarray(ins["input_anchors"],
+                                          dtype=np.float32)
+                               if "input_anchors" in ins else None)
+                elif "instances" in body:  # row format
+                    rows = body["instances"]
+                    images = np.stack([np.asarray(r["input_image"],
+                                                  dtype=np.float32)
+                                       for r in rows])
+                    metas = np.stack([np.asarray(r["input_image_meta"],
+                                                 dtype=np.float32)
+                                      for r in rows])
+                    anchors = None
+                    if rows and "input_anchors" in rows[0]:
+                        anchors = np.asarray(rows[0]["input_anchors"],
+                                             dtype=np.float32)[None]
+                else:
+                    return JSONResponse(
+                        {"error": "missing inputs/instances"},
+                        status_code=400)
+                t0 = time.monotonic()
+                outputs, _ = outer.state.predict(images, metas, anchors)
+                if _PROM:
+                    LAT.observe(time.monotonic() - t0)
+                return {"outputs": {k: v.tolist()
+                                    for k, v in outputs.items()}}
+            except Exception as e:  # noqa: BLE001
+                if _PROM:
+                    ERRS.labels("rest").inc()
+                log.exception("REST predict failed")
+                return JSONResponse({"error": str(e)}, status_code=500)
+
+        return app
+
+    # --------------------------------------------------------- lifecycle
+    def start(self, wait_for_model: bool = True):
+        self.state.maybe_reload()
+        if wait_for_model and self.state.version is None:
+            raise FileNotFoundError(
+                f"no servable version under {self.state.base_path}")
+        if self.metrics_port and _PROM:
+            start_http_server(self.metrics_port)
+        self._grpc_server = self._build_grpc()
+        self._grpc_server.start()
+
+        if self.rest_port is not None:
+            import uvicorn
+            app = self._build_rest_app()
+            config = uvicorn.Config(app, host="0.0.0.0",
+                                    port=self.rest_port,
+                                    log_level="warning")
+            self._uvicorn_server = uvicorn.Server(config)
+            self._rest_thread = threading.Thread(
+                target=self._uvicorn_server.run, daemon=True)
+            self._rest_thread.start()
+
+        self._poll_thread = threading.Thread(target=self._poll_loop,
+                                             daemon=True)
+        self._poll_thread.start()
+        log.info("ModelServer up: grpc=:%d rest=:%s model=%s v%s",
+                 self.grpc_port, self.rest_port, self.state.name,
+                 self.state.version)
+
+    def _poll_loop(self):
+        while not self._stop.wait(self.poll_interval_s):
+            try:
+                self.state.maybe_reload()
+            except Exception:  # noqa: BLE001
+                log.exception("version reload failed")
+
+    def stop(self, grace: float = 1.0):
+        self._stop.set()
+        if self._grpc_server is not None:
+            self._grpc_server.stop(grace).wait()
+        if self._uvicorn_server is not None:
+            self._uvicorn_server.should_exit = True
+        if self.state.batcher is not None:
+            self.state.batcher.close()
+
+    def wait(self):
+        self._grpc_server.wait_for_termination()
+
+
+def main():  # pragma: no cover
+    import argparse
+    ap = argparse.ArgumentParser(description="maskrcnn_amd model server")
+    ap.add_argument("--model_name", default="mask_rcnn")
+    ap.add_argument("--model_base_path", required=True)
+    ap.add_argument("--port", type=int, default=8500)
+    ap.add_argument("--rest_api_port", type=int, default=8501)
+    ap.add_argument("--metrics_port", type=int, default=None)
+    ap.add_argument("--enable_batching", action="store_true")
+    ap.add_argument("--max_batch_size", type=int, default=32)
+    ap.add_argument("--batch_timeout_micros", type=int, default=2000)
+    args = ap.parse_args()
+    logging.basicConfig(level=logging.INFO)
+    server = ModelServer(
+        args.model_name, args.model_base_path, grpc_port=args.port,
+        rest_port=args.rest_api_port, metrics_port=args.metrics_port,
+        max_batch_size=args.max_batch_size if args.enable_batching else 1,
+        batch_timeout_micros=args.batch_timeout_micros)
+    server.start()
+    server.wait()
+
+
+if __name__ == "__main__":  # pragma: no cover
+    main()

@@ -1,0 +1,81 @@
+"""REST (:8501-style) Predict round trip against the FastAPI facade."""
+
+import socket
+import time
+
+import numpy as np
+import pytest
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+@pytest.fixture(scope="module")
+def rest_server(tmp_path_factory):
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+    from maskrcnn_amd.serving.server import ModelServer
+
+    class SmallConfig(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 500
+        POST_NMS_ROIS_INFERENCE = 50
+        DETECTION_MIN_CONFIDENCE = 0.0
+
+    cfg = SmallConfig()
+    tmp = tmp_path_factory.mktemp("rest")
+    m = MaskRCNN("inference", cfg, model_dir=str(tmp))
+    base = str(tmp / "models" / "mask_rcnn")
+    export_saved_model(m, cfg, base, version=1)
+    port = _free_port()
+    server = ModelServer("mask_rcnn", base, grpc_port=0, rest_port=port,
+                         max_batch_size=2, batch_timeout_micros=1000,
+                         config=cfg)
+    server.start()
+    # Wait for uvicorn to come up.
+    import requests
+    for _ in range(100):
+        try:
+            requests.get(f"http://127.0.0.1:{port}/v1/models/mask_rcnn",
+                         timeout=1)
+            break
+        except Exception:
+            time.sleep(0.1)
+    yield server, cfg, port
+    server.stop()
+
+
+def test_rest_status(rest_server):
+    import requests
+    server, cfg, port = rest_server
+    r = requests.get(f"http://127.0.0.1:{port}/v1/models/mask_rcnn",
+                     timeout=10)
+    assert r.status_code == 200
+    st = r.json()["model_version_status"][0]
+    assert st["state"] == "AVAILABLE"
+    assert st["version"] == "1"
+
+
+def test_rest_predict_columnar(rest_server):
+    server, cfg, port = rest_server
+    from maskrcnn_amd.serving.client import RestServingClient
+    client = RestServingClient(f"http://127.0.0.1:{port}", cfg)
+    img = (np.random.rand(100, 90, 3) * 255).astype(np.uint8)
+    results = client.detect([img])
+    r = results[0]
+    assert r["rois"].shape[1] == 4
+    assert r["masks"].shape[:2] == img.shape[:2]
+
+
+def test_rest_unknown_model_404(rest_server):
+    import requests
+    server, cfg, port = rest_server
+    r = requests.post(f"http://127.0.0.1:{port}/v1/models/nope:predict",
+                      json={"inputs": {}}, timeout=10)
+    assert r.status_code == 404
