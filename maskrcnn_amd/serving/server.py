@@ -183,7 +183,7 @@ class ModelServer:
 
     # -------------------------------------------------------------- rest
     def _build_rest_app(self):
-        from fastapi import FastAPI, Request
+        from fastapi import Body, FastAPI
         from fastapi.responses import JSONResponse
 
         app = FastAPI(title="maskrcnn_amd model server")
@@ -201,7 +201,7 @@ class ModelServer:
             }]}
 
         @app.post("/v1/models/{model_and_verb:path}")
-        async def predict(model_and_verb: str, request: Request):
+        async def predict(model_and_verb: str, body: dict = Body(...)):
             if _PROM:
                 REQS.labels("rest").inc()
             if not model_and_verb.endswith(":predict"):
@@ -211,7 +211,6 @@ class ModelServer:
             if name != outer.state.name:
                 return JSONResponse(
                     {"error": f"model {name} not found"}, status_code=404)
-            body = await request.json()
             try:
                 if "inputs" in body:       # columnar format
                     ins = body["inputs"]
