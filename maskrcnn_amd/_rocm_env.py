@@ -2,22 +2,24 @@
 
 MIOpen's immediate mode (torch default) resolves most gfx950 bf16-NHWC
 conv shapes to NAIVE fallback kernels — measured at 93% of the training
-step (profiles/r01_naive_conv.md). Two measured fixes:
+step (profiles/r01_naive_conv.md). Fixes:
 
   1. MIOPEN_FIND_MODE=5 (FAST): MIOpen's heuristic solver selection picks
-     the tuned CK/iGEMM solvers without any search. Must be exported
-     BEFORE libMIOpen loads (i.e. before `import torch`) —
-     stage_miopen_env() below is torch-free and invoked from
-     maskrcnn_amd/__init__ and bench.py.
-  2. torch.backends.cudnn.benchmark = True.
+     the tuned CK/iGEMM solvers with no search. Empirically the variable
+     is only honored when it is in the environment the PROCESS started
+     with (A/B on MI355X: python-set os.environ before `import torch` is
+     ignored; shell-exported works). Entry-point scripts therefore call
+     ensure_env_or_reexec() which re-execs the interpreter once with the
+     variable exported. Library imports never re-exec and never set the
+     variable (a python-side setdefault would defeat the re-exec check
+     while doing nothing for libMIOpen).
+  2. torch.backends.cudnn.benchmark = True (per-shape Find for any shape
+     the heuristics cover badly; warmup-time cost only).
 
 Measured (R101-FPN bf16 train, batch 2, 1024², 1 MI355X):
   immediate mode (naive convs):        61-64 ms/step
-  FIND_MODE=5 + benchmark:             47.4 ms/step, warmup ~10 s
+  FIND_MODE=5 + benchmark:             47.2 ms/step, warmup ~10 s
   exhaustive Find + benchmark:         46.5 ms/step, warmup ~100 s
-A shipped user find-db was tried and REJECTED: the db recorded naive
-winners for several wrw shapes and made steady state worse than the
-heuristic fallback.
 """
 
 from __future__ import annotations
@@ -25,45 +27,36 @@ from __future__ import annotations
 import os
 import sys
 
-_staged = False
 _bench_set = False
 
 _ENV = {"MIOPEN_FIND_MODE": "5"}
-
-
-def stage_miopen_env(local_rank: int = 0) -> None:
-    """Export MIOpen env knobs. Torch-free: call before `import torch`."""
-    global _staged
-    if _staged:
-        return
-    _staged = True
-    for k, v in _ENV.items():
-        os.environ.setdefault(k, v)
+_SENTINEL = "MASKRCNN_AMD_NO_REEXEC"
 
 
 def ensure_env_or_reexec() -> None:
-    """Guarantee the MIOpen env is visible to libMIOpen by re-exec'ing the
-    interpreter once if needed.
+    """Re-exec the interpreter once with the MIOpen env exported.
 
-    Empirically (A/B on MI355X), MIOPEN_FIND_MODE set via os.environ before
-    `import torch` is NOT honored (exhaustive Find still runs, +100 s
-    warmup), while the same variable exported by the parent shell is. So
-    entry-point scripts (bench.py, __graft_entry__) call this first: if the
-    variable is missing from the inherited environment, set it and execv
-    the same command line. Library imports never re-exec.
+    Call this FIRST in entry-point scripts (before importing torch or any
+    module that does). No-ops when the parent shell already exported the
+    variables, when re-exec already happened (sentinel), or on failure.
     """
-    missing = [k for k in _ENV if k not in os.environ]
-    if not missing:  # respect any user-set values
+    if os.environ.get(_SENTINEL):
         return
-    if os.environ.get("MASKRCNN_AMD_NO_REEXEC"):
+    missing = [k for k in _ENV if k not in os.environ]
+    if not missing:  # parent shell set them — nothing to do
         return
     for k in missing:
         os.environ[k] = _ENV[k]
-    os.environ["MASKRCNN_AMD_NO_REEXEC"] = "1"  # loop guard
+    os.environ[_SENTINEL] = "1"  # loop guard
     try:
         os.execv(sys.executable, [sys.executable] + sys.argv)
     except OSError:
         pass
+
+
+def stage_miopen_env(local_rank: int = 0) -> None:
+    """Deprecated no-op kept for import compatibility (see module doc:
+    python-side env writes do not reach libMIOpen)."""
 
 
 # Backwards-compatible alias (older call sites).
@@ -71,9 +64,8 @@ stage_miopen_db = stage_miopen_env
 
 
 def setup_rocm_perf(local_rank: int = 0) -> None:
-    """Full setup: env staging + cudnn.benchmark (call after torch import)."""
+    """Torch-side knobs (call after torch import)."""
     global _bench_set
-    stage_miopen_env(local_rank)
     if _bench_set:
         return
     _bench_set = True

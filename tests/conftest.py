@@ -1,3 +1,16 @@
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
+
+if os.path.exists("/dev/kfd"):
+    # GPU box: export MIOPEN_FIND_MODE before torch loads libMIOpen
+    # (maskrcnn_amd/_rocm_env.py) — avoids naive-conv fallback in GPU tests.
+    from maskrcnn_amd._rocm_env import ensure_env_or_reexec
+
+    ensure_env_or_reexec()
+
 import numpy as np
 import pytest
 import torch
