@@ -104,6 +104,76 @@ def make_synthetic_batch(config, device, seed=0):
     }
 
 
+def _phase_timing(model, config, batch, anchors, meta, optimizer,
+                  use_bf16, n_iters=5):
+    """Per-phase wall breakdown of a training step (GPU-synced per phase;
+    adds sync overhead, so the per-phase SUM exceeds the free-running
+    step time — use it for ranking, not absolutes)."""
+    import time as _t
+
+    from maskrcnn_amd.model.losses import compute_losses
+
+    def sync():
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+
+    acc = {}
+
+    def mark(name, t0):
+        sync()
+        t1 = _t.perf_counter()
+        acc[name] = acc.get(name, 0.0) + (t1 - t0)
+        return t1
+
+    for _ in range(n_iters):
+        sync()
+        t = _t.perf_counter()
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                            enabled=use_bf16):
+            feats = model.extract_features(batch["image"])
+            t = mark("backbone+fpn", t)
+            rpn_logits, rpn_class, rpn_bbox = model.rpn(feats)
+            t = mark("rpn", t)
+            rois_p = model.proposal_layer(rpn_class, rpn_bbox, anchors,
+                                          config.POST_NMS_ROIS_TRAINING)
+            t = mark("proposal(topk+nms)", t)
+            rois, tci, tdel, tmask = model.detection_target_layer(
+                rois_p, batch["gt_class_ids"], batch["gt_boxes"],
+                batch["gt_masks"])
+            t = mark("detection_targets", t)
+            cl_logits, cl_probs, cl_bbox = model.classifier_head(
+                rois, feats[:4])
+            t = mark("classifier_head", t)
+            mask_logits = model.mask_head(rois, feats[:4])
+            t = mark("mask_head", t)
+            losses = compute_losses(
+                config,
+                {"rpn_match": batch["rpn_match"],
+                 "rpn_bbox": batch["rpn_bbox"],
+                 "active_class_ids": meta["active_class_ids"]},
+                {"rpn_class_logits": rpn_logits, "rpn_bbox": rpn_bbox,
+                 "target_class_ids": tci, "target_deltas": tdel,
+                 "target_masks": tmask, "mrcnn_class_logits": cl_logits,
+                 "mrcnn_bbox": cl_bbox, "mrcnn_mask": mask_logits})
+            t = mark("losses", t)
+        optimizer.zero_grad(set_to_none=True)
+        losses["total_loss"].backward()
+        t = mark("backward", t)
+        torch.nn.utils.clip_grad_norm_(
+            (p for p in model.parameters() if p.requires_grad),
+            config.GRADIENT_CLIP_NORM)
+        optimizer.step()
+        t = mark("clip+opt_step", t)
+    total = sum(acc.values())
+    print("=== phase timing (avg ms over "
+          f"{n_iters} iters; per-phase synced) ===", file=sys.stderr)
+    for k, v in sorted(acc.items(), key=lambda kv: -kv[1]):
+        print(f"  {k:22s} {1000 * v / n_iters:8.2f} ms "
+              f"({100 * v / total:5.1f}%)", file=sys.stderr)
+    print(f"  {'TOTAL':22s} {1000 * total / n_iters:8.2f} ms",
+          file=sys.stderr)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -114,6 +184,8 @@ def main():
     ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     ap.add_argument("--backend", default=None,
                     help="override dist backend (nccl|gloo)")
+    ap.add_argument("--phase-timing", action="store_true",
+                    help="print a per-phase GPU/wall breakdown to stderr")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -188,6 +260,10 @@ def main():
 
     for _ in range(args.warmup):
         one_step()
+
+    if args.phase_timing and rank == 0:
+        _phase_timing(model, config, batch, anchors, meta, optimizer,
+                      use_bf16, n_iters=5)
 
     if distributed:
         import torch.distributed as dist

@@ -67,7 +67,7 @@ def main():
                          max_batch_size=args.max_batch,
                          batch_timeout_micros=args.batch_timeout_micros,
                          config=cfg)
-    server.start()
+    server.start(warmup=True)
 
     client = GrpcServingClient(f"localhost:{server.grpc_port}", cfg,
                                send_anchors=args.send_anchors)

@@ -22,9 +22,16 @@ class DynamicBatcher:
     def __init__(self, engine, max_batch_size: int = 32,
                  batch_timeout_micros: int = 2000,
                  max_enqueued_batches: int = 128,
-                 num_batch_threads: int = 1):
+                 num_batch_threads: int = 1,
+                 pad_to_max: bool = True):
         self.engine = engine
         self.max_batch_size = max_batch_size
+        # TF-Serving's BatchingSession pads every batch to max_batch_size
+        # (SURVEY.md N11). Besides matching that behavior, padding keeps
+        # the conv batch shape FIXED, which matters on MIOpen: every new
+        # batch size triggers a per-shape Find (tens of seconds at
+        # 1024²x32). One shape = one warmup.
+        self.pad_to_max = pad_to_max
         self.batch_timeout_s = batch_timeout_micros / 1e6
         self.queue: "queue.Queue" = queue.Queue(
             maxsize=max_enqueued_batches * max(1, max_batch_size))
@@ -88,7 +95,19 @@ class DynamicBatcher:
             images = np.concatenate([b[0] for b in batch], axis=0)
             metas = np.concatenate([b[1] for b in batch], axis=0)
             anchors = batch[0][2]
+            real_n = images.shape[0]
+            if self.pad_to_max and real_n < self.max_batch_size:
+                pad = self.max_batch_size - real_n
+                images = np.concatenate(
+                    [images, np.zeros((pad,) + images.shape[1:],
+                                      images.dtype)], axis=0)
+                # Padded metas replicate row 0 (real) so shape fields
+                # parsed from the meta stay valid.
+                metas = np.concatenate(
+                    [metas, np.repeat(metas[:1], pad, axis=0)], axis=0)
             out = self.engine.predict(images, metas, anchors)
+            if self.pad_to_max:
+                out = {k: v[:real_n] for k, v in out.items()}
             self.stats["batches"] += 1
             self.stats["requests"] += len(batch)
             self.stats["batch_size_sum"] += images.shape[0]

@@ -89,7 +89,21 @@ class ModelState:
         if batcher is None:
             raise RuntimeError(f"model {self.name} has no loaded version")
         fut = batcher.submit(images, metas, anchors)
-        return fut.result(timeout=60.0), version
+        return fut.result(timeout=300.0), version
+
+    def warmup(self, image_shape=None):
+        """Run one dummy max-batch predict so MIOpen Find / algo caches are
+        hot before traffic (TF-Serving model_warmup equivalent)."""
+        eng = self.engine
+        if eng is None:
+            return
+        H, W, _ = image_shape or eng.config.IMAGE_SHAPE
+        B = self.max_batch_size
+        images = np.zeros((B, H, W, 3), np.float32)
+        metas = np.zeros((B, eng.config.IMAGE_META_SIZE), np.float32)
+        metas[:, 4:7] = (H, W, 3)
+        metas[:, 7:11] = (0, 0, H, W)
+        eng.predict(images, metas, None)
 
 
 class ModelServer:
@@ -252,11 +266,14 @@ class ModelServer:
         return app
 
     # --------------------------------------------------------- lifecycle
-    def start(self, wait_for_model: bool = True):
+    def start(self, wait_for_model: bool = True, warmup: bool = False):
         self.state.maybe_reload()
         if wait_for_model and self.state.version is None:
             raise FileNotFoundError(
                 f"no servable version under {self.state.base_path}")
+        if warmup:
+            log.info("warmup predict at max batch size...")
+            self.state.warmup()
         if self.metrics_port and _PROM:
             start_http_server(self.metrics_port)
         self._grpc_server = self._build_grpc()
