@@ -30,6 +30,7 @@ import numpy as np  # noqa: E402
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--requests", type=int, default=64)
+    ap.add_argument("--client-timeout", type=float, default=300.0)
     ap.add_argument("--concurrency", type=int, default=16)
     ap.add_argument("--warmup", type=int, default=8)
     ap.add_argument("--image-size", type=int, default=1024)
@@ -40,6 +41,11 @@ def main():
                     help="include the [A,4] anchor tensor in each request "
                          "(reference client behavior)")
     args = ap.parse_args()
+
+    import logging
+    logging.basicConfig(
+        level=logging.INFO, stream=sys.stderr,
+        format="%(asctime)s %(name)s %(message)s")
 
     import torch
     from maskrcnn_amd.config import Config
@@ -70,7 +76,8 @@ def main():
     server.start(warmup=True)
 
     client = GrpcServingClient(f"localhost:{server.grpc_port}", cfg,
-                               send_anchors=args.send_anchors)
+                               send_anchors=args.send_anchors,
+                               timeout=args.client_timeout)
     # Pre-mold one synthetic image; clients reuse the molded arrays
     # (the serving benchmark measures the server, not PIL).
     img = (np.random.rand(800, 1000, 3) * 255).astype(np.uint8)

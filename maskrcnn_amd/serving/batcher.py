@@ -10,12 +10,16 @@ mixed-size configs get per-shape buckets, SURVEY.md §7 hard-part 5).
 
 from __future__ import annotations
 
+import logging
 import queue
 import threading
+import time
 from concurrent.futures import Future
 from typing import Dict, Optional
 
 import numpy as np
+
+log = logging.getLogger(__name__)
 
 
 class DynamicBatcher:
@@ -105,7 +109,10 @@ class DynamicBatcher:
                 # parsed from the meta stay valid.
                 metas = np.concatenate(
                     [metas, np.repeat(metas[:1], pad, axis=0)], axis=0)
+            t0 = time.monotonic()
             out = self.engine.predict(images, metas, anchors)
+            log.info("batch of %d (padded %d) in %.3fs", real_n,
+                     images.shape[0], time.monotonic() - t0)
             if self.pad_to_max:
                 out = {k: v[:real_n] for k, v in out.items()}
             self.stats["batches"] += 1
