@@ -39,6 +39,22 @@ from .targets import DetectionTargetLayer
 class MaskRCNNModel(nn.Module):
     """The network graph (reference MaskRCNN.build, model.py:~1750-2030)."""
 
+    #: when True, forward() logs a GPU-synced wall time per stage
+    #: (diagnostic only; adds syncs).
+    phase_log = False
+
+    def _mark(self, name, t_prev):
+        if not self.phase_log:
+            return t_prev
+        import logging
+        import time
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        t = time.perf_counter()
+        logging.getLogger(__name__).info(
+            "phase %-18s %8.3f s", name, t - t_prev)
+        return t
+
     def __init__(self, config: Config):
         super().__init__()
         self.config = config
@@ -74,16 +90,21 @@ class MaskRCNNModel(nn.Module):
         Returns a dict of named outputs matching the serving signature.
         """
         cfg = self.config
+        import time as _time
+        t = _time.perf_counter() if self.phase_log else 0
         p2, p3, p4, p5, p6 = self.extract_features(images)
+        t = self._mark("backbone+fpn", t)
         rpn_feature_maps = [p2, p3, p4, p5, p6]
         mrcnn_feature_maps = [p2, p3, p4, p5]
 
         rpn_class_logits, rpn_class, rpn_bbox = self.rpn(rpn_feature_maps)
+        t = self._mark("rpn", t)
 
         proposal_count = (cfg.POST_NMS_ROIS_TRAINING if mode == "training"
                           else cfg.POST_NMS_ROIS_INFERENCE)
         rpn_rois = self.proposal_layer(rpn_class, rpn_bbox, anchors,
                                        proposal_count)
+        t = self._mark("proposal", t)
 
         if mode == "training":
             rois, target_class_ids, target_deltas, target_masks = \
@@ -109,12 +130,15 @@ class MaskRCNNModel(nn.Module):
         else:
             mrcnn_class_logits, mrcnn_class, mrcnn_bbox = \
                 self.classifier_head(rpn_rois, mrcnn_feature_maps)
+            t = self._mark("classifier_head", t)
             detections = self.detection_layer(rpn_rois, mrcnn_class,
                                               mrcnn_bbox, image_meta)
+            t = self._mark("detection_layer", t)
             detection_boxes = detections[..., :4]
             # Mask head runs on the FINAL detections (reference :~2000).
             mrcnn_mask = torch.sigmoid(
                 self.mask_head(detection_boxes, mrcnn_feature_maps))
+            t = self._mark("mask_head", t)
             return {
                 "mrcnn_detection": detections,
                 "mrcnn_class": mrcnn_class,
