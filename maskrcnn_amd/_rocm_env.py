@@ -71,4 +71,9 @@ def setup_rocm_perf(local_rank: int = 0) -> None:
     _bench_set = True
     import torch
     if torch.cuda.is_available():
-        torch.backends.cudnn.benchmark = True
+        # With MIOPEN_FIND_MODE=5, benchmark=False routes torch through
+        # MIOpen's non-exhaustive Find (heuristic selection, no search);
+        # benchmark=True forces the 100-265 s exhaustive search per shape.
+        # MASKRCNN_CUDNN_BENCHMARK=1 re-enables the exhaustive path.
+        torch.backends.cudnn.benchmark = bool(
+            os.environ.get("MASKRCNN_CUDNN_BENCHMARK"))
