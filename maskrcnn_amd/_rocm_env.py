@@ -71,9 +71,12 @@ def setup_rocm_perf(local_rank: int = 0) -> None:
     _bench_set = True
     import torch
     if torch.cuda.is_available():
-        # With MIOPEN_FIND_MODE=5, benchmark=False routes torch through
-        # MIOpen's non-exhaustive Find (heuristic selection, no search);
-        # benchmark=True forces the 100-265 s exhaustive search per shape.
-        # MASKRCNN_CUDNN_BENCHMARK=1 re-enables the exhaustive path.
-        torch.backends.cudnn.benchmark = bool(
-            os.environ.get("MASKRCNN_CUDNN_BENCHMARK"))
+        # benchmark=True: exhaustive MIOpen Find per conv shape at first
+        # use (~100 s warmup for the R101 train step, once per process) →
+        # 46.5 ms/step steady. benchmark=False (immediate mode) resolves
+        # many bf16-NHWC shapes to naive kernels → 60-64 ms/step. The
+        # Find results land in MIOpen's user find-db, so later processes
+        # on the same machine skip the search entirely.
+        # MASKRCNN_NO_CUDNN_BENCHMARK=1 opts out.
+        torch.backends.cudnn.benchmark = not bool(
+            os.environ.get("MASKRCNN_NO_CUDNN_BENCHMARK"))

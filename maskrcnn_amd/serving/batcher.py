@@ -110,11 +110,10 @@ class DynamicBatcher:
                 metas = np.concatenate(
                     [metas, np.repeat(metas[:1], pad, axis=0)], axis=0)
             t0 = time.monotonic()
-            out = self.engine.predict(images, metas, anchors)
+            out = self.engine.predict(images, metas, anchors,
+                                      trim_to=real_n)
             log.info("batch of %d (padded %d) in %.3fs", real_n,
                      images.shape[0], time.monotonic() - t0)
-            if self.pad_to_max:
-                out = {k: v[:real_n] for k, v in out.items()}
             self.stats["batches"] += 1
             self.stats["requests"] += len(batch)
             self.stats["batch_size_sum"] += images.shape[0]

@@ -44,7 +44,8 @@ class InferenceEngine:
 
     def predict(self, input_image: np.ndarray,
                 input_image_meta: np.ndarray,
-                input_anchors: Optional[np.ndarray] = None
+                input_anchors: Optional[np.ndarray] = None,
+                trim_to: Optional[int] = None
                 ) -> Dict[str, np.ndarray]:
         """Run the Predict signature.
 
@@ -75,17 +76,24 @@ class InferenceEngine:
             with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
                                 enabled=self.use_bf16):
                 out = self.model(images, metas, anchors, mode="inference")
-        # Torch [B,D,C,mh,mw] -> TF signature layout [B,D,mh,mw,C].
-        mask = out["mrcnn_mask"].float().permute(0, 1, 3, 4, 2)
-        return {
-            "mrcnn_detection": out["mrcnn_detection"].float().cpu().numpy(),
-            "mrcnn_class": out["mrcnn_class"].float().cpu().numpy(),
-            "mrcnn_bbox": out["mrcnn_bbox"].float().cpu().numpy(),
-            "mrcnn_mask": mask.cpu().numpy(),
-            "rois": out["rois"].float().cpu().numpy(),
-            "rpn_class": out["rpn_class"].float().cpu().numpy(),
-            "rpn_bbox": out["rpn_bbox"].float().cpu().numpy(),
-        }
+            # Trim the padded batch rows GPU-side BEFORE DtoH: a padded
+            # batch-32 output set is ~1 GB; real rows are what the caller
+            # needs (batcher pad-to-max).
+            k = trim_to if trim_to is not None else B
+            # Torch [B,D,C,mh,mw] -> TF signature layout [B,D,mh,mw,C].
+            mask = out["mrcnn_mask"][:k].float().permute(0, 1, 3, 4, 2)
+            return {
+                "mrcnn_detection":
+                    out["mrcnn_detection"][:k].float().cpu().numpy(),
+                "mrcnn_class":
+                    out["mrcnn_class"][:k].float().cpu().numpy(),
+                "mrcnn_bbox":
+                    out["mrcnn_bbox"][:k].float().cpu().numpy(),
+                "mrcnn_mask": mask.contiguous().cpu().numpy(),
+                "rois": out["rois"][:k].float().cpu().numpy(),
+                "rpn_class": out["rpn_class"][:k].float().cpu().numpy(),
+                "rpn_bbox": out["rpn_bbox"][:k].float().cpu().numpy(),
+            }
 
     @classmethod
     def from_saved_model(cls, export_dir: str, device=None,
