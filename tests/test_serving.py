@@ -181,10 +181,13 @@ class _FakeEngine:
     def __init__(self):
         self.calls = []
 
-    def predict(self, images, metas, anchors=None):
+    def predict(self, images, metas, anchors=None, trim_to=None):
         self.calls.append(images.shape[0])
         time.sleep(0.01)
-        return {"out": images.sum(axis=(1, 2, 3))}
+        out = images.sum(axis=(1, 2, 3))
+        if trim_to is not None:
+            out = out[:trim_to]
+        return {"out": out}
 
 
 def test_batcher_merges_requests():
