@@ -72,7 +72,8 @@ def make_synthetic_batch(config, device, seed=0):
                         dtype=np.float32)
 
     for b in range(B):
-        n = rng.randint(8, 30)  # typical COCO instance count range
+        # Typical COCO instance count range, capped by config.
+        n = min(int(rng.randint(8, 30)), config.MAX_GT_INSTANCES)
         ys = rng.uniform(0, H - 64, n)
         xs = rng.uniform(0, W - 64, n)
         hs = rng.uniform(32, H / 3, n)
