@@ -21,6 +21,8 @@ void roi_align_bwd_launch(const void*, const float*, const int*, float*, int,
 long nms_workspace_bytes(int N);
 void nms_launch(const float*, int, float, int, void*, int*, int*,
                 hipStream_t);
+void fpn_fuse_launch(const void*, const void*, const float*, const void*,
+                     void*, int, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -114,6 +116,38 @@ torch::Tensor nms(torch::Tensor boxes, torch::Tensor scores,
   return order.index_select(0, keep.slice(0, 0, n).to(torch::kLong));
 }
 
+torch::Tensor fpn_fuse(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                       c10::optional<torch::Tensor> up) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "x must be [B,C,H,W] cuda");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+  TORCH_CHECK(is_channels_last(x), "x must be channels_last (NHWC)");
+  const int B = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  auto w2 = w.reshape({w.size(0), -1}).contiguous().to(torch::kBFloat16);
+  const int Cout = w2.size(0);
+  TORCH_CHECK(w2.size(1) == Cin, "weight Cin mismatch");
+  TORCH_CHECK(Cin % 32 == 0, "Cin must be a multiple of 32");
+  auto bias_f = bias.to(torch::kFloat).contiguous();
+  const void* up_ptr = nullptr;
+  torch::Tensor up_t;
+  if (up.has_value()) {
+    up_t = up.value();
+    TORCH_CHECK(is_channels_last(up_t) &&
+                    up_t.scalar_type() == torch::kBFloat16,
+                "up must be channels_last bf16");
+    TORCH_CHECK(up_t.size(1) == Cout && up_t.size(2) == H / 2 &&
+                    up_t.size(3) == W / 2,
+                "up must be [B,Cout,H/2,W/2]");
+    up_ptr = up_t.data_ptr();
+  }
+  auto out = torch::empty(
+      {B, Cout, H, W},
+      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  fpn_fuse_launch(x.data_ptr(), w2.data_ptr(), bias_f.data_ptr<float>(),
+                  up_ptr, out.data_ptr(), B, H, W, Cin, Cout,
+                  cur_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -122,4 +156,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("roi_align_backward", &roi_align_backward,
         "RoIAlign backward (CDNA4, atomic scatter)");
   m.def("nms", &nms, "bitmask greedy NMS (CDNA4)");
+  m.def("fpn_fuse", &fpn_fuse,
+        "fused 1x1 conv + bias + nearest-up2 add (CDNA4 MFMA)",
+        py::arg("x"), py::arg("w"), py::arg("bias"),
+        py::arg("up") = py::none());
 }

@@ -5,9 +5,11 @@ Replicates the FPN section of reference MaskRCNN.build (mrcnn/model.py:
 3x3 output smoothing convs, P6 = stride-2 maxpool of P5. RPN consumes
 [P2..P6]; the heads consume [P2..P5].
 
-On GPU the lateral+top-down+add step runs as one fused HIP kernel
-(csrc/hip/fpn_fuse.hip, N7) when enabled; the module is numerically
-identical either way.
+MI355X-native: on GPU the lateral 1x1 conv + upsample + add runs as ONE
+hand-written CDNA4 MFMA kernel per level (csrc/hip/fpn_fuse.hip, N7),
+replacing three kernels and a full P-map HBM round trip; the 3x3
+smoothing convs stay on MIOpen. Numerics are identical either way
+(ops/fpn_fuse.py falls back to torch off-GPU).
 """
 
 from __future__ import annotations
@@ -15,9 +17,12 @@ from __future__ import annotations
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.fpn_fuse import fpn_fuse
+
 
 class FPN(nn.Module):
-    def __init__(self, in_channels=(256, 512, 1024, 2048), out_channels=256):
+    def __init__(self, in_channels=(256, 512, 1024, 2048), out_channels=256,
+                 use_fused=True):
         super().__init__()
         c2, c3, c4, c5 = in_channels
         self.fpn_c5p5 = nn.Conv2d(c5, out_channels, 1)
@@ -29,15 +34,22 @@ class FPN(nn.Module):
         self.fpn_p4 = nn.Conv2d(out_channels, out_channels, 3, padding=1)
         self.fpn_p5 = nn.Conv2d(out_channels, out_channels, 3, padding=1)
         self.out_channels = out_channels
+        self.use_fused = use_fused
 
     def forward(self, c2, c3, c4, c5):
-        p5 = self.fpn_c5p5(c5)
-        p4 = self.fpn_c4p4(c4) + F.interpolate(p5, scale_factor=2,
-                                               mode="nearest")
-        p3 = self.fpn_c3p3(c3) + F.interpolate(p4, scale_factor=2,
-                                               mode="nearest")
-        p2 = self.fpn_c2p2(c2) + F.interpolate(p3, scale_factor=2,
-                                               mode="nearest")
+        if self.use_fused:
+            p5 = fpn_fuse(c5, self.fpn_c5p5.weight, self.fpn_c5p5.bias)
+            p4 = fpn_fuse(c4, self.fpn_c4p4.weight, self.fpn_c4p4.bias, p5)
+            p3 = fpn_fuse(c3, self.fpn_c3p3.weight, self.fpn_c3p3.bias, p4)
+            p2 = fpn_fuse(c2, self.fpn_c2p2.weight, self.fpn_c2p2.bias, p3)
+        else:
+            p5 = self.fpn_c5p5(c5)
+            p4 = self.fpn_c4p4(c4) + F.interpolate(p5, scale_factor=2,
+                                                   mode="nearest")
+            p3 = self.fpn_c3p3(c3) + F.interpolate(p4, scale_factor=2,
+                                                   mode="nearest")
+            p2 = self.fpn_c2p2(c2) + F.interpolate(p3, scale_factor=2,
+                                                   mode="nearest")
         p2 = self.fpn_p2(p2)
         p3 = self.fpn_p3(p3)
         p4 = self.fpn_p4(p4)

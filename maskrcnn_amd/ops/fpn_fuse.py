@@ -1,0 +1,77 @@
+"""Fused FPN lateral + top-down op (HIP MFMA forward, torch backward).
+
+P_i = Conv1x1(C_i; W, b) + NearestUp2(P_{i+1})  — SURVEY.md N7.
+
+Forward runs the hand-written CDNA4 kernel (csrc/hip/fpn_fuse.hip) when
+on GPU with bf16 channels-last tensors; the CPU/odd-dtype fallback and the
+backward are composed from torch ops (the backward of a 1x1 conv is two
+GEMMs — hipBLASLt territory, no custom kernel needed).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from ._ext import load_ext
+
+
+def _fallback(x, w, b, up):
+    out = F.conv2d(x, w.view(w.size(0), -1, 1, 1), b.to(x.dtype))
+    if up is not None:
+        out = out + F.interpolate(up, scale_factor=2, mode="nearest")
+    return out
+
+
+class _FpnFuseFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, up):
+        ext = load_ext()
+        use_hip = (ext is not None and x.is_cuda and
+                   x.dtype == torch.bfloat16 and
+                   x.is_contiguous(memory_format=torch.channels_last) and
+                   x.size(1) % 32 == 0)
+        ctx.save_for_backward(x, w)
+        ctx.has_up = up is not None
+        ctx.up_dtype = up.dtype if up is not None else None
+        if use_hip:
+            out = ext.fpn_fuse(
+                x, w.to(torch.bfloat16), b.float(),
+                up.to(torch.bfloat16).contiguous(
+                    memory_format=torch.channels_last)
+                if up is not None else None)
+            return out
+        with torch.no_grad():
+            return _fallback(x, w, b, up)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        x, w = ctx.saved_tensors
+        go = grad_out
+        Cout = w.size(0)
+        w2 = w.view(Cout, -1)
+        # grad wrt input: 1x1 "transposed" conv = GEMM with W^T.
+        grad_x = None
+        if ctx.needs_input_grad[0]:
+            wt = w2.t().reshape(w2.size(1), Cout, 1, 1).to(go.dtype)
+            grad_x = F.conv2d(go, wt).to(x.dtype)
+        grad_w = None
+        if ctx.needs_input_grad[1]:
+            gof = go.permute(1, 0, 2, 3).reshape(Cout, -1).float()
+            xf = x.permute(1, 0, 2, 3).reshape(x.size(1), -1).float()
+            grad_w = (gof @ xf.t()).view_as(w)
+        grad_b = go.float().sum(dim=(0, 2, 3)) \
+            if ctx.needs_input_grad[2] else None
+        grad_up = None
+        if ctx.has_up and ctx.needs_input_grad[3]:
+            # Nearest-up2 backward = 2x2 sum pooling.
+            grad_up = (F.avg_pool2d(go.float(), 2) * 4.0) \
+                .to(ctx.up_dtype)
+        return grad_x, grad_w, grad_b, grad_up
+
+
+def fpn_fuse(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
+             up: torch.Tensor = None) -> torch.Tensor:
+    """x: [B,Cin,H,W]; weight: [Cout,Cin,1,1] or [Cout,Cin];
+    bias: [Cout]; up: [B,Cout,H/2,W/2] or None."""
+    return _FpnFuseFn.apply(x, weight, bias, up)
