@@ -23,6 +23,8 @@ void nms_launch(const float*, int, float, int, void*, int*, int*,
                 hipStream_t);
 void fpn_fuse_launch(const void*, const void*, const float*, const void*,
                      void*, int, int, int, int, int, hipStream_t);
+void conv3x3_fwd_launch(const void*, const void*, const float*, void*, int,
+                        int, int, int, int, bool, hipStream_t);
 }
 
 namespace {
@@ -148,6 +150,32 @@ torch::Tensor fpn_fuse(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   return out;
 }
 
+torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
+                          torch::Tensor bias, bool relu) {
+  // x: [B,Cin,H,W] channels_last bf16; w_packed: [9, Cout, Cin] bf16
+  // (pre-transposed tap-major by the python wrapper); bias: [Cout] f32.
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "x must be [B,C,H,W] cuda");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+  TORCH_CHECK(is_channels_last(x), "x must be channels_last (NHWC)");
+  TORCH_CHECK(w_packed.dim() == 3 && w_packed.size(0) == 9,
+              "w_packed must be [9, Cout, Cin]");
+  TORCH_CHECK(w_packed.scalar_type() == torch::kBFloat16 &&
+                  w_packed.is_contiguous(),
+              "w_packed must be contiguous bf16");
+  const int B = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  const int Cout = w_packed.size(1);
+  TORCH_CHECK(w_packed.size(2) == Cin, "weight Cin mismatch");
+  TORCH_CHECK(Cin % 32 == 0, "Cin must be a multiple of 32");
+  auto bias_f = bias.to(torch::kFloat).contiguous();
+  auto out = torch::empty(
+      {B, Cout, H, W},
+      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  conv3x3_fwd_launch(x.data_ptr(), w_packed.data_ptr(),
+                     bias_f.data_ptr<float>(), out.data_ptr(), B, H, W,
+                     Cin, Cout, relu, cur_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -160,4 +188,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused 1x1 conv + bias + nearest-up2 add (CDNA4 MFMA)",
         py::arg("x"), py::arg("w"), py::arg("bias"),
         py::arg("up") = py::none());
+  m.def("conv3x3_fwd", &conv3x3_fwd,
+        "3x3 conv s1p1 + bias + optional ReLU (CDNA4 MFMA implicit GEMM)",
+        py::arg("x"), py::arg("w_packed"), py::arg("bias"),
+        py::arg("relu") = false);
 }

@@ -1,0 +1,82 @@
+"""Fused 3x3 conv + frozen-BN + ReLU (HIP MFMA forward, torch backward).
+
+The SURVEY.md N1 op: frozen BatchNorm folds into the conv (w_eff = w *
+scale_per_outchannel; bias_eff = shift + scale * conv_bias), computed
+differentiably in torch so training still updates w/bias/BN-affine; the
+folded conv+ReLU then runs as one hand-written CDNA4 implicit-GEMM MFMA
+kernel (csrc/hip/conv3x3.hip). Backward uses torch.nn.grad conv grads
+(MIOpen-backed) — the forward fusion is the custom part.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from ._ext import load_ext
+
+
+def pack_weight(w: torch.Tensor) -> torch.Tensor:
+    """[Cout, Cin, 3, 3] -> contiguous [9, Cout, Cin] bf16 (tap-major)."""
+    Cout, Cin, _, _ = w.shape
+    return w.permute(2, 3, 0, 1).reshape(9, Cout, Cin) \
+        .contiguous().to(torch.bfloat16)
+
+
+class _Conv3x3Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, relu):
+        ctx.save_for_backward(x, w)
+        ctx.relu = relu
+        ext = load_ext()
+        use_hip = (ext is not None and x.is_cuda and
+                   x.dtype == torch.bfloat16 and
+                   x.is_contiguous(memory_format=torch.channels_last) and
+                   x.size(1) % 32 == 0)
+        if use_hip:
+            out = ext.conv3x3_fwd(x, pack_weight(w), b.float(), relu)
+        else:
+            with torch.no_grad():
+                out = F.conv2d(x, w.to(x.dtype), b.to(x.dtype), padding=1)
+                if relu:
+                    out = F.relu(out)
+        ctx.save_out = out if relu else None
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        x, w = ctx.saved_tensors
+        if ctx.relu:
+            grad_out = grad_out * (ctx.save_out > 0).to(grad_out.dtype)
+        grad_x = grad_w = grad_b = None
+        wd = w.to(grad_out.dtype)
+        if ctx.needs_input_grad[0]:
+            grad_x = torch.nn.grad.conv2d_input(
+                list(x.shape), wd, grad_out, padding=1).to(x.dtype)
+        if ctx.needs_input_grad[1]:
+            grad_w = torch.nn.grad.conv2d_weight(
+                x.to(grad_out.dtype), list(w.shape), grad_out,
+                padding=1).to(w.dtype)
+        if ctx.needs_input_grad[2]:
+            grad_b = grad_out.float().sum(dim=(0, 2, 3))
+        return grad_x, grad_w, grad_b, None
+
+
+def conv3x3_bias_relu(x, weight, bias, relu=True):
+    """3x3 stride-1 pad-1 conv + bias (+ReLU). weight [Cout,Cin,3,3]."""
+    return _Conv3x3Fn.apply(x, weight, bias, relu)
+
+
+def conv3x3_bn_relu(x, weight, conv_bias, bn_weight, bn_bias,
+                    running_mean, running_var, eps=1e-5, relu=True):
+    """Fold frozen BN into the conv, then run the fused kernel.
+
+    y = BN(conv(x, w) + cb) = conv(x, w * s) + (cb - mean) * s + bn_b,
+    with s = bn_w / sqrt(var + eps). The fold is differentiable, so
+    training updates every parameter while the fwd runs one kernel.
+    """
+    scale = bn_weight * (running_var + eps).rsqrt()
+    w_eff = weight * scale.view(-1, 1, 1, 1)
+    b_eff = bn_bias + (conv_bias - running_mean) * scale \
+        if conv_bias is not None else bn_bias - running_mean * scale
+    return _Conv3x3Fn.apply(x, w_eff, b_eff, relu)
