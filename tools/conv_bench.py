@@ -1,0 +1,92 @@
+#!/usr/bin/env python3
+"""A/B micro-benchmark: hand-written conv3x3/fpn_fuse MFMA kernels vs
+MIOpen (torch F.conv2d with benchmark Find) on R101-FPN shapes."""
+
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
+from maskrcnn_amd._rocm_env import ensure_env_or_reexec  # noqa: E402
+
+ensure_env_or_reexec()
+
+import torch  # noqa: E402
+import torch.nn.functional as F  # noqa: E402
+
+
+def timeit(fn, iters=30, warmup=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1e3  # ms
+
+
+def main():
+    from maskrcnn_amd._rocm_env import setup_rocm_perf
+    from maskrcnn_amd.ops._ext import require_ext
+    from maskrcnn_amd.ops.conv3x3 import pack_weight
+
+    setup_rocm_perf()
+    ext = require_ext()
+    dev = torch.device("cuda:0")
+    # (B, Cin, H, W, Cout) — the R101-FPN 3x3 shapes at batch 2 / 1024²
+    # plus the head shape at 400 rois.
+    shapes = [
+        (2, 64, 256, 256, 64),     # res2 3x3
+        (2, 128, 128, 128, 128),   # res3 3x3
+        (2, 256, 64, 64, 256),     # res4 3x3
+        (2, 512, 32, 32, 512),     # res5 3x3
+        (2, 256, 256, 256, 256),   # fpn_p2 3x3
+        (2, 256, 64, 64, 256),     # fpn_p4 3x3
+        (400, 256, 14, 14, 256),   # mask head conv
+        (2, 256, 256, 256, 512),   # rpn shared (P2)
+    ]
+    print(f"{'shape':>28} {'ours_ms':>9} {'miopen_ms':>10} "
+          f"{'ours_TF':>8} {'ratio':>6}")
+    for (B, Cin, H, W, Cout) in shapes:
+        x = torch.randn(B, Cin, H, W, device=dev).bfloat16() \
+            .to(memory_format=torch.channels_last)
+        w = (torch.randn(Cout, Cin, 3, 3, device=dev) * 0.05).bfloat16() \
+            .to(memory_format=torch.channels_last)
+        b = torch.randn(Cout, device=dev)
+        wp = pack_weight(w.float())
+        bf = b.float()
+
+        ours = timeit(lambda: ext.conv3x3_fwd(x, wp, bf, True))
+        ref = timeit(lambda: F.relu(F.conv2d(x, w, b.bfloat16(),
+                                             padding=1)))
+        flops = 2.0 * B * H * W * Cout * Cin * 9
+        tf = flops / (ours * 1e-3) / 1e12
+        print(f"{str((B, Cin, H, W, Cout)):>28} {ours:9.3f} {ref:10.3f} "
+              f"{tf:8.1f} {ref / ours:6.2f}x")
+
+    # fpn_fuse vs unfused (conv1x1 + interpolate + add).
+    print("\nfpn_fuse (1x1 + up2 + add):")
+    for (B, Cin, H, W, Cout) in [(2, 2048, 32, 32, 256),
+                                 (2, 1024, 64, 64, 256),
+                                 (2, 512, 128, 128, 256),
+                                 (2, 256, 256, 256, 256)]:
+        x = torch.randn(B, Cin, H, W, device=dev).bfloat16() \
+            .to(memory_format=torch.channels_last)
+        w = (torch.randn(Cout, Cin, 1, 1, device=dev) * 0.05)
+        wb = w.bfloat16().to(memory_format=torch.channels_last)
+        b = torch.randn(Cout, device=dev)
+        up = torch.randn(B, Cout, H // 2, W // 2, device=dev).bfloat16() \
+            .to(memory_format=torch.channels_last)
+        w2 = w.reshape(Cout, Cin).contiguous().bfloat16()
+
+        ours = timeit(lambda: ext.fpn_fuse(x, w2, b, up))
+        ref = timeit(lambda: F.conv2d(x, wb, b.bfloat16()) +
+                     F.interpolate(up, scale_factor=2, mode="nearest"))
+        print(f"{str((B, Cin, H, W, Cout)):>28} {ours:9.3f} {ref:10.3f} "
+              f"{'':8} {ref / ours:6.2f}x")
+
+
+if __name__ == "__main__":
+    main()
