@@ -13,6 +13,8 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
+from ..ops.conv_policy import use_hip_conv3x3
+
 
 class FrozenBatchNorm2d(nn.Module):
     """BatchNorm with statistics and affine params frozen.
@@ -52,6 +54,9 @@ class Bottleneck(nn.Module):
     """1x1 -> 3x3 -> 1x1 bottleneck with identity or projection shortcut.
 
     Mirrors reference conv_block / identity_block (model.py:~60-150).
+    With frozen BN the middle 3x3 runs as ONE hand-written CDNA4 MFMA
+    kernel (conv + folded BN + ReLU) on the shapes
+    where it beats MIOpen (ops/conv_policy.py).
     """
     expansion = 4
 
@@ -67,11 +72,20 @@ class Bottleneck(nn.Module):
         self.bn3 = make_norm(out_ch, train_bn)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
+        self._fused_mid = not train_bn  # frozen BN folds into the conv
 
     def forward(self, x):
         identity = x
         out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
+        if self._fused_mid and out.is_cuda and use_hip_conv3x3(
+                out.shape[0], out.shape[1], out.shape[2], out.shape[3]):
+            from ..ops.conv3x3 import conv3x3_bn_relu
+            bn = self.bn2
+            out = conv3x3_bn_relu(out, self.conv2.weight, self.conv2.bias,
+                                  bn.weight, bn.bias, bn.running_mean,
+                                  bn.running_var, eps=bn.eps, relu=True)
+        else:
+            out = self.relu(self.bn2(self.conv2(out)))
         out = self.bn3(self.conv3(out))
         if self.downsample is not None:
             identity = self.downsample(x)

@@ -13,6 +13,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.conv_policy import use_hip_conv3x3
+
 
 class RPN(nn.Module):
     def __init__(self, in_channels=256, anchors_per_location=3,
@@ -29,7 +31,15 @@ class RPN(nn.Module):
     def forward_level(self, x):
         """One pyramid level -> (logits [B,A_l,2], probs [B,A_l,2],
         deltas [B,A_l,4])."""
-        shared = F.relu(self.rpn_conv_shared(x))
+        conv = self.rpn_conv_shared
+        if (x.is_cuda and conv.stride[0] == 1 and
+                use_hip_conv3x3(x.shape[0], x.shape[1], x.shape[2],
+                                x.shape[3])):
+            from ..ops.conv3x3 import conv3x3_bias_relu
+            shared = conv3x3_bias_relu(x, conv.weight, conv.bias,
+                                       relu=True)
+        else:
+            shared = F.relu(conv(x))
         # [B, 2A, H, W] -> [B, H, W, 2A] -> [B, H*W*A, 2]
         cls = self.rpn_class_raw(shared)
         B, _, H, W = cls.shape
