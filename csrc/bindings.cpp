@@ -8,6 +8,7 @@
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
+#include <unordered_map>
 
 extern "C" {
 void roi_align_fwd_launch(const void*, const float*, const int*, void*, int,
@@ -25,6 +26,8 @@ void fpn_fuse_launch(const void*, const void*, const float*, const void*,
                      void*, int, int, int, int, int, hipStream_t);
 void conv3x3_fwd_launch(const void*, const void*, const float*, void*, int,
                         int, int, int, int, bool, hipStream_t);
+void conv3x3_v2_launch(const void*, const void*, const float*, const void*,
+                       void*, int, int, int, int, int, bool, hipStream_t);
 }
 
 namespace {
@@ -151,7 +154,8 @@ torch::Tensor fpn_fuse(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
 }
 
 torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
-                          torch::Tensor bias, bool relu) {
+                          torch::Tensor bias, bool relu,
+                          int64_t version = 0) {
   // x: [B,Cin,H,W] channels_last bf16; w_packed: [9, Cout, Cin] bf16
   // (pre-transposed tap-major by the python wrapper); bias: [Cout] f32.
   TORCH_CHECK(x.is_cuda() && x.dim() == 4, "x must be [B,C,H,W] cuda");
@@ -170,9 +174,31 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
   auto out = torch::empty(
       {B, Cout, H, W},
       x.options().memory_format(at::MemoryFormat::ChannelsLast));
-  conv3x3_fwd_launch(x.data_ptr(), w_packed.data_ptr(),
-                     bias_f.data_ptr<float>(), out.data_ptr(), B, H, W,
-                     Cin, Cout, relu, cur_stream());
+  const long M = (long)B * H * W;
+  bool use_v2;
+  if (version == 1) use_v2 = false;
+  else if (version == 2) use_v2 = true;
+  else  // auto: the 128x128 glds kernel needs enough M x N tiles
+    use_v2 = (M >= 8192 && Cout >= 128);
+  if (use_v2) {
+    // Persistent 64-B zero page per device (glds cannot predicate
+    // out-of-range lanes; they DMA from here instead).
+    static std::unordered_map<int, torch::Tensor> zero_pages;
+    const int dev = x.get_device();
+    auto it = zero_pages.find(dev);
+    if (it == zero_pages.end()) {
+      it = zero_pages.emplace(
+          dev, torch::zeros({32}, x.options())).first;
+    }
+    conv3x3_v2_launch(x.data_ptr(), w_packed.data_ptr(),
+                      bias_f.data_ptr<float>(), it->second.data_ptr(),
+                      out.data_ptr(), B, H, W, Cin, Cout, relu,
+                      cur_stream());
+  } else {
+    conv3x3_fwd_launch(x.data_ptr(), w_packed.data_ptr(),
+                       bias_f.data_ptr<float>(), out.data_ptr(), B, H, W,
+                       Cin, Cout, relu, cur_stream());
+  }
   return out;
 }
 
@@ -191,5 +217,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_fwd", &conv3x3_fwd,
         "3x3 conv s1p1 + bias + optional ReLU (CDNA4 MFMA implicit GEMM)",
         py::arg("x"), py::arg("w_packed"), py::arg("bias"),
-        py::arg("relu") = false);
+        py::arg("relu") = false, py::arg("version") = 0);
 }

@@ -86,6 +86,29 @@ def test_gpu_kernel_matches_reference(cin, cout, h, w):
 
 
 @pytest.mark.gpu
+@pytest.mark.parametrize("cin,cout,h,w,batch", [
+    (256, 256, 64, 64, 2),    # exact-tile case
+    (256, 256, 50, 30, 3),    # ragged M
+    (64, 192, 64, 64, 2),     # Cout not multiple of 128
+    (128, 512, 32, 32, 2),
+])
+def test_gpu_v2_kernel_matches_reference(cin, cout, h, w, batch):
+    """The 128x128-tile glds variant, forced via version=2."""
+    from maskrcnn_amd.ops._ext import require_ext
+    from maskrcnn_amd.ops.conv3x3 import pack_weight
+    ext = require_ext()
+    dev = torch.device("cuda:0")
+    torch.manual_seed(0)
+    x = torch.randn(batch, cin, h, w, device=dev).bfloat16() \
+        .to(memory_format=torch.channels_last)
+    wt = (torch.randn(cout, cin, 3, 3, device=dev) * 0.05)
+    b = torch.randn(cout, device=dev)
+    out = ext.conv3x3_fwd(x, pack_weight(wt), b.float(), True, 2)
+    ref = F.relu(F.conv2d(x.float(), wt, b, padding=1))
+    torch.testing.assert_close(out.float(), ref, atol=0.3, rtol=0.05)
+
+
+@pytest.mark.gpu
 def test_gpu_border_zero_padding():
     """Border pixels must see zero-padding, not garbage."""
     dev = torch.device("cuda:0")

@@ -47,8 +47,7 @@ def main():
         (400, 256, 14, 14, 256),   # mask head conv
         (2, 256, 256, 256, 512),   # rpn shared (P2)
     ]
-    print(f"{'shape':>28} {'ours_ms':>9} {'miopen_ms':>10} "
-          f"{'ours_TF':>8} {'ratio':>6}")
+    print(f"{'shape':>28}  v1 / v2 / miopen")
     for (B, Cin, H, W, Cout) in shapes:
         x = torch.randn(B, Cin, H, W, device=dev).bfloat16() \
             .to(memory_format=torch.channels_last)
@@ -58,13 +57,16 @@ def main():
         wp = pack_weight(w.float())
         bf = b.float()
 
-        ours = timeit(lambda: ext.conv3x3_fwd(x, wp, bf, True))
+        v1 = timeit(lambda: ext.conv3x3_fwd(x, wp, bf, True, 1))
+        v2 = timeit(lambda: ext.conv3x3_fwd(x, wp, bf, True, 2))
         ref = timeit(lambda: F.relu(F.conv2d(x, w, b.bfloat16(),
                                              padding=1)))
         flops = 2.0 * B * H * W * Cout * Cin * 9
-        tf = flops / (ours * 1e-3) / 1e12
-        print(f"{str((B, Cin, H, W, Cout)):>28} {ours:9.3f} {ref:10.3f} "
-              f"{tf:8.1f} {ref / ours:6.2f}x")
+        tf1 = flops / (v1 * 1e-3) / 1e12
+        tf2 = flops / (v2 * 1e-3) / 1e12
+        print(f"{str((B, Cin, H, W, Cout)):>28} v1={v1:7.3f}ms({tf1:5.0f}TF)"
+              f" v2={v2:7.3f}ms({tf2:5.0f}TF) miopen={ref:7.3f}ms "
+              f"best_ratio={ref / min(v1, v2):5.2f}x")
 
     # fpn_fuse vs unfused (conv1x1 + interpolate + add).
     print("\nfpn_fuse (1x1 + up2 + add):")
