@@ -179,7 +179,7 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
   if (version == 1) use_v2 = false;
   else if (version == 2) use_v2 = true;
   else  // auto: the 128x128 glds kernel needs enough M x N tiles
-    use_v2 = (M >= 8192 && Cout >= 128);
+    use_v2 = (M >= 32768 && Cout >= 128);
   if (use_v2) {
     // Persistent 64-B zero page per device (glds cannot predicate
     // out-of-range lanes; they DMA from here instead).

@@ -29,7 +29,8 @@ class PyramidROIAlign(nn.Module):
         """
         rois: [B, N, 4] normalized (zero rows = padding)
         feature_maps: [P2, P3, P4, P5], each [B, C, H_l, W_l]
-        Returns [B, N, C, pool, pool].
+        Returns [B*N, C, pool, pool] (flat roi-major; channels_last on GPU
+        so the head convs hit the HIP conv path).
         """
         B, N, _ = rois.shape
         C = feature_maps[0].shape[1]
@@ -48,8 +49,13 @@ class PyramidROIAlign(nn.Module):
         # Padding rows (zero area) -> level 2; output stays zero via mask.
         valid = (h > 0) & (w > 0)
 
-        out = feature_maps[0].new_zeros(
-            (B * N, C, self.pool_size, self.pool_size))
+        cl = feature_maps[0].is_cuda and feature_maps[0].is_contiguous(
+            memory_format=torch.channels_last)
+        out = torch.empty(
+            (B * N, C, self.pool_size, self.pool_size),
+            dtype=feature_maps[0].dtype, device=feature_maps[0].device,
+            memory_format=(torch.channels_last if cl
+                           else torch.contiguous_format)).zero_()
         for i, level in enumerate(range(2, 6)):
             sel = ((levels == level) & valid).nonzero(as_tuple=True)[0]
             if sel.numel() == 0:
@@ -59,4 +65,4 @@ class PyramidROIAlign(nn.Module):
                                sampling_ratio=self.sampling_ratio,
                                mode=self.mode)
             out[sel] = pooled.to(out.dtype)
-        return out.reshape(B, N, C, self.pool_size, self.pool_size)
+        return out
