@@ -63,10 +63,13 @@ def build_signature(config) -> SignatureDef:
 
 
 def export_saved_model(model, config, base_dir: str, version: int = 1,
-                       extra_meta: dict = None) -> str:
+                       extra_meta: dict = None,
+                       weight_dtype: str = "fp32") -> str:
     """Export a (torch) Mask R-CNN model as a SavedModel-layout directory.
 
     model: MaskRCNNModel (or the MaskRCNN facade; .model is used).
+    weight_dtype: "fp32" or "fp8" (e4m3fn per-out-channel quantized
+    conv/linear weights — BASELINE config #5).
     Returns the version directory path.
     """
     import torch
@@ -83,11 +86,17 @@ def export_saved_model(model, config, base_dir: str, version: int = 1,
 
     state = {k: v.detach().cpu().contiguous()
              for k, v in net.state_dict().items()}
+    if weight_dtype == "fp8":
+        from ..utils.fp8 import quantize_state_dict
+        state = quantize_state_dict(state)
+    elif weight_dtype != "fp32":
+        raise ValueError(f"unsupported weight_dtype {weight_dtype!r}")
     shard = os.path.join(var_dir, "variables.data-00000-of-00001")
     save_file(state, shard)
 
     index = {
         "format": "maskrcnn_amd.safetensors.v1",
+        "weight_dtype": weight_dtype,
         "shards": ["variables.data-00000-of-00001"],
         "num_tensors": len(state),
         "config": {k: v for k, v in config.to_dict().items()
@@ -119,6 +128,9 @@ def load_saved_model(export_dir: str):
     state = {}
     for shard in index["shards"]:
         state.update(load_file(os.path.join(var_dir, shard)))
+    if index.get("weight_dtype") == "fp8":
+        from ..utils.fp8 import dequantize_state_dict
+        state = dequantize_state_dict(state)
     return state, index.get("config", {}), sigs
 
 
