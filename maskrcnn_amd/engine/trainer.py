@@ -54,6 +54,15 @@ class Trainer:
         self.history = []
         from .._rocm_env import setup_rocm_perf
         setup_rocm_perf()
+        # DDP over RCCL/xGMI (or gloo on CPU) when launched under
+        # torchrun — replaces the reference's ParallelModel.
+        import torch.distributed as dist
+        self.distributed = dist.is_available() and dist.is_initialized()
+        self.rank = dist.get_rank() if self.distributed else 0
+        self.ddp = None
+        if self.distributed:
+            from ..distributed import DistributedTrainer
+            self.ddp = DistributedTrainer(self.model, self.config)
 
     def _autocast(self):
         use_bf16 = (self.config.COMPUTE_DTYPE == "bf16" and
@@ -87,11 +96,36 @@ class Trainer:
                 outputs)
         optimizer.zero_grad(set_to_none=True)
         losses["total_loss"].backward()
+        if self.ddp is not None:
+            self.ddp.finalize_backward()
         torch.nn.utils.clip_grad_norm_(
             (p for p in self.model.parameters() if p.requires_grad),
             self.config.GRADIENT_CLIP_NORM)
         optimizer.step()
         return {k: float(v.detach()) for k, v in losses.items()}
+
+    def _log_scalars(self, epoch, means, seconds):
+        """Per-epoch scalars: CSV always; TensorBoard when importable
+        (reference used a TensorBoard callback, model.py:~2340)."""
+        if self.rank != 0:
+            return
+        import csv
+        path = os.path.join(self.wrapper.log_dir, "metrics.csv")
+        new = not os.path.exists(path)
+        with open(path, "a", newline="") as f:
+            wtr = csv.writer(f)
+            if new:
+                wtr.writerow(["epoch", "seconds"] + sorted(means))
+            wtr.writerow([epoch, round(seconds, 2)] +
+                         [round(means[k], 6) for k in sorted(means)])
+        try:
+            from torch.utils.tensorboard import SummaryWriter
+            if not hasattr(self, "_tb"):
+                self._tb = SummaryWriter(self.wrapper.log_dir)
+            for k, v in means.items():
+                self._tb.add_scalar(k, v, epoch)
+        except ImportError:
+            pass
 
     def fit(self, learning_rate, epochs, layers="all", steps_per_epoch=None):
         cfg = self.config
@@ -99,12 +133,17 @@ class Trainer:
         optimizer = build_optimizer(self.model, cfg, learning_rate)
         loader = make_data_loader(self.train_dataset, cfg,
                                   augmentation=self.augmentation,
-                                  num_workers=self.num_workers)
+                                  num_workers=self.num_workers,
+                                  distributed=self.distributed)
         steps_per_epoch = steps_per_epoch or cfg.STEPS_PER_EPOCH
-        os.makedirs(self.wrapper.log_dir, exist_ok=True)
+        if self.rank == 0:
+            os.makedirs(self.wrapper.log_dir, exist_ok=True)
 
         while self.wrapper.epoch < epochs:
             epoch = self.wrapper.epoch
+            if self.distributed and loader.sampler is not None and \
+                    hasattr(loader.sampler, "set_epoch"):
+                loader.sampler.set_epoch(epoch)
             t0 = time.time()
             step = 0
             epoch_losses = {}
@@ -124,7 +163,9 @@ class Trainer:
             dt = time.time() - t0
             log.info("epoch %d: %.1fs %s", epoch, dt,
                      {k: round(v, 4) for k, v in means.items()})
-            ckpt = self.wrapper.checkpoint_path.format(epoch=epoch)
-            self.wrapper.save_weights(ckpt)
+            self._log_scalars(epoch, means, dt)
+            if self.rank == 0:
+                ckpt = self.wrapper.checkpoint_path.format(epoch=epoch)
+                self.wrapper.save_weights(ckpt)
             self.wrapper.epoch += 1
         return self.history
