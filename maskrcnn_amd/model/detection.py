@@ -39,56 +39,65 @@ class DetectionLayer(nn.Module):
         deltas: [B, N, C, 4]
         image_meta: [B, META]
         Returns detections [B, DETECTION_MAX_INSTANCES, 6].
+
+        Batched over images AND classes in ONE NMS launch (the reference
+        serializes an 80-class map_fn per image; a python per-image loop
+        would likewise serialize at serving batch sizes).
         """
         cfg = self.config
         B, N, C = probs.shape
+        device = rois.device
         meta = parse_image_meta_t(image_meta)
         image_shape = meta["image_shape"][0]
-        # Window in normalized coords (reference :~700).
+        # Windows in normalized coords (reference :~700), per image.
         windows = norm_boxes_t(meta["window"].float(),
                                (float(image_shape[0]),
-                                float(image_shape[1])))
+                                float(image_shape[1])))  # [B, 4]
 
-        out = rois.new_zeros((B, cfg.DETECTION_MAX_INSTANCES, 6))
-        for b in range(B):
-            det = self._one_image(rois[b].float(), probs[b].float(),
-                                  deltas[b].float(), windows[b])
-            out[b, :det.shape[0]] = det
-        return out
-
-    def _one_image(self, rois, probs, deltas, window):
-        cfg = self.config
-        device = rois.device
-        N = rois.shape[0]
-        class_ids = probs.argmax(dim=1)                       # [N]
-        idx = torch.arange(N, device=device)
-        class_scores = probs[idx, class_ids]
-        deltas_specific = deltas[idx, class_ids]              # [N, 4]
+        probs = probs.float()
+        class_ids = probs.argmax(dim=2)                        # [B, N]
+        class_scores = probs.gather(
+            2, class_ids.unsqueeze(-1)).squeeze(-1)            # [B, N]
+        deltas_specific = deltas.float().gather(
+            2, class_ids[..., None, None].expand(-1, -1, 1, 4)) \
+            .squeeze(2)                                        # [B, N, 4]
 
         refined = apply_box_deltas_t(
-            rois, deltas_specific * self.bbox_std_dev)
-        refined = clip_boxes_t(refined, window)
+            rois.float(), deltas_specific * self.bbox_std_dev)
+        # Clip each image to its own window ([B,1] bounds broadcast).
+        wy1, wx1, wy2, wx2 = windows.unsqueeze(2).unbind(dim=1)
+        y1 = torch.clamp(refined[..., 0], wy1, wy2)
+        x1 = torch.clamp(refined[..., 1], wx1, wx2)
+        y2 = torch.clamp(refined[..., 2], wy1, wy2)
+        x2 = torch.clamp(refined[..., 3], wx1, wx2)
+        refined = torch.stack([y1, x1, y2, x2], dim=-1)
 
         keep = class_ids > 0
         if cfg.DETECTION_MIN_CONFIDENCE:
             keep &= class_scores >= cfg.DETECTION_MIN_CONFIDENCE
-        keep_ix = keep.nonzero(as_tuple=True)[0]
-        if keep_ix.numel() == 0:
-            return rois.new_zeros((0, 6))
+        flat_ix = keep.reshape(-1).nonzero(as_tuple=True)[0]
+        D = cfg.DETECTION_MAX_INSTANCES
+        out = rois.new_zeros((B, D, 6), dtype=torch.float32)
+        if flat_ix.numel() == 0:
+            return out
 
-        pre_boxes = refined[keep_ix]
-        pre_scores = class_scores[keep_ix]
-        pre_class = class_ids[keep_ix]
-
-        nms_keep = batched_nms(pre_boxes, pre_scores, pre_class,
+        img = flat_ix // N
+        boxes_f = refined.reshape(-1, 4)[flat_ix]
+        scores_f = class_scores.reshape(-1)[flat_ix]
+        cls_f = class_ids.reshape(-1)[flat_ix]
+        # NMS group = (image, class).
+        group = img * (C + 1) + cls_f
+        nms_keep = batched_nms(boxes_f, scores_f, group,
                                cfg.DETECTION_NMS_THRESHOLD)
-        # Reference also caps per-class keep at DETECTION_MAX_INSTANCES
-        # before the global top-k; the global cap below subsumes it.
-        nms_keep = nms_keep[:cfg.DETECTION_MAX_INSTANCES]
-        boxes = pre_boxes[nms_keep]
-        scores = pre_scores[nms_keep]
-        cls = pre_class[nms_keep]
-        order = scores.argsort(descending=True)
-        boxes, scores, cls = boxes[order], scores[order], cls[order]
-        return torch.cat([boxes, cls[:, None].float(),
-                          scores[:, None]], dim=1)
+        if nms_keep.numel() == 0:
+            return out
+        kimg = img[nms_keep]                   # global score-desc order
+        onehot = (kimg.unsqueeze(1) ==
+                  torch.arange(B, device=device).unsqueeze(0))
+        rank = onehot.long().cumsum(dim=0)[
+            torch.arange(kimg.numel(), device=device), kimg] - 1
+        sel = rank < D
+        rows = torch.cat([boxes_f[nms_keep], cls_f[nms_keep, None].float(),
+                          scores_f[nms_keep, None]], dim=1)
+        out[kimg[sel], rank[sel]] = rows[sel]
+        return out

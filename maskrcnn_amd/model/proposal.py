@@ -14,7 +14,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ..ops import nms
+from ..ops import batched_nms, nms
 from ..utils.boxes import apply_box_deltas_t, clip_boxes_t
 
 
@@ -53,10 +53,25 @@ class ProposalLayer(nn.Module):
         window = torch.tensor([0., 0., 1., 1.], device=boxes.device)
         boxes = clip_boxes_t(boxes, window)
 
+        # One NMS launch for ALL images (image-index offset trick) —
+        # B sequential NMS calls each cost a host sync; at serving
+        # batch 32 that serializes badly.
+        P = boxes.shape[1]
+        flat_boxes = boxes.reshape(-1, 4)
+        flat_scores = top_scores.reshape(-1)
+        img_ix = torch.arange(B, device=boxes.device) \
+            .repeat_interleave(P)
+        keep = batched_nms(flat_boxes, flat_scores, img_ix,
+                           cfg.RPN_NMS_THRESHOLD)
         rois = boxes.new_zeros((B, proposal_count, 4))
-        for b in range(B):
-            keep = nms(boxes[b], top_scores[b], cfg.RPN_NMS_THRESHOLD,
-                       max_out=proposal_count)
-            n = keep.shape[0]
-            rois[b, :n] = boxes[b, keep]
+        if keep.numel():
+            img = img_ix[keep]                         # [K], score-desc
+            # Per-image rank of each kept box (global order preserves
+            # per-image score order).
+            onehot = (img.unsqueeze(1) ==
+                      torch.arange(B, device=img.device).unsqueeze(0))
+            rank = onehot.long().cumsum(dim=0)[
+                torch.arange(img.numel(), device=img.device), img] - 1
+            sel = rank < proposal_count
+            rois[img[sel], rank[sel]] = flat_boxes[keep[sel]]
         return rois
