@@ -76,8 +76,9 @@ class FPNMaskHead(nn.Module):
         one HIP MFMA kernel on GPU (14x14 maps: measured 1.19x over
         MIOpen before counting the fused BN/ReLU, profiles/r01_conv_ab.md)."""
         from ..ops.conv_policy import use_hip_conv3x3
-        if not (self._fused and x.is_cuda and use_hip_conv3x3(
-                x.shape[0], x.shape[1], x.shape[2], x.shape[3])):
+        if not (self._fused and x.is_cuda and not self.training and
+                use_hip_conv3x3(x.shape[0], x.shape[1], x.shape[2],
+                                x.shape[3])):
             return self.convs(x)
         from ..ops.conv3x3 import conv3x3_bn_relu
         for i in range(4):

@@ -77,8 +77,12 @@ class Bottleneck(nn.Module):
     def forward(self, x):
         identity = x
         out = self.relu(self.bn1(self.conv1(x)))
-        if self._fused_mid and out.is_cuda and use_hip_conv3x3(
-                out.shape[0], out.shape[1], out.shape[2], out.shape[3]):
+        # HIP conv in eval/serving; during training the MIOpen
+        # autograd path measured faster end-to-end (fwd win < bwd-algo
+        # delta) — revisit with the conv3x3 bwd kernel.
+        if self._fused_mid and out.is_cuda and not self.training and \
+                use_hip_conv3x3(out.shape[0], out.shape[1], out.shape[2],
+                                out.shape[3]):
             from ..ops.conv3x3 import conv3x3_bn_relu
             bn = self.bn2
             out = conv3x3_bn_relu(out, self.conv2.weight, self.conv2.bias,

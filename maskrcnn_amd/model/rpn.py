@@ -32,7 +32,7 @@ class RPN(nn.Module):
         """One pyramid level -> (logits [B,A_l,2], probs [B,A_l,2],
         deltas [B,A_l,4])."""
         conv = self.rpn_conv_shared
-        if (x.is_cuda and conv.stride[0] == 1 and
+        if (x.is_cuda and not self.training and conv.stride[0] == 1 and
                 use_hip_conv3x3(x.shape[0], x.shape[1], x.shape[2],
                                 x.shape[3])):
             from ..ops.conv3x3 import conv3x3_bias_relu

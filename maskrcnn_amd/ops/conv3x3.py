@@ -50,13 +50,19 @@ class _Conv3x3Fn(torch.autograd.Function):
             grad_out = grad_out * (ctx.save_out > 0).to(grad_out.dtype)
         grad_x = grad_w = grad_b = None
         wd = w.to(grad_out.dtype)
+        # aten.convolution_backward = the fused MIOpen backward autograd
+        # itself uses (torch.nn.grad.conv2d_weight takes a far slower
+        # unfold path).
+        gx, gw, _ = torch.ops.aten.convolution_backward(
+            grad_out.contiguous(
+                memory_format=torch.channels_last if grad_out.is_cuda
+                else torch.contiguous_format),
+            x, wd, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+            [ctx.needs_input_grad[0], ctx.needs_input_grad[1], False])
         if ctx.needs_input_grad[0]:
-            grad_x = torch.nn.grad.conv2d_input(
-                list(x.shape), wd, grad_out, padding=1).to(x.dtype)
+            grad_x = gx.to(x.dtype)
         if ctx.needs_input_grad[1]:
-            grad_w = torch.nn.grad.conv2d_weight(
-                x.to(grad_out.dtype), list(w.shape), grad_out,
-                padding=1).to(w.dtype)
+            grad_w = gw.to(w.dtype)
         if ctx.needs_input_grad[2]:
             grad_b = grad_out.float().sum(dim=(0, 2, 3))
         return grad_x, grad_w, grad_b, None

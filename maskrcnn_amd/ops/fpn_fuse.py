@@ -49,17 +49,17 @@ class _FpnFuseFn(torch.autograd.Function):
         x, w = ctx.saved_tensors
         go = grad_out
         Cout = w.size(0)
-        w2 = w.view(Cout, -1)
-        # grad wrt input: 1x1 "transposed" conv = GEMM with W^T.
-        grad_x = None
-        if ctx.needs_input_grad[0]:
-            wt = w2.t().reshape(w2.size(1), Cout, 1, 1).to(go.dtype)
-            grad_x = F.conv2d(go, wt).to(x.dtype)
-        grad_w = None
-        if ctx.needs_input_grad[1]:
-            gof = go.permute(1, 0, 2, 3).reshape(Cout, -1).float()
-            xf = x.permute(1, 0, 2, 3).reshape(x.size(1), -1).float()
-            grad_w = (gof @ xf.t()).view_as(w)
+        w4 = w.view(Cout, -1, 1, 1).to(go.dtype)
+        # aten.convolution_backward = the fused MIOpen 1x1-conv backward.
+        gx, gw, _ = torch.ops.aten.convolution_backward(
+            go.contiguous(
+                memory_format=torch.channels_last if go.is_cuda
+                else torch.contiguous_format),
+            x, w4, None, [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+            [ctx.needs_input_grad[0], ctx.needs_input_grad[1], False])
+        grad_x = gx.to(x.dtype) if ctx.needs_input_grad[0] else None
+        grad_w = gw.view_as(w).to(w.dtype) \
+            if ctx.needs_input_grad[1] else None
         grad_b = go.float().sum(dim=(0, 2, 3)) \
             if ctx.needs_input_grad[2] else None
         grad_up = None
