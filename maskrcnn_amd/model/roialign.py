@@ -49,8 +49,12 @@ class PyramidROIAlign(nn.Module):
         # Padding rows (zero area) -> level 2; output stays zero via mask.
         valid = (h > 0) & (w > 0)
 
-        cl = feature_maps[0].is_cuda and feature_maps[0].is_contiguous(
-            memory_format=torch.channels_last)
+        # channels_last output feeds the HIP conv path in the heads —
+        # eval/serving only (training keeps the round-1-validated NCHW
+        # layout; round-2 A/B pending).
+        cl = (feature_maps[0].is_cuda and not self.training and
+              feature_maps[0].is_contiguous(
+                  memory_format=torch.channels_last))
         out = torch.empty(
             (B * N, C, self.pool_size, self.pool_size),
             dtype=feature_maps[0].dtype, device=feature_maps[0].device,
