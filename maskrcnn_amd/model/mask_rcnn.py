@@ -373,6 +373,53 @@ class MaskRCNN:
         self.model.eval()
         return self.model(images_t, metas_t, anchors, mode="inference")
 
+    @torch.no_grad()
+    def run_graph(self, images, names, mode="inference"):
+        """Evaluate named intermediates (reference run_graph, model.py
+        :~2530 — the inspect_model.ipynb probe mechanism).
+
+        names: list of either forward-output keys (e.g. "rois",
+        "mrcnn_detection") or dotted module names (e.g. "backbone.layer2",
+        "fpn", "rpn") whose outputs are captured via hooks.
+        Returns {name: numpy array / list of arrays}.
+        """
+        captured = {}
+        hooks = []
+        module_map = dict(self.model.named_modules())
+        for name in names:
+            if name in module_map:
+                def make_hook(n):
+                    def hook(_m, _inp, out):
+                        captured[n] = out
+                    return hook
+                hooks.append(
+                    module_map[name].register_forward_hook(
+                        make_hook(name)))
+        try:
+            molded_images, image_metas, _ = self.mold_inputs(images)
+            anchors = self.anchor_cache.get(molded_images[0].shape,
+                                            device=self.device)
+            outputs = self.predict_molded(molded_images, image_metas,
+                                          anchors)
+        finally:
+            for h in hooks:
+                h.remove()
+        for name in names:
+            if name in outputs:
+                captured[name] = outputs[name]
+
+        def to_np(v):
+            if isinstance(v, torch.Tensor):
+                return v.detach().float().cpu().numpy()
+            if isinstance(v, (list, tuple)):
+                return [to_np(x) for x in v]
+            return v
+        missing = [n for n in names if n not in captured]
+        if missing:
+            raise KeyError(f"run_graph: unknown names {missing}; "
+                           f"outputs={list(outputs)}")
+        return {n: to_np(captured[n]) for n in names}
+
     # ------------------------------------------------------------ training
     def compute_losses(self, inputs, outputs):
         return compute_losses(self.config, inputs, outputs)
