@@ -3,9 +3,10 @@
 dynamic-batched inference on 1 MI355X.
 
 Starts an in-process ModelServer (gRPC only) on a random-init R101-FPN
-export, fires concurrent Predict requests from client threads (pre-molded
-synthetic 1024x1024 images, anchors server-side), and reports images/s +
-p50/p99 latency as one JSON line.
+export, then drives it with MULTI-PROCESS gRPC clients (each subprocess
+has its own channel and GIL; single-process clients were the round-1
+bottleneck) firing pre-molded synthetic 1024x1024 Predict requests.
+Reports images/s + p50/p99 latency as one JSON line.
 """
 
 from __future__ import annotations
@@ -13,9 +14,9 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import subprocess
 import sys
 import tempfile
-import threading
 import time
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(
@@ -26,33 +27,34 @@ ensure_env_or_reexec()
 
 import numpy as np  # noqa: E402
 
+WORKER = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                      "serving_client_worker.py")
+
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--requests", type=int, default=64)
-    ap.add_argument("--client-timeout", type=float, default=300.0)
-    ap.add_argument("--concurrency", type=int, default=16)
+    ap.add_argument("--requests", type=int, default=64,
+                    help="total timed requests across all clients")
+    ap.add_argument("--client-procs", type=int, default=4)
+    ap.add_argument("--client-threads", type=int, default=8)
     ap.add_argument("--warmup", type=int, default=8)
     ap.add_argument("--image-size", type=int, default=1024)
     ap.add_argument("--max-batch", type=int, default=32)
     ap.add_argument("--batch-timeout-micros", type=int, default=5000)
     ap.add_argument("--backbone", default="resnet101")
-    ap.add_argument("--send-anchors", action="store_true",
-                    help="include the [A,4] anchor tensor in each request "
-                         "(reference client behavior)")
+    ap.add_argument("--weight-dtype", default="fp32",
+                    choices=["fp32", "fp8"])
     args = ap.parse_args()
 
     import logging
-    logging.basicConfig(
-        level=logging.INFO, stream=sys.stderr,
-        format="%(asctime)s %(name)s %(message)s")
+    logging.basicConfig(level=logging.INFO, stream=sys.stderr,
+                        format="%(asctime)s %(name)s %(message)s")
 
     import torch
     from maskrcnn_amd.config import Config
     from maskrcnn_amd.model.mask_rcnn import MaskRCNNModel
     from maskrcnn_amd.serving.savedmodel import export_saved_model
     from maskrcnn_amd.serving.server import ModelServer
-    from maskrcnn_amd.serving.client import GrpcServingClient
 
     class ServeConfig(Config):
         NAME = "serve_bench"
@@ -66,7 +68,8 @@ def main():
     torch.manual_seed(0)
     model = MaskRCNNModel(cfg)
     base = tempfile.mkdtemp(prefix="serve_bench_")
-    export_saved_model(model, cfg, base, version=1)
+    export_saved_model(model, cfg, base, version=1,
+                       weight_dtype=args.weight_dtype)
     del model
 
     server = ModelServer("mask_rcnn", base, grpc_port=0, rest_port=None,
@@ -74,52 +77,43 @@ def main():
                          batch_timeout_micros=args.batch_timeout_micros,
                          config=cfg)
     server.start(warmup=True)
+    target = f"127.0.0.1:{server.grpc_port}"
 
-    client = GrpcServingClient(f"localhost:{server.grpc_port}", cfg,
-                               send_anchors=args.send_anchors,
-                               timeout=args.client_timeout)
-    # Pre-mold one synthetic image; clients reuse the molded arrays
-    # (the serving benchmark measures the server, not PIL).
-    img = (np.random.rand(800, 1000, 3) * 255).astype(np.uint8)
-    molded, metas, windows = client.mold_inputs([img])
-    anchors = None
-    if args.send_anchors:
-        anchors = client.anchor_cache.get(molded[0].shape).numpy() \
-            .astype(np.float32)[None]
+    def run_clients(total_requests):
+        per = max(1, total_requests // args.client_procs)
+        procs = [subprocess.Popen(
+            [sys.executable, WORKER, "--target", target,
+             "--requests", str(per),
+             "--threads", str(args.client_threads),
+             "--image-size", str(args.image_size)],
+            stdout=subprocess.PIPE, text=True)
+            for _ in range(args.client_procs)]
+        lats = []
+        elapsed = 0.0
+        n_err = 0
+        for p in procs:
+            out, _ = p.communicate(timeout=1800)
+            assert p.returncode == 0, "client worker failed"
+            d = json.loads(out.strip().splitlines()[-1])
+            lats.extend(d["lats"])
+            elapsed = max(elapsed, d["elapsed"])
+            n_err += d.get("n_errors", 0)
+            if d.get("errors"):
+                print("client errors:", d["errors"], file=sys.stderr)
+        if n_err:
+            print(f"WARNING: {n_err} client requests errored",
+                  file=sys.stderr)
+        return lats, elapsed
 
-    latencies = []
-    lat_lock = threading.Lock()
-    counter = {"n": -args.warmup}
+    run_clients(args.warmup)  # warmup (incl. channel setup)
+    t0 = time.perf_counter()
+    lats, elapsed = run_clients(args.requests)
+    wall = time.perf_counter() - t0
 
-    def worker():
-        while True:
-            with lat_lock:
-                if counter["n"] >= args.requests:
-                    return
-                counter["n"] += 1
-                warm = counter["n"] <= 0
-            t0 = time.perf_counter()
-            client.predict_raw(molded, metas, anchors)
-            dt = time.perf_counter() - t0
-            if not warm:
-                with lat_lock:
-                    latencies.append(dt)
-
-    # Warmup request serially first (MIOpen find etc.).
-    client.predict_raw(molded, metas, anchors)
-
-    threads = [threading.Thread(target=worker)
-               for _ in range(args.concurrency)]
-    t_start = time.perf_counter()
-    for t in threads:
-        t.start()
-    for t in threads:
-        t.join()
-    elapsed = time.perf_counter() - t_start
-
-    lats = np.array(sorted(latencies))
-    n = len(lats)
+    lats_a = np.array(sorted(lats))
+    n = len(lats_a)
     qps = n / elapsed
+    st = server.state.batcher.stats
     out = {
         "metric": "serving_images_per_sec",
         "value": round(qps, 3),
@@ -131,18 +125,18 @@ def main():
         "higher_is_better": True,
         "scaling": "weak",
         "vs_baseline": None,
-        "dtype": "bf16",
+        "dtype": "bf16" if args.weight_dtype == "fp32" else "fp8-weights",
         "data": "synthetic",
-        "p50_ms": round(1000 * float(np.percentile(lats, 50)), 2),
-        "p99_ms": round(1000 * float(np.percentile(lats, 99)), 2),
-        "mean_batch": round(server.state.batcher.stats["batch_size_sum"] /
-                            max(server.state.batcher.stats["batches"], 1),
+        "p50_ms": round(1000 * float(np.percentile(lats_a, 50)), 2),
+        "p99_ms": round(1000 * float(np.percentile(lats_a, 99)), 2),
+        "mean_batch": round(st["batch_size_sum"] / max(st["batches"], 1),
                             2),
+        "wall_s": round(wall, 2),
         "config": {
             "model": f"{args.backbone}-FPN maskrcnn serving",
             "global_batch": args.max_batch,
             "image_size": args.image_size,
-            "concurrency": args.concurrency,
+            "clients": f"{args.client_procs}p x {args.client_threads}t",
             "parallelism": "dp1",
         },
     }
