@@ -41,7 +41,10 @@ def load_image_gt(dataset, config, image_id, augmentation=None,
             image = np.fliplr(image).copy()
             mask = np.fliplr(mask).copy()
     elif callable(augmentation):
-        image, mask = augmentation(image, mask)
+        try:
+            image, mask = augmentation(image, mask, rng)
+        except TypeError:  # simple (image, mask) callables
+            image, mask = augmentation(image, mask)
 
     # Exclude instances whose mask vanished in resize/crop.
     _idx = np.sum(mask, axis=(0, 1)) > 0
@@ -61,6 +64,59 @@ def load_image_gt(dataset, config, image_id, augmentation=None,
     image_meta = compose_image_meta(image_id, original_shape, image.shape,
                                     window, scale, active_class_ids)
     return image, image_meta, class_ids, bbox, mask
+
+
+def generate_random_rois(image_shape, count, gt_boxes, rng=None):
+    """Random roi proposals for RPN-free head training (reference
+    generate_random_rois, model.py:~1460): 90% placed around GT boxes
+    (2x-box neighborhoods), 10% anywhere. Returns [count, 4] pixel rois.
+    """
+    rng = rng or np.random
+    rois = np.zeros((count, 4), dtype=np.int32)
+    if gt_boxes is not None and len(gt_boxes):
+        rois_per_box = int(0.9 * count / gt_boxes.shape[0])
+        for i in range(gt_boxes.shape[0]):
+            gy1, gx1, gy2, gx2 = gt_boxes[i][:4]
+            h = gy2 - gy1
+            w = gx2 - gx1
+            r_y1 = max(gy1 - h, 0)
+            r_y2 = min(gy2 + h, image_shape[0])
+            r_x1 = max(gx1 - w, 0)
+            r_x2 = min(gx2 + w, image_shape[1])
+            while True:
+                y1y2 = rng.randint(r_y1, r_y2, (rois_per_box, 2))
+                x1x2 = rng.randint(r_x1, r_x2, (rois_per_box, 2))
+                threshold = 1
+                y1y2 = y1y2[np.abs(y1y2[:, 0] - y1y2[:, 1]) >=
+                            threshold][:rois_per_box]
+                x1x2 = x1x2[np.abs(x1x2[:, 0] - x1x2[:, 1]) >=
+                            threshold][:rois_per_box]
+                if y1y2.shape[0] == rois_per_box and \
+                        x1x2.shape[0] == rois_per_box:
+                    break
+            y1, y2 = np.split(np.sort(y1y2, axis=1), 2, axis=1)
+            x1, x2 = np.split(np.sort(x1x2, axis=1), 2, axis=1)
+            box_rois = np.hstack([y1, x1, y2, x2])
+            rois[rois_per_box * i:rois_per_box * (i + 1)] = box_rois
+        remaining_count = count - (rois_per_box * gt_boxes.shape[0])
+    else:
+        remaining_count = count
+    # The rest anywhere in the image.
+    while True:
+        y1y2 = rng.randint(0, image_shape[0], (remaining_count, 2))
+        x1x2 = rng.randint(0, image_shape[1], (remaining_count, 2))
+        threshold = 1
+        y1y2 = y1y2[np.abs(y1y2[:, 0] - y1y2[:, 1]) >=
+                    threshold][:remaining_count]
+        x1x2 = x1x2[np.abs(x1x2[:, 0] - x1x2[:, 1]) >=
+                    threshold][:remaining_count]
+        if y1y2.shape[0] == remaining_count and \
+                x1x2.shape[0] == remaining_count:
+            break
+    y1, y2 = np.split(np.sort(y1y2, axis=1), 2, axis=1)
+    x1, x2 = np.split(np.sort(x1x2, axis=1), 2, axis=1)
+    rois[-remaining_count:] = np.hstack([y1, x1, y2, x2])
+    return rois
 
 
 def build_rpn_targets(image_shape, anchors, gt_class_ids, gt_boxes, config,

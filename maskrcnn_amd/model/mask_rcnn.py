@@ -82,7 +82,8 @@ class MaskRCNNModel(nn.Module):
         return self.fpn(c2, c3, c4, c5)
 
     def forward(self, images, image_meta, anchors, gt_class_ids=None,
-                gt_boxes=None, gt_masks=None, mode="inference"):
+                gt_boxes=None, gt_masks=None, mode="inference",
+                input_rois=None):
         """
         images: [B, 3, H, W] molded; anchors: [A, 4] normalized;
         gt_*: training mode only (gt_boxes normalized, gt_masks
@@ -107,8 +108,13 @@ class MaskRCNNModel(nn.Module):
         t = self._mark("proposal", t)
 
         if mode == "training":
+            # USE_RPN_ROIS=False: train heads on externally provided rois
+            # (reference input_rois path, model.py:~1915).
+            target_rois = rpn_rois if cfg.USE_RPN_ROIS else input_rois
+            assert target_rois is not None, \
+                "USE_RPN_ROIS=False requires input_rois"
             rois, target_class_ids, target_deltas, target_masks = \
-                self.detection_target_layer(rpn_rois, gt_class_ids,
+                self.detection_target_layer(target_rois, gt_class_ids,
                                             gt_boxes, gt_masks)
             mrcnn_class_logits, mrcnn_class, mrcnn_bbox = \
                 self.classifier_head(rois, mrcnn_feature_maps)
@@ -234,6 +240,37 @@ class MaskRCNN:
                 raise RuntimeError(f"Missing keys: {allowed_missing[:5]}...")
         self.set_log_dir(path)
         return missing if exclude else []
+
+    def get_imagenet_weights(self):
+        """Reference downloads ResNet50-notop .h5 from the Keras release
+        bucket (model.py:~2100). This build is offline by design."""
+        raise RuntimeError(
+            "ImageNet bootstrap weights cannot be downloaded in this "
+            "offline build; pass a local checkpoint to load_weights() "
+            "instead.")
+
+    def get_trainable_layers(self):
+        """Modules that own at least one trainable parameter
+        (reference get_trainable_layers)."""
+        out = []
+        for name, module in self.model.named_modules():
+            if name and any(p.requires_grad
+                            for p in module.parameters(recurse=False)):
+                out.append((name, module))
+        return out
+
+    def find_trainable_layer(self, name: str):
+        """Innermost trainable module containing `name`
+        (reference find_trainable_layer/ancestor analog)."""
+        parts = name.split(".")
+        modules = dict(self.model.named_modules())
+        for i in range(len(parts), 0, -1):
+            cand = ".".join(parts[:i])
+            m = modules.get(cand)
+            if m is not None and any(p.requires_grad
+                                     for p in m.parameters(recurse=False)):
+                return cand, m
+        return None
 
     # ------------------------------------------------------- trainability
     LAYER_REGEX = {

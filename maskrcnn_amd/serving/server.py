@@ -22,7 +22,9 @@ import numpy as np
 from .batcher import DynamicBatcher
 from .engine import InferenceEngine
 from .savedmodel import find_latest_version
-from .wireproto import (ModelSpec, PredictRequest, PredictResponse,
+from .savedmodel import build_signature
+from .wireproto import (GetModelMetadataRequest, GetModelMetadataResponse,
+                        ModelSpec, PredictRequest, PredictResponse,
                         numpy_to_tensor, tensor_to_numpy)
 
 log = logging.getLogger(__name__)
@@ -180,11 +182,28 @@ class ModelServer:
                 log.exception("Predict failed")
                 context.abort(grpc.StatusCode.INTERNAL, str(e))
 
+        def get_model_metadata(request: GetModelMetadataRequest, context):
+            name = request.model_spec.name or outer.state.name
+            if name != outer.state.name:
+                context.abort(grpc.StatusCode.NOT_FOUND,
+                              f"unknown model {name!r}")
+            eng = outer.state.engine
+            sigs = {"serving_default": build_signature(eng.config)} \
+                if eng is not None else {}
+            return GetModelMetadataResponse(
+                model_spec=ModelSpec(name=outer.state.name,
+                                     version=outer.state.version),
+                signature_defs=sigs)
+
         handler = grpc.method_handlers_generic_handler(
             "tensorflow.serving.PredictionService",
             {"Predict": grpc.unary_unary_rpc_method_handler(
                 predict,
                 request_deserializer=PredictRequest.decode,
+                response_serializer=lambda r: r.encode()),
+             "GetModelMetadata": grpc.unary_unary_rpc_method_handler(
+                get_model_metadata,
+                request_deserializer=GetModelMetadataRequest.decode,
                 response_serializer=lambda r: r.encode())})
         server = grpc.server(
             cf.ThreadPoolExecutor(max_workers=16),

@@ -477,6 +477,86 @@ class SignatureDef:
         return s
 
 
+@dataclass
+class GetModelMetadataRequest:
+    """tensorflow.serving.GetModelMetadataRequest:
+    1 model_spec, 2 metadata_field repeated string."""
+    model_spec: ModelSpec = field(default_factory=ModelSpec)
+    metadata_field: List[str] = field(default_factory=list)
+
+    def encode(self) -> bytes:
+        out = bytearray()
+        _w_len_field(out, 1, self.model_spec.encode())
+        for f in self.metadata_field:
+            _w_string_field(out, 2, f)
+        return bytes(out)
+
+    @classmethod
+    def decode(cls, buf: bytes) -> "GetModelMetadataRequest":
+        r = cls()
+        pos = 0
+        while pos < len(buf):
+            fno, wt, v, pos = _r_field(buf, pos)
+            if fno == 1 and wt == _WT_LEN:
+                r.model_spec = ModelSpec.decode(v)
+            elif fno == 2 and wt == _WT_LEN:
+                r.metadata_field.append(v.decode("utf-8"))
+        return r
+
+
+SIGNATURE_DEF_MAP_TYPE_URL = \
+    "type.googleapis.com/tensorflow.serving.SignatureDefMap"
+
+
+@dataclass
+class GetModelMetadataResponse:
+    """tensorflow.serving.GetModelMetadataResponse:
+    1 model_spec, 2 metadata map<string, google.protobuf.Any>.
+    The 'signature_def' entry packs a SignatureDefMap
+    (1 signature_def map<string, SignatureDef>)."""
+    model_spec: ModelSpec = field(default_factory=ModelSpec)
+    signature_defs: Dict[str, SignatureDef] = field(default_factory=dict)
+
+    def encode(self) -> bytes:
+        sdm = bytearray()
+        for name, sig in self.signature_defs.items():
+            _w_len_field(sdm, 1, _encode_map_entry(name, sig.encode()))
+        any_msg = bytearray()
+        _w_string_field(any_msg, 1, SIGNATURE_DEF_MAP_TYPE_URL)
+        _w_len_field(any_msg, 2, bytes(sdm))
+        out = bytearray()
+        _w_len_field(out, 1, self.model_spec.encode())
+        _w_len_field(out, 2,
+                     _encode_map_entry("signature_def", bytes(any_msg)))
+        return bytes(out)
+
+    @classmethod
+    def decode(cls, buf: bytes) -> "GetModelMetadataResponse":
+        r = cls()
+        pos = 0
+        while pos < len(buf):
+            fno, wt, v, pos = _r_field(buf, pos)
+            if fno == 1 and wt == _WT_LEN:
+                r.model_spec = ModelSpec.decode(v)
+            elif fno == 2 and wt == _WT_LEN:
+                key, any_bytes = _decode_map_entry(v)
+                if key != "signature_def":
+                    continue
+                p = 0
+                payload = b""
+                while p < len(any_bytes):
+                    f2, w2, v2, p = _r_field(any_bytes, p)
+                    if f2 == 2 and w2 == _WT_LEN:
+                        payload = v2
+                p = 0
+                while p < len(payload):
+                    f2, w2, v2, p = _r_field(payload, p)
+                    if f2 == 1 and w2 == _WT_LEN:
+                        name, sv = _decode_map_entry(v2)
+                        r.signature_defs[name] = SignatureDef.decode(sv)
+        return r
+
+
 def encode_saved_model(signature_defs: Dict[str, SignatureDef],
                        tags=("serve",)) -> bytes:
     """tensorflow.SavedModel: 1 schema_version, 2 meta_graphs.
