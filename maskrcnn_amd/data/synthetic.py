@@ -50,10 +50,12 @@ class ShapesDataset(Dataset):
     def _random_shape(self, rng, height, width):
         shape = self.SHAPES[rng.randint(0, 3)]
         color = tuple([rng.randint(0, 255) for _ in range(3)])
-        buffer = 20
+        # Reference used a fixed 20 px buffer at 128²; scale it so small
+        # test images work too.
+        buffer = max(4, height // 7)
         y = rng.randint(buffer, height - buffer - 1)
         x = rng.randint(buffer, width - buffer - 1)
-        s = rng.randint(buffer, height // 4)
+        s = rng.randint(buffer, max(height // 4, buffer + 1))
         return shape, color, (x, y, s)
 
     def _random_image(self, rng, height, width):

@@ -1,0 +1,72 @@
+"""End-to-end learning evidence on the shapes fixture (CPU, bounded).
+
+Overfit a tiny shapes set and require real detections (AP50) on the
+trained images — exercises every stage: data plane, RPN/proposal
+matching, target sampling, heads, losses, optimizer, detection post,
+unmold, and the evaluator. (The reference's train_shapes.ipynb smoke,
+SURVEY §4, as an automated test; generalization runs are bench/GPU
+territory, not unit-test territory.)
+"""
+
+import numpy as np
+import pytest
+import torch
+
+
+@pytest.mark.timeout(600)
+def test_shapes_overfit_produces_detections():
+    from maskrcnn_amd.data.loader import make_data_loader
+    from maskrcnn_amd.data.synthetic import ShapesConfig, ShapesDataset
+    from maskrcnn_amd.engine.evaluator import evaluate_dataset
+    from maskrcnn_amd.engine.trainer import Trainer, build_optimizer
+    from maskrcnn_amd.model import MaskRCNN
+
+    class C(ShapesConfig):
+        IMAGES_PER_GPU = 2
+        BACKBONE = "resnet50"
+        PRE_NMS_LIMIT = 500
+        POST_NMS_ROIS_TRAINING = 200
+        POST_NMS_ROIS_INFERENCE = 50
+        TRAIN_ROIS_PER_IMAGE = 32
+        DETECTION_MIN_CONFIDENCE = 0.0
+        IMAGE_MIN_DIM = 64
+        IMAGE_MAX_DIM = 64
+        RPN_ANCHOR_SCALES = (8, 16, 24, 32, 48)
+
+    cfg = C()
+    torch.manual_seed(0)
+    np.random.seed(0)
+    ds = ShapesDataset()
+    ds.load_shapes(8, 64, 64, seed=10)
+    ds.prepare()
+
+    class InfC(C):
+        IMAGES_PER_GPU = 1
+
+    m = MaskRCNN("training", cfg, model_dir="/tmp/conv_logs")
+    m_inf = MaskRCNN("inference", InfC(), model_dir="/tmp/conv_logs")
+    m_inf.model.load_state_dict(m.model.state_dict())
+    base_ap = evaluate_dataset(m_inf, ds, verbose=0)["bbox"]["AP50"]
+
+    tr = Trainer(m, ds)
+    opt = build_optimizer(m.model, cfg, 0.02)
+    loader = make_data_loader(ds, cfg, shuffle=True, seed=1)
+    it = iter(loader)
+    first = last = None
+    for step in range(250):
+        try:
+            batch = next(it)
+        except StopIteration:
+            it = iter(loader)
+            batch = next(it)
+        losses = tr.train_step(batch, opt)
+        if step == 0:
+            first = losses["total_loss"]
+        last = losses["total_loss"]
+
+    assert last < first, (first, last)
+
+    m_inf.model.load_state_dict(m.model.state_dict())
+    stats = evaluate_dataset(m_inf, ds, verbose=0)
+    ap = stats["bbox"]["AP50"]
+    assert ap > max(base_ap + 0.1, 0.3), (base_ap, ap, stats)

@@ -1,0 +1,83 @@
+"""Property-based tests (hypothesis): wire-proto codec and box math."""
+
+import numpy as np
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from maskrcnn_amd.serving.wireproto import (PredictRequest, TensorProto,
+                                            ModelSpec, numpy_to_tensor,
+                                            tensor_to_numpy)
+from maskrcnn_amd.utils import boxes as B
+from maskrcnn_amd.utils.masks import rle_decode, rle_encode
+from maskrcnn_amd.data.coco import rle_from_string, rle_to_string
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.lists(st.integers(1, 6), min_size=1, max_size=4),
+       st.sampled_from([np.float32, np.int32, np.int64, np.bool_]))
+def test_tensor_roundtrip_any_shape(shape, dtype):
+    rng = np.random.RandomState(0)
+    if dtype == np.bool_:
+        arr = rng.rand(*shape) > 0.5
+    elif dtype == np.float32:
+        arr = (rng.randn(*shape) * 1e3).astype(dtype)
+    else:
+        arr = rng.randint(-10**6, 10**6, size=shape).astype(dtype)
+    t = numpy_to_tensor(arr)
+    back = tensor_to_numpy(TensorProto.decode(t.encode()))
+    np.testing.assert_array_equal(back, arr)
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.text(min_size=0, max_size=40),
+       st.one_of(st.none(), st.integers(-2**40, 2**40)),
+       st.text(min_size=0, max_size=20))
+def test_model_spec_roundtrip(name, version, sig):
+    m = ModelSpec(name=name, version=version, signature_name=sig)
+    d = ModelSpec.decode(m.encode())
+    assert d.name == name
+    assert d.version == version
+    assert d.signature_name == sig
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(1, 40), st.integers(1, 40), st.integers(0, 2**31))
+def test_rle_string_roundtrip_fuzz(h, w, seed):
+    rng = np.random.RandomState(seed % (2**31))
+    mask = rng.rand(h, w) > rng.rand()
+    rle = rle_encode(mask)
+    back = rle_from_string(rle_to_string(rle), rle["size"])
+    np.testing.assert_array_equal(rle_decode(back), mask)
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(1, 64), st.integers(0, 2**31))
+def test_box_refinement_roundtrip_fuzz(n, seed):
+    rng = np.random.RandomState(seed % (2**31))
+    src = rng.rand(n, 4)
+    src[:, 2:] = src[:, :2] + rng.rand(n, 2) + 0.05
+    dst = rng.rand(n, 4)
+    dst[:, 2:] = dst[:, :2] + rng.rand(n, 2) + 0.05
+    rec = B.apply_box_deltas(src, B.box_refinement(src, dst))
+    np.testing.assert_allclose(rec, dst, atol=1e-4)
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.integers(2, 100), st.floats(0.1, 0.9), st.integers(0, 2**31))
+def test_nms_invariants(n, thresh, seed):
+    import torch
+    from maskrcnn_amd.ops import nms
+    rng = np.random.RandomState(seed % (2**31))
+    boxes = rng.rand(n, 4).astype(np.float32)
+    boxes[:, 2:] = boxes[:, :2] + rng.rand(n, 2).astype(np.float32) + 0.01
+    scores = rng.rand(n).astype(np.float32)
+    keep = nms(torch.from_numpy(boxes), torch.from_numpy(scores), thresh)
+    k = keep.numpy()
+    # Kept indices unique; scores non-increasing; survivors mutually
+    # below the IoU threshold.
+    assert len(set(k.tolist())) == len(k)
+    s = scores[k]
+    assert (np.diff(s) <= 1e-6).all()
+    ov = B.compute_overlaps(boxes[k], boxes[k])
+    np.fill_diagonal(ov, 0)
+    assert (ov <= thresh + 1e-5).all()
