@@ -10,7 +10,6 @@ from __future__ import annotations
 
 import os
 import subprocess
-import sys
 import sysconfig
 from concurrent.futures import ThreadPoolExecutor
 from pathlib import Path

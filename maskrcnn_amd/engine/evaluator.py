@@ -9,7 +9,6 @@ from __future__ import annotations
 
 import time
 
-import numpy as np
 
 from ..utils.boxes import extract_bboxes
 from ..utils.cocoeval import COCOEvalLite

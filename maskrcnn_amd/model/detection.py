@@ -18,7 +18,7 @@ import torch
 import torch.nn as nn
 
 from ..ops import batched_nms
-from ..utils.boxes import apply_box_deltas_t, clip_boxes_t, norm_boxes_t
+from ..utils.boxes import apply_box_deltas_t, norm_boxes_t
 from ..utils.meta import parse_image_meta_t
 
 

@@ -12,7 +12,6 @@ deconv 2x2/2 -> 1x1 conv num_classes + sigmoid -> [B, R, C, 28, 28].
 
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
 import torch.nn.functional as F
 

@@ -14,7 +14,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ..ops import batched_nms, nms
+from ..ops import batched_nms
 from ..utils.boxes import apply_box_deltas_t, clip_boxes_t
 
 

@@ -9,7 +9,6 @@ if a cuda tensor reaches an op and the extension is absent, we raise.
 from __future__ import annotations
 
 import importlib
-import os
 
 _ext = None
 _tried = False

@@ -23,7 +23,7 @@ from __future__ import annotations
 import torch
 import torch.nn.functional as F
 
-from ._ext import load_ext, require_ext
+from ._ext import require_ext
 
 
 class _RoiAlignHIP(torch.autograd.Function):

@@ -15,8 +15,7 @@ import random
 
 import numpy as np
 
-from .utils.boxes import compute_overlaps
-from .utils.metrics import compute_matches, trim_zeros
+from .utils.metrics import compute_matches
 
 
 def _plt():

@@ -4,8 +4,8 @@ import numpy as np
 from hypothesis import given, settings
 from hypothesis import strategies as st
 
-from maskrcnn_amd.serving.wireproto import (PredictRequest, TensorProto,
-                                            ModelSpec, numpy_to_tensor,
+from maskrcnn_amd.serving.wireproto import (ModelSpec, TensorProto,
+                                            numpy_to_tensor,
                                             tensor_to_numpy)
 from maskrcnn_amd.utils import boxes as B
 from maskrcnn_amd.utils.masks import rle_decode, rle_encode

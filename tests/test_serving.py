@@ -2,7 +2,6 @@
 full gRPC + REST Predict round trip on CPU (shapes-scale model)."""
 
 import os
-import threading
 import time
 
 import numpy as np
