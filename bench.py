@@ -24,7 +24,10 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 # (profiles/r01_naive_conv.md); re-exec once if the shell didn't set it.
 from maskrcnn_amd._rocm_env import ensure_env_or_reexec  # noqa: E402
 
-ensure_env_or_reexec()
+if __name__ == "__main__":
+    # re-exec only when run as a script (library imports must never
+    # replace the host process; see maskrcnn_amd/_rocm_env.py)
+    ensure_env_or_reexec()
 
 import numpy as np  # noqa: E402
 import torch  # noqa: E402

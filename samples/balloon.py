@@ -22,7 +22,10 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))))
 from maskrcnn_amd._rocm_env import ensure_env_or_reexec  # noqa: E402
 
-ensure_env_or_reexec()
+if __name__ == "__main__":
+    # re-exec only when run as a script (library imports must never
+    # replace the host process; see maskrcnn_amd/_rocm_env.py)
+    ensure_env_or_reexec()
 
 from maskrcnn_amd.config import Config  # noqa: E402
 from maskrcnn_amd.data.dataset import Dataset  # noqa: E402

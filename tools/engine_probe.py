@@ -3,7 +3,10 @@
 import os, sys, time, logging
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from maskrcnn_amd._rocm_env import ensure_env_or_reexec
-ensure_env_or_reexec()
+if __name__ == "__main__":
+    # re-exec only when run as a script (library imports must never
+    # replace the host process; see maskrcnn_amd/_rocm_env.py)
+    ensure_env_or_reexec()
 logging.basicConfig(level=logging.INFO, format="%(asctime)s %(message)s")
 import numpy as np
 import torch
