@@ -28,6 +28,8 @@ void conv3x3_fwd_launch(const void*, const void*, const float*, void*, int,
                         int, int, int, int, bool, hipStream_t);
 void conv3x3_v2_launch(const void*, const void*, const float*, const void*,
                        void*, int, int, int, int, int, bool, hipStream_t);
+void conv3x3_v3_launch(const void*, const void*, const float*, const void*,
+                       void*, int, int, int, int, int, bool, hipStream_t);
 }
 
 namespace {
@@ -175,12 +177,10 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
       {B, Cout, H, W},
       x.options().memory_format(at::MemoryFormat::ChannelsLast));
   const long M = (long)B * H * W;
-  bool use_v2;
-  if (version == 1) use_v2 = false;
-  else if (version == 2) use_v2 = true;
-  else  // auto: the 128x128 glds kernel needs enough M x N tiles
-    use_v2 = (M >= 32768 && Cout >= 128);
-  if (use_v2) {
+  int ver = (int)version;
+  if (ver == 0)  // auto: the 128x128 glds kernels need enough M x N tiles
+    ver = (M >= 32768 && Cout >= 128) ? 2 : 1;
+  if (ver >= 2) {
     // Persistent 64-B zero page per device (glds cannot predicate
     // out-of-range lanes; they DMA from here instead).
     static std::unordered_map<int, torch::Tensor> zero_pages;
@@ -190,10 +190,10 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
       it = zero_pages.emplace(
           dev, torch::zeros({32}, x.options())).first;
     }
-    conv3x3_v2_launch(x.data_ptr(), w_packed.data_ptr(),
-                      bias_f.data_ptr<float>(), it->second.data_ptr(),
-                      out.data_ptr(), B, H, W, Cin, Cout, relu,
-                      cur_stream());
+    auto launch = (ver == 3) ? conv3x3_v3_launch : conv3x3_v2_launch;
+    launch(x.data_ptr(), w_packed.data_ptr(), bias_f.data_ptr<float>(),
+           it->second.data_ptr(), out.data_ptr(), B, H, W, Cin, Cout,
+           relu, cur_stream());
   } else {
     conv3x3_fwd_launch(x.data_ptr(), w_packed.data_ptr(),
                        bias_f.data_ptr<float>(), out.data_ptr(), B, H, W,

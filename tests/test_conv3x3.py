@@ -86,14 +86,16 @@ def test_gpu_kernel_matches_reference(cin, cout, h, w):
 
 
 @pytest.mark.gpu
+@pytest.mark.parametrize("version", [2, 3])
 @pytest.mark.parametrize("cin,cout,h,w,batch", [
     (256, 256, 64, 64, 2),    # exact-tile case
     (256, 256, 50, 30, 3),    # ragged M
     (64, 192, 64, 64, 2),     # Cout not multiple of 128
     (128, 512, 32, 32, 2),
 ])
-def test_gpu_v2_kernel_matches_reference(cin, cout, h, w, batch):
-    """The 128x128-tile glds variant, forced via version=2."""
+def test_gpu_v2_kernel_matches_reference(cin, cout, h, w, batch, version):
+    """The 128x128-tile glds variants (v2 = 2-buffer __syncthreads,
+    v3 = 3-buffer counted-vmcnt + raw barriers + XCD swizzle)."""
     from maskrcnn_amd.ops._ext import require_ext
     from maskrcnn_amd.ops.conv3x3 import pack_weight
     ext = require_ext()
@@ -103,9 +105,30 @@ def test_gpu_v2_kernel_matches_reference(cin, cout, h, w, batch):
         .to(memory_format=torch.channels_last)
     wt = (torch.randn(cout, cin, 3, 3, device=dev) * 0.05)
     b = torch.randn(cout, device=dev)
-    out = ext.conv3x3_fwd(x, pack_weight(wt), b.float(), True, 2)
+    out = ext.conv3x3_fwd(x, pack_weight(wt), b.float(), True, version)
     ref = F.relu(F.conv2d(x.float(), wt, b, padding=1))
     torch.testing.assert_close(out.float(), ref, atol=0.3, rtol=0.05)
+
+
+@pytest.mark.gpu
+def test_gpu_v3_repeated_runs_stable():
+    """Race screen for the counted-vmcnt schedule: byte-identical results
+    across repeated launches (cdna_hip_programming.md two-lane
+    discipline for sync-structure edits)."""
+    from maskrcnn_amd.ops._ext import require_ext
+    from maskrcnn_amd.ops.conv3x3 import pack_weight
+    ext = require_ext()
+    dev = torch.device("cuda:0")
+    torch.manual_seed(1)
+    x = torch.randn(2, 256, 64, 64, device=dev).bfloat16() \
+        .to(memory_format=torch.channels_last)
+    wt = (torch.randn(256, 256, 3, 3, device=dev) * 0.05)
+    wp = pack_weight(wt)
+    b = torch.zeros(256, device=dev).float()
+    first = ext.conv3x3_fwd(x, wp, b, False, 3)
+    for _ in range(10):
+        again = ext.conv3x3_fwd(x, wp, b, False, 3)
+        assert torch.equal(first, again)
 
 
 @pytest.mark.gpu
