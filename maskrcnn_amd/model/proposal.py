@@ -53,25 +53,32 @@ class ProposalLayer(nn.Module):
         window = torch.tensor([0., 0., 1., 1.], device=boxes.device)
         boxes = clip_boxes_t(boxes, window)
 
-        # One NMS launch for ALL images (image-index offset trick) —
-        # B sequential NMS calls each cost a host sync; at serving
-        # batch 32 that serializes badly.
+        # NMS for a GROUP of images per launch (image-index offset
+        # trick): per-image NMS costs one host sync each (serializes at
+        # serving batch 32), but one launch over ALL images makes the
+        # bitmask workspace quadratic in B*P (4.6 GB at 32x6000). Chunk
+        # so each launch stays <= ~16k boxes.
         P = boxes.shape[1]
-        flat_boxes = boxes.reshape(-1, 4)
-        flat_scores = top_scores.reshape(-1)
-        img_ix = torch.arange(B, device=boxes.device) \
-            .repeat_interleave(P)
-        keep = batched_nms(flat_boxes, flat_scores, img_ix,
-                           cfg.RPN_NMS_THRESHOLD)
+        group = max(1, 16384 // P)
         rois = boxes.new_zeros((B, proposal_count, 4))
-        if keep.numel():
+        for g0 in range(0, B, group):
+            g1 = min(g0 + group, B)
+            nb = g1 - g0
+            flat_boxes = boxes[g0:g1].reshape(-1, 4)
+            flat_scores = top_scores[g0:g1].reshape(-1)
+            img_ix = torch.arange(nb, device=boxes.device) \
+                .repeat_interleave(P)
+            keep = batched_nms(flat_boxes, flat_scores, img_ix,
+                               cfg.RPN_NMS_THRESHOLD)
+            if not keep.numel():
+                continue
             img = img_ix[keep]                         # [K], score-desc
             # Per-image rank of each kept box (global order preserves
             # per-image score order).
             onehot = (img.unsqueeze(1) ==
-                      torch.arange(B, device=img.device).unsqueeze(0))
+                      torch.arange(nb, device=img.device).unsqueeze(0))
             rank = onehot.long().cumsum(dim=0)[
                 torch.arange(img.numel(), device=img.device), img] - 1
             sel = rank < proposal_count
-            rois[img[sel], rank[sel]] = flat_boxes[keep[sel]]
+            rois[g0 + img[sel], rank[sel]] = flat_boxes[keep[sel]]
         return rois
