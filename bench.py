@@ -190,6 +190,9 @@ def main():
                     help="override dist backend (nccl|gloo)")
     ap.add_argument("--phase-timing", action="store_true",
                     help="print a per-phase GPU/wall breakdown to stderr")
+    ap.add_argument("--torch-profile", metavar="DIR", default=None,
+                    help="capture a kineto (torch.profiler) chrome trace "
+                         "of 3 steady steps into DIR")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -268,6 +271,26 @@ def main():
     if args.phase_timing and rank == 0:
         _phase_timing(model, config, batch, anchors, meta, optimizer,
                       use_bf16, n_iters=5)
+
+    if args.torch_profile and rank == 0:
+        from torch.profiler import (ProfilerActivity, profile,
+                                    tensorboard_trace_handler)
+        activities = [ProfilerActivity.CPU]
+        if use_cuda:
+            activities.append(ProfilerActivity.CUDA)
+        with profile(activities=activities, record_shapes=False,
+                     with_stack=False) as prof:
+            for _ in range(3):
+                one_step()
+            if use_cuda:
+                torch.cuda.synchronize()
+        os.makedirs(args.torch_profile, exist_ok=True)
+        trace = os.path.join(args.torch_profile, "bench_trace.json")
+        prof.export_chrome_trace(trace)
+        print(f"torch.profiler trace -> {trace}", file=sys.stderr)
+        print(prof.key_averages().table(
+            sort_by="cuda_time_total" if use_cuda else "cpu_time_total",
+            row_limit=25), file=sys.stderr)
 
     if distributed:
         import torch.distributed as dist
