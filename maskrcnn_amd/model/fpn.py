@@ -40,8 +40,11 @@ class FPN(nn.Module):
         # Fused path in eval/serving (measured 1.4-1.9x, profiles/
         # r01_conv_ab.md). In training the end-to-end delta was within
         # noise vs MIOpen but trended slower on one box — kept on the
-        # known-good path until the round-2 A/B (same backward either way).
-        if self.use_fused and not self.training:
+        # known-good path until the round-2 A/B (same backward either
+        # way). MASKRCNN_TRAIN_FUSED_FPN=1 enables it for that A/B.
+        import os
+        if self.use_fused and (not self.training or
+                               os.environ.get("MASKRCNN_TRAIN_FUSED_FPN")):
             p5 = fpn_fuse(c5, self.fpn_c5p5.weight, self.fpn_c5p5.bias)
             p4 = fpn_fuse(c4, self.fpn_c4p4.weight, self.fpn_c4p4.bias, p5)
             p3 = fpn_fuse(c3, self.fpn_c3p3.weight, self.fpn_c3p3.bias, p4)

@@ -80,7 +80,11 @@ class Bottleneck(nn.Module):
         # HIP conv in eval/serving; during training the MIOpen
         # autograd path measured faster end-to-end (fwd win < bwd-algo
         # delta) — revisit with the conv3x3 bwd kernel.
-        if self._fused_mid and out.is_cuda and not self.training and \
+        # MASKRCNN_TRAIN_HIP_CONV=1 enables the round-2 A/B.
+        import os
+        if self._fused_mid and out.is_cuda and \
+                (not self.training or
+                 os.environ.get("MASKRCNN_TRAIN_HIP_CONV")) and \
                 use_hip_conv3x3(out.shape[0], out.shape[1], out.shape[2],
                                 out.shape[3]):
             from ..ops.conv3x3 import conv3x3_bn_relu

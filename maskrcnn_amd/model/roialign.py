@@ -51,8 +51,11 @@ class PyramidROIAlign(nn.Module):
 
         # channels_last output feeds the HIP conv path in the heads —
         # eval/serving only (training keeps the round-1-validated NCHW
-        # layout; round-2 A/B pending).
-        cl = (feature_maps[0].is_cuda and not self.training and
+        # layout; MASKRCNN_TRAIN_CL_ROI=1 enables the round-2 A/B).
+        import os
+        cl = (feature_maps[0].is_cuda and
+              (not self.training or
+               os.environ.get("MASKRCNN_TRAIN_CL_ROI")) and
               feature_maps[0].is_contiguous(
                   memory_format=torch.channels_last))
         out = torch.empty(

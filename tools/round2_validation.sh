@@ -1,0 +1,59 @@
+#!/bin/bash
+# Round-2 opening GPU session plan — spend the fresh 90 GPU-min budget
+# front-loaded. Run pieces via gpurun, NOT all at once; each block is
+# independently useful. See NOTES_ROUND2.md for context.
+#
+#   gpurun --timeout 1200 -- 'bash tools/round2_validation.sh block1'
+set -x
+cd "${GRAFT_REPO_ROOT:-/root/repo}"
+mkdir -p gpurun_out
+
+case "${1:-block1}" in
+block1)  # v3 kernel validation + conv A/B (~4 min)
+  timeout 400 python -m pytest tests/test_conv3x3.py -m gpu -q \
+      --runxfail > gpurun_out/r2_conv_tests.log 2>&1
+  tail -4 gpurun_out/r2_conv_tests.log
+  timeout 400 python tools/conv_bench.py > gpurun_out/r2_conv_ab.log 2>&1
+  grep -v libdrm gpurun_out/r2_conv_ab.log
+  ;;
+block2)  # training-path A/Bs (one box; ~12 min incl. one find phase)
+  run() { timeout 500 python bench.py --steps 15 --warmup 5 \
+          > "gpurun_out/r2_$1.json" 2>/dev/null; cat "gpurun_out/r2_$1.json"; }
+  echo "=== baseline"; run base
+  echo "=== fused fpn";        MASKRCNN_TRAIN_FUSED_FPN=1 run fpn
+  echo "=== hip conv";         MASKRCNN_TRAIN_HIP_CONV=1 run conv
+  echo "=== cl roi";           MASKRCNN_TRAIN_CL_ROI=1 run clroi
+  echo "=== all three";        MASKRCNN_TRAIN_FUSED_FPN=1 \
+      MASKRCNN_TRAIN_HIP_CONV=1 MASKRCNN_TRAIN_CL_ROI=1 run all
+  ;;
+block3)  # serving bench (~8 min incl. 265 s batch-32 find)
+  timeout 900 python bench/serving_bench.py --requests 96 \
+      --client-procs 4 --client-threads 8 --warmup 16 \
+      > gpurun_out/r2_serving.json 2> gpurun_out/r2_serving.err
+  cat gpurun_out/r2_serving.json
+  grep -E "WARNING|batch of" gpurun_out/r2_serving.err | tail -5
+  ;;
+block4)  # steady-state rocprof of the tuned step (profile AFTER warmup
+         # by profiling a second process on the warm box: find-db is
+         # box-local, so run bench once unprofiled first)
+  cd /tmp && export TMPDIR=/tmp && cd "$GRAFT_REPO_ROOT"
+  timeout 400 python bench.py --steps 3 --warmup 2 > /dev/null 2>&1
+  mkdir -p gpurun_out/r2_prof
+  timeout 500 rocprofv3 --kernel-trace --stats --output-format csv \
+      -d gpurun_out/r2_prof -o r2 -- \
+      python bench.py --steps 15 --warmup 3 \
+      > gpurun_out/r2_prof_bench.json 2>/dev/null
+  cat gpurun_out/r2_prof_bench.json
+  python tools/gap_analysis.py "gpurun_out/r2_prof/*kernel_trace.csv"
+  find gpurun_out/r2_prof -name "*kernel_trace*" -delete
+  python tools/prof_summary.py "gpurun_out/r2_prof/*kernel_stats.csv" 25
+  ;;
+block5)  # PMC counters for the hand kernels (own run, never with traces)
+  cd /tmp && export TMPDIR=/tmp && cd "$GRAFT_REPO_ROOT"
+  mkdir -p gpurun_out/r2_pmc
+  timeout 500 rocprofv3 --pmc SQ_VALU_MFMA_BUSY_CYCLES,SQ_WAVE_CYCLES,SQ_LDS_BANK_CONFLICT \
+      --output-format csv -d gpurun_out/r2_pmc -o pmc -- \
+      python tools/conv_bench.py > gpurun_out/r2_pmc.log 2>&1
+  ls gpurun_out/r2_pmc
+  ;;
+esac
