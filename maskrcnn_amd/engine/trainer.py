@@ -61,8 +61,16 @@ class Trainer:
         self.rank = dist.get_rank() if self.distributed else 0
         self.ddp = None
         if self.distributed:
-            from ..distributed import DistributedTrainer
-            self.ddp = DistributedTrainer(self.model, self.config)
+            self._rebuild_ddp()
+
+    def _rebuild_ddp(self):
+        """(Re)build the bucketed all-reduce wrapper over the CURRENT
+        trainable set — must re-run after set_trainable() changes which
+        params get grads (buckets of frozen params would never fill)."""
+        from ..distributed import DistributedTrainer
+        if self.ddp is not None:
+            self.ddp.remove_hooks()
+        self.ddp = DistributedTrainer(self.model, self.config)
 
     def _autocast(self):
         use_bf16 = (self.config.COMPUTE_DTYPE == "bf16" and
@@ -130,6 +138,8 @@ class Trainer:
     def fit(self, learning_rate, epochs, layers="all", steps_per_epoch=None):
         cfg = self.config
         self.wrapper.set_trainable(layers)
+        if self.distributed:
+            self._rebuild_ddp()
         optimizer = build_optimizer(self.model, cfg, learning_rate)
         loader = make_data_loader(self.train_dataset, cfg,
                                   augmentation=self.augmentation,
