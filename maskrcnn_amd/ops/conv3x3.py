@@ -29,6 +29,10 @@ class _Conv3x3Fn(torch.autograd.Function):
         ctx.save_for_backward(x, w)
         ctx.relu = relu
         ext = load_ext()
+        if x.is_cuda and ext is None:
+            # Never silently degrade to eager on a GPU box.
+            from ._ext import require_ext
+            require_ext()
         use_hip = (ext is not None and x.is_cuda and
                    x.dtype == torch.bfloat16 and
                    x.is_contiguous(memory_format=torch.channels_last) and

@@ -27,6 +27,10 @@ class _FpnFuseFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, up):
         ext = load_ext()
+        if x.is_cuda and ext is None:
+            # Never silently degrade to eager on a GPU box.
+            from ._ext import require_ext
+            require_ext()
         use_hip = (ext is not None and x.is_cuda and
                    x.dtype == torch.bfloat16 and
                    x.is_contiguous(memory_format=torch.channels_last) and
