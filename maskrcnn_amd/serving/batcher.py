@@ -49,11 +49,15 @@ class DynamicBatcher:
         self.stats = {"batches": 0, "requests": 0, "batch_size_sum": 0}
 
     def submit(self, input_image: np.ndarray, input_image_meta: np.ndarray,
-               input_anchors: Optional[np.ndarray] = None) -> Future:
+               input_anchors: Optional[np.ndarray] = None,
+               outputs: Optional[list] = None) -> Future:
         """Enqueue one request (batch dim must be 1..max_batch_size).
-        Returns a Future resolving to the outputs dict (numpy)."""
+        `outputs` restricts which signature outputs are computed back to
+        the host (None = all 7). Returns a Future resolving to the
+        outputs dict (numpy)."""
         fut: Future = Future()
-        self.queue.put((input_image, input_image_meta, input_anchors, fut))
+        self.queue.put((input_image, input_image_meta, input_anchors,
+                        outputs, fut))
         return fut
 
     def close(self):
@@ -99,6 +103,10 @@ class DynamicBatcher:
             images = np.concatenate([b[0] for b in batch], axis=0)
             metas = np.concatenate([b[1] for b in batch], axis=0)
             anchors = batch[0][2]
+            # Union of the batch's output filters (None = everything).
+            outputs = None
+            if all(b[3] is not None for b in batch):
+                outputs = sorted({n for b in batch for n in b[3]})
             real_n = images.shape[0]
             if self.pad_to_max and real_n < self.max_batch_size:
                 pad = self.max_batch_size - real_n
@@ -111,14 +119,14 @@ class DynamicBatcher:
                     [metas, np.repeat(metas[:1], pad, axis=0)], axis=0)
             t0 = time.monotonic()
             out = self.engine.predict(images, metas, anchors,
-                                      trim_to=real_n)
+                                      trim_to=real_n, outputs=outputs)
             log.info("batch of %d (padded %d) in %.3fs", real_n,
                      images.shape[0], time.monotonic() - t0)
             self.stats["batches"] += 1
             self.stats["requests"] += len(batch)
             self.stats["batch_size_sum"] += images.shape[0]
             off = 0
-            for img, _, _, fut in batch:
+            for img, _, _, _, fut in batch:
                 k = img.shape[0]
                 sliced = {name: arr[off:off + k]
                           for name, arr in out.items()}

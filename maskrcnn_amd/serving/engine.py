@@ -42,10 +42,15 @@ class InferenceEngine:
         self.anchor_cache = AnchorCache(config)
         self._lock = threading.Lock()  # one inference stream per engine
 
+    #: the 7 signature outputs (S1 order)
+    OUTPUT_NAMES = ("mrcnn_detection", "mrcnn_class", "mrcnn_bbox",
+                    "mrcnn_mask", "rois", "rpn_class", "rpn_bbox")
+
     def predict(self, input_image: np.ndarray,
                 input_image_meta: np.ndarray,
                 input_anchors: Optional[np.ndarray] = None,
-                trim_to: Optional[int] = None
+                trim_to: Optional[int] = None,
+                outputs: Optional[list] = None
                 ) -> Dict[str, np.ndarray]:
         """Run the Predict signature.
 
@@ -76,24 +81,21 @@ class InferenceEngine:
             with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
                                 enabled=self.use_bf16):
                 out = self.model(images, metas, anchors, mode="inference")
-            # Trim the padded batch rows GPU-side BEFORE DtoH: a padded
-            # batch-32 output set is ~1 GB; real rows are what the caller
-            # needs (batcher pad-to-max).
+            # Trim the padded batch rows GPU-side BEFORE DtoH (a padded
+            # batch-32 output set is ~1 GB) and transfer only the
+            # requested outputs (a full set adds ~12 MB/image of rpn/roi
+            # tensors most clients never read).
             k = trim_to if trim_to is not None else B
-            # Torch [B,D,C,mh,mw] -> TF signature layout [B,D,mh,mw,C].
-            mask = out["mrcnn_mask"][:k].float().permute(0, 1, 3, 4, 2)
-            return {
-                "mrcnn_detection":
-                    out["mrcnn_detection"][:k].float().cpu().numpy(),
-                "mrcnn_class":
-                    out["mrcnn_class"][:k].float().cpu().numpy(),
-                "mrcnn_bbox":
-                    out["mrcnn_bbox"][:k].float().cpu().numpy(),
-                "mrcnn_mask": mask.contiguous().cpu().numpy(),
-                "rois": out["rois"][:k].float().cpu().numpy(),
-                "rpn_class": out["rpn_class"][:k].float().cpu().numpy(),
-                "rpn_bbox": out["rpn_bbox"][:k].float().cpu().numpy(),
-            }
+            wanted = [n for n in self.OUTPUT_NAMES
+                      if outputs is None or n in outputs]
+            result = {}
+            for name in wanted:
+                t = out[name][:k].float()
+                if name == "mrcnn_mask":
+                    # Torch [B,D,C,mh,mw] -> TF layout [B,D,mh,mw,C].
+                    t = t.permute(0, 1, 3, 4, 2).contiguous()
+                result[name] = t.cpu().numpy()
+            return result
 
     @classmethod
     def from_saved_model(cls, export_dir: str, device=None,

@@ -84,13 +84,13 @@ class ModelState:
             VERSION.set(latest)
         return True
 
-    def predict(self, images, metas, anchors=None):
+    def predict(self, images, metas, anchors=None, outputs=None):
         with self._lock:
             batcher = self.batcher
             version = self.version
         if batcher is None:
             raise RuntimeError(f"model {self.name} has no loaded version")
-        fut = batcher.submit(images, metas, anchors)
+        fut = batcher.submit(images, metas, anchors, outputs=outputs)
         return fut.result(timeout=300.0), version
 
     def warmup(self, image_shape=None):
@@ -149,7 +149,11 @@ class ModelServer:
         if "input_anchors" in request.inputs:
             anchors = tensor_to_numpy(
                 request.inputs["input_anchors"]).astype(np.float32)
-        outputs, version = self.state.predict(images, metas, anchors)
+        # The filter is pushed into the engine: unrequested outputs skip
+        # the DtoH transfer entirely.
+        outputs, version = self.state.predict(
+            images, metas, anchors,
+            outputs=list(request.output_filter) or None)
         if request.output_filter:
             outputs = {k: v for k, v in outputs.items()
                        if k in request.output_filter}

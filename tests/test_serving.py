@@ -180,7 +180,8 @@ class _FakeEngine:
     def __init__(self):
         self.calls = []
 
-    def predict(self, images, metas, anchors=None, trim_to=None):
+    def predict(self, images, metas, anchors=None, trim_to=None,
+                outputs=None):
         self.calls.append(images.shape[0])
         time.sleep(0.01)
         out = images.sum(axis=(1, 2, 3))
