@@ -49,18 +49,34 @@ def _box_iou_matrix(dt, gt, iscrowd):
     return ious
 
 
-def _mask_iou_matrix(dt, gt, iscrowd):
-    dms = [m if isinstance(m, np.ndarray) else rle_decode(m) for m in dt]
-    gms = [m if isinstance(m, np.ndarray) else rle_decode(m) for m in gt]
+def _mask_iou_matrix(dt, gt, iscrowd, decode_cache=None):
+    def dec(m):
+        if isinstance(m, np.ndarray):
+            return m
+        if decode_cache is None:
+            return rle_decode(m)
+        key = id(m)
+        cached = decode_cache.get(key)
+        if cached is None:
+            cached = rle_decode(m)
+            decode_cache[key] = cached
+        return cached
+
+    dms = [dec(m) for m in dt]
+    gms = [dec(m) for m in gt]
+    dareas = [int(m.sum()) for m in dms]
+    gareas = [int(m.sum()) for m in gms]
     ious = np.zeros((len(dms), len(gms)))
     for j, gm in enumerate(gms):
-        garea = gm.sum()
         for i, dm in enumerate(dms):
+            # Quick reject via area bound before touching pixels.
+            if dareas[i] == 0 or gareas[j] == 0:
+                continue
             inter = np.logical_and(dm, gm).sum()
             if inter == 0:
                 continue
-            darea = dm.sum()
-            union = darea if iscrowd[j] else darea + garea - inter
+            union = dareas[i] if iscrowd[j] \
+                else dareas[i] + gareas[j] - inter
             ious[i, j] = inter / max(union, 1e-10)
     return ious
 
@@ -91,6 +107,9 @@ class COCOEvalLite:
         for a in dt_anns:
             self.dt_by_img_cat[(a["image_id"], a["category_id"])].append(a)
             self.imgs.add(a["image_id"])
+        # RLE decode cache: an annotation's mask is needed once per
+        # (category, IoU-pass); decoding dominates segm eval without it.
+        self._decode_cache: dict = {}
 
     def _match_image(self, img, cat, area_rng):
         """Returns (dt_scores, dt_matched [T, D], dt_ignored [T, D],
@@ -112,7 +131,7 @@ class COCOEvalLite:
         else:
             ious = _mask_iou_matrix([d["segmentation"] for d in dts],
                                     [g["segmentation"] for g in gts],
-                                    iscrowd)
+                                    iscrowd, self._decode_cache)
         T = len(self.iou_thrs)
         D = len(dts)
         G = len(gts)
