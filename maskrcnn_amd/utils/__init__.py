@@ -10,7 +10,9 @@ from .boxes import (  # noqa: F401
     apply_box_deltas_t, clip_boxes_t, box_refinement_t, overlaps_t,
 )
 from .anchors import generate_anchors, generate_pyramid_anchors  # noqa: F401
-from .image import resize_image, resize, mold_image, unmold_image  # noqa: F401
+from .image import (  # noqa: F401
+    resize_image, resize, mold_image, unmold_image, download_trained_weights,
+)
 from .masks import (  # noqa: F401
     resize_mask, minimize_mask, expand_mask, mold_mask, unmold_mask,
 )

@@ -138,3 +138,13 @@ def mold_image(images: np.ndarray, config) -> np.ndarray:
 def unmold_image(normalized_images: np.ndarray, config) -> np.ndarray:
     """Reverse of mold_image."""
     return (normalized_images + np.array(config.MEAN_PIXEL)).astype(np.uint8)
+
+
+def download_trained_weights(coco_model_path, verbose=1):
+    """Reference utils.download_trained_weights fetched the COCO .h5 from
+    the Mask_RCNN GitHub releases (utils.py:~870). This build is offline
+    by design — kept for API parity with an actionable error."""
+    raise RuntimeError(
+        "Pretrained COCO weights cannot be downloaded in this offline "
+        "build. Train with samples/coco.py or provide a local checkpoint "
+        f"at {coco_model_path!r}.")
