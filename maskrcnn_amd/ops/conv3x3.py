@@ -23,6 +23,13 @@ def pack_weight(w: torch.Tensor) -> torch.Tensor:
         .contiguous().to(torch.bfloat16)
 
 
+def rotate_for_dgrad(w: torch.Tensor) -> torch.Tensor:
+    """Weight transform making bwd-data a FORWARD conv:
+    dgrad(x) = conv3x3(grad_out, w') with
+    w'[ci, co, r, s] = w[co, ci, 2-r, 2-s]  (s1p1, 3x3)."""
+    return w.flip(2, 3).permute(1, 0, 2, 3)
+
+
 class _Conv3x3Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, relu):
@@ -54,16 +61,33 @@ class _Conv3x3Fn(torch.autograd.Function):
             grad_out = grad_out * (ctx.save_out > 0).to(grad_out.dtype)
         grad_x = grad_w = grad_b = None
         wd = w.to(grad_out.dtype)
-        # aten.convolution_backward = the fused MIOpen backward autograd
-        # itself uses (torch.nn.grad.conv2d_weight takes a far slower
-        # unfold path).
+        go_cl = grad_out.contiguous(
+            memory_format=torch.channels_last if grad_out.is_cuda
+            else torch.contiguous_format)
+        # bwd-data of a 3x3 s1p1 conv IS a forward 3x3 conv with
+        # rotated-transposed weights — reuse the hand MFMA kernel where
+        # its policy wins (note the role swap: the dgrad's "Cin" is this
+        # conv's Cout).
+        ext = load_ext()
+        use_hip_dgrad = (
+            ctx.needs_input_grad[0] and ext is not None and
+            grad_out.is_cuda and go_cl.dtype == torch.bfloat16 and
+            go_cl.is_contiguous(memory_format=torch.channels_last) and
+            w.size(0) % 32 == 0)
+        if use_hip_dgrad:
+            from .conv_policy import use_hip_conv3x3
+            use_hip_dgrad = use_hip_conv3x3(
+                go_cl.size(0), w.size(0), go_cl.size(2), go_cl.size(3))
+        need_aten_gx = ctx.needs_input_grad[0] and not use_hip_dgrad
         gx, gw, _ = torch.ops.aten.convolution_backward(
-            grad_out.contiguous(
-                memory_format=torch.channels_last if grad_out.is_cuda
-                else torch.contiguous_format),
-            x, wd, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
-            [ctx.needs_input_grad[0], ctx.needs_input_grad[1], False])
-        if ctx.needs_input_grad[0]:
+            go_cl, x, wd, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+            [need_aten_gx, ctx.needs_input_grad[1], False])
+        if use_hip_dgrad:
+            wr = pack_weight(rotate_for_dgrad(w))
+            zb = torch.zeros(w.size(1), device=w.device,
+                             dtype=torch.float32)
+            grad_x = ext.conv3x3_fwd(go_cl, wr, zb, False).to(x.dtype)
+        elif ctx.needs_input_grad[0]:
             grad_x = gx.to(x.dtype)
         if ctx.needs_input_grad[1]:
             grad_w = gw.to(w.dtype)

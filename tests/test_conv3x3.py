@@ -168,3 +168,39 @@ def test_gpu_asymmetric_transpose_detector():
     out = conv3x3_bias_relu(x, wt, b, relu=False).float()
     assert out[0, 17, 9, 8].item() == pytest.approx(2.0, rel=0.02)
     assert out.abs().sum().item() == pytest.approx(2.0, rel=0.02)
+
+
+def test_rotate_for_dgrad_math():
+    """conv2d_input(gout, w) == conv2d(gout, rotate_for_dgrad(w)) for
+    3x3 s1p1 — the transform the HIP dgrad path relies on."""
+    from maskrcnn_amd.ops.conv3x3 import rotate_for_dgrad
+    torch.manual_seed(3)
+    x_shape = [2, 16, 9, 7]
+    w = torch.randn(8, 16, 3, 3)
+    gout = torch.randn(2, 8, 9, 7)
+    ref = torch.nn.grad.conv2d_input(x_shape, w, gout, padding=1)
+    via_fwd = F.conv2d(gout, rotate_for_dgrad(w), padding=1)
+    torch.testing.assert_close(via_fwd, ref, atol=1e-5, rtol=1e-5)
+
+
+@pytest.mark.gpu
+def test_gpu_dgrad_via_forward_kernel():
+    """The HIP dgrad path (forward kernel on rotated weights) against
+    aten's conv backward."""
+    from maskrcnn_amd.ops.conv3x3 import conv3x3_bias_relu
+    dev = torch.device("cuda:0")
+    torch.manual_seed(4)
+    x = torch.randn(2, 64, 32, 32, device=dev).bfloat16() \
+        .to(memory_format=torch.channels_last).requires_grad_(True)
+    w = (torch.randn(64, 64, 3, 3, device=dev) * 0.05).requires_grad_(True)
+    b = torch.zeros(64, device=dev, requires_grad=True)
+    out = conv3x3_bias_relu(x, w, b, relu=False)
+    g = torch.randn_like(out)
+    out.backward(g)
+    gx_hip = x.grad.clone()
+
+    x2 = x.detach().float().clone().requires_grad_(True)
+    ref = F.conv2d(x2, w.detach(), b.detach(), padding=1)
+    ref.backward(g.float())
+    torch.testing.assert_close(gx_hip.float(), x2.grad, atol=0.3,
+                               rtol=0.05)
