@@ -112,6 +112,53 @@ class Trainer:
         optimizer.step()
         return {k: float(v.detach()) for k, v in losses.items()}
 
+    @torch.no_grad()
+    def _validate(self, steps=None):
+        """Mean losses over VALIDATION_STEPS val batches (the reference's
+        fit_generator validation_data pass; keras val_* scalars)."""
+        if self.val_dataset is None:
+            return {}
+        cfg = self.config
+        steps = steps or cfg.VALIDATION_STEPS
+        loader = make_data_loader(self.val_dataset, cfg, shuffle=False,
+                                  distributed=self.distributed)
+        from ..utils.meta import parse_image_meta_t
+        totals = {}
+        n = 0
+        it = iter(loader)
+        anchors = self.wrapper.anchor_cache.get(cfg.IMAGE_SHAPE,
+                                                device=self.device)
+        was_training = self.model.training
+        self.model.train()  # training graph (targets + 5 losses)
+        for _ in range(steps):
+            try:
+                batch = next(it)
+            except StopIteration:
+                break
+            batch = {k: v.to(self.device, non_blocking=True)
+                     for k, v in batch.items()}
+            images = batch["image"]
+            if cfg.CHANNELS_LAST and self.device.type == "cuda":
+                images = images.to(memory_format=torch.channels_last)
+            meta = parse_image_meta_t(batch["image_meta"])
+            with self._autocast():
+                outputs = self.model(
+                    images, batch["image_meta"], anchors,
+                    gt_class_ids=batch["gt_class_ids"],
+                    gt_boxes=batch["gt_boxes"],
+                    gt_masks=batch["gt_masks"], mode="training")
+                losses = self.wrapper.compute_losses(
+                    {"rpn_match": batch["rpn_match"],
+                     "rpn_bbox": batch["rpn_bbox"],
+                     "active_class_ids": meta["active_class_ids"]},
+                    outputs)
+            for k, v in losses.items():
+                totals[k] = totals.get(k, 0.0) + float(v)
+            n += 1
+        if not was_training:
+            self.model.eval()
+        return {f"val_{k}": v / n for k, v in totals.items()} if n else {}
+
     def _log_scalars(self, epoch, means, seconds):
         """Per-epoch scalars: CSV always; TensorBoard when importable
         (reference used a TensorBoard callback, model.py:~2340)."""
@@ -169,6 +216,7 @@ class Trainer:
                     epoch_losses[k] = epoch_losses.get(k, 0.0) + v
                 step += 1
             means = {k: v / step for k, v in epoch_losses.items()}
+            means.update(self._validate())
             self.history.append(means)
             dt = time.time() - t0
             log.info("epoch %d: %.1fs %s", epoch, dt,

@@ -138,6 +138,11 @@ class ModelServer:
         sig = request.model_spec.signature_name or "serving_default"
         if sig != "serving_default":
             raise KeyError(f"unknown signature {sig!r}")
+        want_version = request.model_spec.version
+        if want_version is not None and want_version != self.state.version:
+            raise KeyError(
+                f"model {name!r} version {want_version} not loaded "
+                f"(serving {self.state.version})")
         try:
             images = tensor_to_numpy(request.inputs["input_image"]) \
                 .astype(np.float32)
