@@ -47,7 +47,8 @@ def build(verbose: bool = True) -> Path:
     ]
     incs = [f"-I{p}" for p in inc] + [f"-I{py_inc}", f"-I{CSRC}"]
 
-    sources = sorted(CSRC.glob("hip/*.hip")) + [CSRC / "bindings.cpp"]
+    sources = sorted(CSRC.glob("hip/*.hip")) + \
+        sorted(CSRC.glob("*.cpp"))
     objs = []
 
     def compile_one(src: Path) -> Path:

@@ -204,6 +204,8 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
 
 }  // namespace
 
+void register_rle(pybind11::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("roi_align_forward", &roi_align_forward,
         "RoIAlign / crop_and_resize forward (CDNA4)");
@@ -214,6 +216,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused 1x1 conv + bias + nearest-up2 add (CDNA4 MFMA)",
         py::arg("x"), py::arg("w"), py::arg("bias"),
         py::arg("up") = py::none());
+  register_rle(m);
   m.def("conv3x3_fwd", &conv3x3_fwd,
         "3x3 conv s1p1 + bias + optional ReLU (CDNA4 MFMA implicit GEMM)",
         py::arg("x"), py::arg("w_packed"), py::arg("bias"),

@@ -50,6 +50,22 @@ def _box_iou_matrix(dt, gt, iscrowd):
 
 
 def _mask_iou_matrix(dt, gt, iscrowd, decode_cache=None):
+    # Fast path: all-RLE inputs go through the native run-walk IoU
+    # (csrc/rle.cpp) without ever decoding to pixels.
+    if all(isinstance(m, dict) for m in dt) and \
+            all(isinstance(m, dict) for m in gt):
+        from ..ops._ext import load_ext
+        ext = load_ext()
+        if ext is not None and hasattr(ext, "rle_iou_matrix"):
+            import torch
+            dts = [torch.tensor(m["counts"], dtype=torch.long)
+                   for m in dt]
+            gts = [torch.tensor(m["counts"], dtype=torch.long)
+                   for m in gt]
+            return ext.rle_iou_matrix(
+                dts, gts,
+                torch.tensor(iscrowd, dtype=torch.bool)).numpy()
+
     def dec(m):
         if isinstance(m, np.ndarray):
             return m

@@ -80,13 +80,28 @@ def unmold_mask(mask: np.ndarray, bbox, image_shape):
 # RLE codec (COCO-compatible; replaces pycocotools maskApi.c for eval)
 # --------------------------------------------------------------------------
 
+def _rle_ext():
+    """The native C++ RLE codec (csrc/rle.cpp, pycocotools maskApi
+    parity) when the extension is built; None otherwise."""
+    from ..ops._ext import load_ext
+    ext = load_ext()
+    return ext if ext is not None and hasattr(ext, "rle_encode") else None
+
+
 def rle_encode(mask: np.ndarray) -> dict:
     """Encode a binary [H, W] mask as COCO uncompressed RLE.
 
     Column-major (Fortran) order, counts alternate 0s/1s starting with 0s —
-    the pycocotools convention (reference N12).
+    the pycocotools convention (reference N12). Uses the native C++ codec
+    when built.
     """
     h, w = mask.shape
+    ext = _rle_ext()
+    if ext is not None:
+        import torch
+        counts = ext.rle_encode(
+            torch.from_numpy(np.ascontiguousarray(mask))).tolist()
+        return {"size": [h, w], "counts": counts}
     flat = np.asfortranarray(mask).flatten(order="F").astype(np.uint8)
     diffs = np.flatnonzero(flat[1:] != flat[:-1]) + 1
     runs = np.diff(np.concatenate([[0], diffs, [flat.size]]))
@@ -100,6 +115,11 @@ def rle_decode(rle: dict) -> np.ndarray:
     """Decode COCO uncompressed RLE to a binary [H, W] mask."""
     h, w = rle["size"]
     counts = rle["counts"]
+    ext = _rle_ext()
+    if ext is not None:
+        import torch
+        return ext.rle_decode(torch.tensor(counts, dtype=torch.long),
+                              h, w).numpy()
     flat = np.zeros(h * w, dtype=bool)
     pos = 0
     val = False
