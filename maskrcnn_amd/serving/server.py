@@ -46,75 +46,101 @@ except ImportError:  # pragma: no cover
 
 
 class ModelState:
-    """One served model: engine + batcher + version, hot-swappable."""
+    """One served model: engine(s) + batcher(s) + version, hot-swappable.
+
+    devices=["cuda:0", ..., "cuda:7"] serves one REPLICA per GPU behind
+    one front door (BASELINE config #4's 8-replica mode; serving DP needs
+    no collectives — SURVEY §2.4). Requests round-robin to the replica
+    with the shortest queue.
+    """
 
     def __init__(self, name: str, base_path: str, device=None,
-                 max_batch_size: int = 32, batch_timeout_micros: int = 2000,
-                 config=None):
+                 devices=None, max_batch_size: int = 32,
+                 batch_timeout_micros: int = 2000, config=None):
         self.name = name
         self.base_path = base_path
-        self.device = device
+        self.devices = list(devices) if devices else [device]
         self.max_batch_size = max_batch_size
         self.batch_timeout_micros = batch_timeout_micros
         self.config_override = config
         self.version: Optional[int] = None
-        self.engine: Optional[InferenceEngine] = None
-        self.batcher: Optional[DynamicBatcher] = None
+        self.engines: list = []
+        self.batchers: list = []
+        self._rr = 0
         self._lock = threading.Lock()
+
+    # Single-replica compat accessors.
+    @property
+    def engine(self):
+        return self.engines[0] if self.engines else None
+
+    @property
+    def batcher(self):
+        return self.batchers[0] if self.batchers else None
 
     def maybe_reload(self) -> bool:
         latest = find_latest_version(self.base_path)
         if latest is None or latest == self.version:
             return False
         export_dir = os.path.join(self.base_path, str(latest))
-        log.info("loading %s version %d from %s", self.name, latest,
-                 export_dir)
-        engine = InferenceEngine.from_saved_model(
-            export_dir, device=self.device, config=self.config_override)
-        batcher = DynamicBatcher(
-            engine, max_batch_size=self.max_batch_size,
+        log.info("loading %s version %d from %s on %s", self.name,
+                 latest, export_dir, self.devices)
+        engines = [InferenceEngine.from_saved_model(
+            export_dir, device=dev, config=self.config_override)
+            for dev in self.devices]
+        batchers = [DynamicBatcher(
+            eng, max_batch_size=self.max_batch_size,
             batch_timeout_micros=self.batch_timeout_micros)
+            for eng in engines]
         with self._lock:
-            old = self.batcher
-            self.engine, self.batcher, self.version = \
-                engine, batcher, latest
-        if old is not None:
-            old.close()
+            old = self.batchers
+            self.engines, self.batchers, self.version = \
+                engines, batchers, latest
+        for b in old:
+            b.close()
         if _PROM:
             VERSION.set(latest)
         return True
 
     def predict(self, images, metas, anchors=None, outputs=None):
         with self._lock:
-            batcher = self.batcher
+            if not self.batchers:
+                raise RuntimeError(
+                    f"model {self.name} has no loaded version")
+            # Shortest-queue replica (round-robin tiebreak).
+            qsizes = [b.queue.qsize() for b in self.batchers]
+            best = min(range(len(qsizes)),
+                       key=lambda i: (qsizes[i], (i - self._rr) %
+                                      len(qsizes)))
+            self._rr = (best + 1) % len(self.batchers)
+            batcher = self.batchers[best]
             version = self.version
-        if batcher is None:
-            raise RuntimeError(f"model {self.name} has no loaded version")
         fut = batcher.submit(images, metas, anchors, outputs=outputs)
         return fut.result(timeout=300.0), version
 
     def warmup(self, image_shape=None):
-        """Run one dummy max-batch predict so MIOpen Find / algo caches are
-        hot before traffic (TF-Serving model_warmup equivalent)."""
-        eng = self.engine
-        if eng is None:
-            return
-        H, W, _ = image_shape or eng.config.IMAGE_SHAPE
-        B = self.max_batch_size
-        images = np.zeros((B, H, W, 3), np.float32)
-        metas = np.zeros((B, eng.config.IMAGE_META_SIZE), np.float32)
-        metas[:, 4:7] = (H, W, 3)
-        metas[:, 7:11] = (0, 0, H, W)
-        eng.predict(images, metas, None)
+        """Run one dummy max-batch predict per replica so MIOpen Find /
+        algo caches are hot before traffic (TF-Serving model_warmup
+        equivalent)."""
+        for eng in self.engines:
+            H, W, _ = image_shape or eng.config.IMAGE_SHAPE
+            B = self.max_batch_size
+            images = np.zeros((B, H, W, 3), np.float32)
+            metas = np.zeros((B, eng.config.IMAGE_META_SIZE), np.float32)
+            metas[:, 4:7] = (H, W, 3)
+            metas[:, 7:11] = (0, 0, H, W)
+            eng.predict(images, metas, None)
 
 
 class ModelServer:
     def __init__(self, model_name: str, model_base_path: str,
                  grpc_port: int = 8500, rest_port: int = 8501,
                  metrics_port: Optional[int] = None, device=None,
-                 max_batch_size: int = 32, batch_timeout_micros: int = 2000,
+                 devices=None, max_batch_size: int = 32,
+                 batch_timeout_micros: int = 2000,
                  poll_interval_s: float = 5.0, config=None):
         self.state = ModelState(model_name, model_base_path, device=device,
+                                devices=devices,
                                 max_batch_size=max_batch_size,
                                 batch_timeout_micros=batch_timeout_micros,
                                 config=config)
@@ -338,8 +364,8 @@ class ModelServer:
             self._grpc_server.stop(grace).wait()
         if self._uvicorn_server is not None:
             self._uvicorn_server.should_exit = True
-        if self.state.batcher is not None:
-            self.state.batcher.close()
+        for b in self.state.batchers:
+            b.close()
 
     def wait(self):
         self._grpc_server.wait_for_termination()
@@ -354,13 +380,19 @@ def main():  # pragma: no cover
     ap.add_argument("--rest_api_port", type=int, default=8501)
     ap.add_argument("--metrics_port", type=int, default=None)
     ap.add_argument("--enable_batching", action="store_true")
+    ap.add_argument("--num_replicas", type=int, default=1,
+                    help="one engine per GPU: cuda:0..cuda:N-1")
     ap.add_argument("--max_batch_size", type=int, default=32)
     ap.add_argument("--batch_timeout_micros", type=int, default=2000)
     args = ap.parse_args()
     logging.basicConfig(level=logging.INFO)
+    devices = None
+    if args.num_replicas > 1:
+        devices = [f"cuda:{i}" for i in range(args.num_replicas)]
     server = ModelServer(
         args.model_name, args.model_base_path, grpc_port=args.port,
         rest_port=args.rest_api_port, metrics_port=args.metrics_port,
+        devices=devices,
         max_batch_size=args.max_batch_size if args.enable_batching else 1,
         batch_timeout_micros=args.batch_timeout_micros)
     server.start()

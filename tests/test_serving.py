@@ -300,3 +300,32 @@ def test_version_hot_reload(served_model, tmp_path):
     while time.time() < deadline and server.state.version != 2:
         time.sleep(0.2)
     assert server.state.version == 2
+
+
+def test_multi_replica_state(tmp_path, shapes_config):
+    """devices=[cpu, cpu] loads two replicas; requests spread across
+    batchers (8-replica serving mode, scaled down for CPU)."""
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+    from maskrcnn_amd.serving.server import ModelState
+
+    m = MaskRCNN("inference", shapes_config, model_dir=str(tmp_path))
+    base = str(tmp_path / "mr" / "mask_rcnn")
+    export_saved_model(m, shapes_config, base, version=1)
+    state = ModelState("mask_rcnn", base, devices=["cpu", "cpu"],
+                       max_batch_size=1, batch_timeout_micros=500,
+                       config=shapes_config)
+    state.maybe_reload()
+    assert len(state.engines) == 2 and len(state.batchers) == 2
+    H = shapes_config.IMAGE_SHAPE[0]
+    images = np.zeros((1, H, H, 3), np.float32)
+    metas = np.zeros((1, shapes_config.IMAGE_META_SIZE), np.float32)
+    metas[:, 4:7] = (H, H, 3)
+    metas[:, 7:11] = (0, 0, H, H)
+    out1, v1 = state.predict(images, metas)
+    out2, v2 = state.predict(images, metas)
+    assert v1 == v2 == 1
+    np.testing.assert_allclose(out1["mrcnn_detection"],
+                               out2["mrcnn_detection"], atol=1e-5)
+    for b in state.batchers:
+        b.close()
