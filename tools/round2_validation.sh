@@ -57,3 +57,13 @@ block5)  # PMC counters for the hand kernels (own run, never with traces)
   ls gpurun_out/r2_pmc
   ;;
 esac
+# (appended) block6: fp8-weight large-image serving (BASELINE config #5,
+# 1333 -> padded 1344; DP=8 scale-out is the driver's 8-GPU tier — this
+# measures the per-GPU replica)
+if [ "${1:-}" = "block6" ]; then
+  timeout 1100 python bench/serving_bench.py --requests 48 \
+      --client-procs 4 --client-threads 8 --warmup 8 \
+      --image-size 1344 --weight-dtype fp8 --max-batch 16 \
+      > gpurun_out/r2_fp8_serving.json 2> gpurun_out/r2_fp8_serving.err
+  cat gpurun_out/r2_fp8_serving.json
+fi
