@@ -23,9 +23,24 @@ class DetectionTargetLayer(nn.Module):
     def __init__(self, config):
         super().__init__()
         self.config = config
+        # Opt-in fully-tensorized sampler: identical selection
+        # DISTRIBUTION with fixed shapes and zero host syncs (the loop
+        # path's nonzero/randperm cost ~15 syncs/step at B=2). Pending
+        # the round-2 perf A/B (tools/round2_validation.sh).
+        import os
+        self.vectorized = bool(
+            os.environ.get("MASKRCNN_VECTORIZED_TARGETS"))
 
     @torch.no_grad()
     def forward(self, proposals, gt_class_ids, gt_boxes, gt_masks):
+        if self.vectorized:
+            return self._forward_vectorized(proposals, gt_class_ids,
+                                            gt_boxes, gt_masks)
+        return self._forward_loop(proposals, gt_class_ids, gt_boxes,
+                                  gt_masks)
+
+    # ------------------------------------------------------ loop variant
+    def _forward_loop(self, proposals, gt_class_ids, gt_boxes, gt_masks):
         """
         proposals:    [B, N, 4] normalized, zero-padded
         gt_class_ids: [B, MAX_GT] (zero-padded; negative = crowd)
@@ -54,6 +69,108 @@ class DetectionTargetLayer(nn.Module):
                 out_cls[b, :n] = cls
                 out_deltas[b, :n] = deltas
                 out_masks[b, :n] = masks
+        return out_rois, out_cls, out_deltas, out_masks
+
+    # ------------------------------------------- vectorized variant
+    @torch.no_grad()
+    def _forward_vectorized(self, proposals, gt_class_ids, gt_boxes,
+                            gt_masks):
+        """Same sampling distribution as the loop variant with FIXED
+        shapes and no host syncs: random subsets are drawn with the
+        rand*mask top-k reservoir trick; counts/quotas stay tensors.
+
+        Layout difference (semantics-neutral): positives occupy slots
+        [0, cap_pos) with invalid slots zeroed, negatives
+        [cap_pos, T) — every consumer (losses, heads) is
+        padding-agnostic via target_class_ids > 0.
+        """
+        cfg = self.config
+        B, P, _ = proposals.shape
+        T = cfg.TRAIN_ROIS_PER_IMAGE
+        cap_pos = int(T * cfg.ROI_POSITIVE_RATIO)
+        neg_slots = T - cap_pos
+        mh, mw = cfg.MASK_SHAPE
+        device = proposals.device
+        std = torch.tensor(cfg.BBOX_STD_DEV, device=device)
+        r = 1.0 / cfg.ROI_POSITIVE_RATIO
+
+        out_rois = proposals.new_zeros((B, T, 4))
+        out_cls = torch.zeros((B, T), dtype=torch.long, device=device)
+        out_deltas = proposals.new_zeros((B, T, 4))
+        out_masks = proposals.new_zeros((B, T, mh, mw))
+
+        for b in range(B):
+            prop = proposals[b].float()                       # [P, 4]
+            ids = gt_class_ids[b].long()                      # [G]
+            boxes = gt_boxes[b].float()                       # [G, 4]
+            valid_prop = prop.abs().sum(dim=1) > 0
+            non_crowd = (ids > 0).float()
+            crowd = (ids < 0).float()
+
+            ov = overlaps_t(prop, boxes)                      # [P, G]
+            ov_nc = ov * non_crowd.unsqueeze(0)
+            roi_iou_max = ov_nc.max(dim=1).values
+            crowd_iou = (ov * crowd.unsqueeze(0)).max(dim=1).values
+            pos_mask = (roi_iou_max >= 0.5) & valid_prop
+            neg_mask = (roi_iou_max < 0.5) & (crowd_iou < 0.001) & \
+                valid_prop
+
+            # Uniform random subset via rand*mask top-k.
+            pos_idx = (torch.rand(P, device=device) *
+                       pos_mask).topk(cap_pos).indices
+            pos_ok = pos_mask[pos_idx]                        # [cap_pos]
+            n_pos = pos_ok.sum()
+            quota = (n_pos.double() * (r - 1.0)).floor()
+            neg_idx = (torch.rand(P, device=device) *
+                       neg_mask).topk(neg_slots).indices
+            neg_ok = neg_mask[neg_idx] & \
+                (torch.arange(neg_slots, device=device) < quota)
+
+            pos_rois = prop[pos_idx] * pos_ok.unsqueeze(1)
+            assign = ov_nc[pos_idx].argmax(dim=1)             # [cap_pos]
+            roi_gt_boxes = boxes[assign]
+            roi_cls = ids[assign].clamp(min=0) * pos_ok.long()
+
+            # Safe refinement (invalid rows later zeroed; clamp avoids
+            # inf*0 = nan from zero-area padding rows).
+            def safe_refine(box, gt):
+                h = (box[:, 2] - box[:, 0]).clamp(min=1e-6)
+                w = (box[:, 3] - box[:, 1]).clamp(min=1e-6)
+                cy = box[:, 0] + 0.5 * h
+                cx = box[:, 1] + 0.5 * w
+                gh = (gt[:, 2] - gt[:, 0]).clamp(min=1e-6)
+                gw = (gt[:, 3] - gt[:, 1]).clamp(min=1e-6)
+                gcy = gt[:, 0] + 0.5 * gh
+                gcx = gt[:, 1] + 0.5 * gw
+                return torch.stack([(gcy - cy) / h, (gcx - cx) / w,
+                                    torch.log(gh / h),
+                                    torch.log(gw / w)], dim=1)
+
+            deltas = safe_refine(prop[pos_idx], roi_gt_boxes) / std
+            deltas = deltas * pos_ok.unsqueeze(1)
+
+            masks_f = gt_masks[b].permute(2, 0, 1).float() \
+                .unsqueeze(1)[assign]                # [cap_pos, 1, H, W]
+            crop_boxes = prop[pos_idx]
+            if cfg.USE_MINI_MASK:
+                y1, x1, y2, x2 = crop_boxes.unbind(dim=1)
+                gy1, gx1, gy2, gx2 = roi_gt_boxes.unbind(dim=1)
+                gh = (gy2 - gy1).clamp(min=1e-6)
+                gw = (gx2 - gx1).clamp(min=1e-6)
+                crop_boxes = torch.stack(
+                    [(y1 - gy1) / gh, (x1 - gx1) / gw,
+                     (y2 - gy1) / gh, (x2 - gx1) / gw], dim=1)
+            box_ind = torch.arange(cap_pos, device=device)
+            mask_t = crop_and_resize(masks_f, crop_boxes, box_ind,
+                                     (mh, mw)).squeeze(1)
+            mask_t = torch.round(mask_t) * pos_ok.unsqueeze(1) \
+                .unsqueeze(2)
+
+            out_rois[b, :cap_pos] = pos_rois
+            out_rois[b, cap_pos:] = prop[neg_idx] * neg_ok.unsqueeze(1)
+            out_cls[b, :cap_pos] = roi_cls
+            out_deltas[b, :cap_pos] = deltas
+            out_masks[b, :cap_pos] = mask_t
         return out_rois, out_cls, out_deltas, out_masks
 
     def _one_image(self, proposals, gt_class_ids, gt_boxes, gt_masks):
