@@ -25,6 +25,7 @@ block2)  # training-path A/Bs (one box; ~12 min incl. one find phase)
   echo "=== cl roi";           MASKRCNN_TRAIN_CL_ROI=1 run clroi
   echo "=== all three";        MASKRCNN_TRAIN_FUSED_FPN=1 \
       MASKRCNN_TRAIN_HIP_CONV=1 MASKRCNN_TRAIN_CL_ROI=1 run all
+  echo "=== vectorized targets"; MASKRCNN_VECTORIZED_TARGETS=1 run vtgt
   ;;
 block3)  # serving bench (~8 min incl. 265 s batch-32 find)
   timeout 900 python bench/serving_bench.py --requests 96 \
