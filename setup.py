@@ -1,0 +1,29 @@
+"""setup.py — `python setup.py build_ext --inplace` builds the in-tree
+gfx950 HIP extension (delegates to build_ext.py, which drives hipcc).
+
+The extension intentionally lives IN-TREE (maskrcnn_amd/_C.so), not in
+site-packages: the repo snapshot shipped to GPU boxes must carry it.
+"""
+
+from __future__ import annotations
+
+from setuptools import find_packages, setup
+from setuptools.command.build_ext import build_ext as _build_ext
+
+
+class HipBuildExt(_build_ext):
+    def run(self):
+        import build_ext as bx
+        bx.build(verbose=True)
+
+
+setup(
+    name="maskrcnn_amd",
+    version="0.1.0",
+    description=("MI355X-native Mask R-CNN training & serving "
+                 "(PyTorch-ROCm + CDNA4 HIP kernels + RCCL)"),
+    packages=find_packages(include=["maskrcnn_amd", "maskrcnn_amd.*"]),
+    package_data={"maskrcnn_amd": ["_C.so"]},
+    cmdclass={"build_ext": HipBuildExt},
+    python_requires=">=3.10",
+)
