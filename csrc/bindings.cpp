@@ -8,6 +8,7 @@
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
+#include <tuple>
 #include <unordered_map>
 
 extern "C" {
@@ -123,6 +124,40 @@ torch::Tensor nms(torch::Tensor boxes, torch::Tensor scores,
   return order.index_select(0, keep.slice(0, 0, n).to(torch::kLong));
 }
 
+std::tuple<torch::Tensor, torch::Tensor> nms_padded(torch::Tensor boxes,
+                                                    torch::Tensor scores,
+                                                    double iou_thresh,
+                                                    int64_t max_out) {
+  // Fixed-shape NMS: keep indices (ORIGINAL order) padded with -1 to
+  // max_out + the kept count as a DEVICE tensor — zero host syncs, so
+  // callers can stay capture-safe (SURVEY §7 hard-part 3).
+  TORCH_CHECK(boxes.is_cuda() && scores.is_cuda(),
+              "nms_padded: GPU tensors only");
+  const int N = boxes.size(0);
+  auto count = torch::zeros({1}, boxes.options().dtype(torch::kInt));
+  if (N == 0) {
+    auto empty = torch::full({max_out}, -1,
+                             boxes.options().dtype(torch::kLong));
+    return {empty, count};
+  }
+  auto order = scores.argsort(0, /*descending=*/true);
+  auto sorted = boxes.index_select(0, order).contiguous().to(torch::kFloat);
+  auto ws = torch::empty({nms_workspace_bytes(N)},
+                         boxes.options().dtype(torch::kByte));
+  auto keep = torch::zeros({std::max<int64_t>(max_out, 1)},
+                           boxes.options().dtype(torch::kInt));
+  nms_launch(sorted.data_ptr<float>(), N, (float)iou_thresh, (int)max_out,
+             (void*)ws.data_ptr(), keep.data_ptr<int>(),
+             count.data_ptr<int>(), cur_stream());
+  auto valid = torch::arange(max_out, boxes.options().dtype(torch::kLong))
+                   .lt(count.to(torch::kLong));
+  auto orig = order.index_select(
+      0, keep.slice(0, 0, max_out).to(torch::kLong));
+  auto padded = torch::where(
+      valid, orig, torch::full_like(orig, -1));
+  return {padded, count};
+}
+
 torch::Tensor fpn_fuse(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                        c10::optional<torch::Tensor> up) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4, "x must be [B,C,H,W] cuda");
@@ -212,6 +247,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("roi_align_backward", &roi_align_backward,
         "RoIAlign backward (CDNA4, atomic scatter)");
   m.def("nms", &nms, "bitmask greedy NMS (CDNA4)");
+  m.def("nms_padded", &nms_padded,
+        "sync-free NMS: (-1-padded keep [max_out], device count)",
+        py::arg("boxes"), py::arg("scores"), py::arg("iou_thresh"),
+        py::arg("max_out"));
   m.def("fpn_fuse", &fpn_fuse,
         "fused 1x1 conv + bias + nearest-up2 add (CDNA4 MFMA)",
         py::arg("x"), py::arg("w"), py::arg("bias"),

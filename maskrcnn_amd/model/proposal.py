@@ -14,7 +14,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ..ops import batched_nms
+from ..ops import batched_nms_padded
 from ..utils.boxes import apply_box_deltas_t, clip_boxes_t
 
 
@@ -54,31 +54,43 @@ class ProposalLayer(nn.Module):
         boxes = clip_boxes_t(boxes, window)
 
         # NMS for a GROUP of images per launch (image-index offset
-        # trick): per-image NMS costs one host sync each (serializes at
-        # serving batch 32), but one launch over ALL images makes the
-        # bitmask workspace quadratic in B*P (4.6 GB at 32x6000). Chunk
-        # so each launch stays <= ~16k boxes.
+        # trick): per-image NMS serializes at serving batch 32, but one
+        # launch over ALL images makes the bitmask workspace quadratic in
+        # boxes/launch (4.6 GB at 32x6000) — chunk to <= ~16k boxes.
+        # Padded NMS + masked scatter keeps the path free of host syncs
+        # (capture-safe; SURVEY §7 hard-part 3).
         P = boxes.shape[1]
         group = max(1, 16384 // P)
-        rois = boxes.new_zeros((B, proposal_count, 4))
+        device = boxes.device
+        # Dummy row/col absorbs invalid scatter targets.
+        rois_buf = boxes.new_zeros((B + 1, proposal_count + 1, 4))
         for g0 in range(0, B, group):
             g1 = min(g0 + group, B)
             nb = g1 - g0
             flat_boxes = boxes[g0:g1].reshape(-1, 4)
             flat_scores = top_scores[g0:g1].reshape(-1)
-            img_ix = torch.arange(nb, device=boxes.device) \
+            img_ix = torch.arange(nb, device=device) \
                 .repeat_interleave(P)
-            keep = batched_nms(flat_boxes, flat_scores, img_ix,
-                               cfg.RPN_NMS_THRESHOLD)
-            if not keep.numel():
-                continue
-            img = img_ix[keep]                         # [K], score-desc
+            # Full-size keep: a global-score cap could starve one image
+            # of its proposal_count while another image's over-quota
+            # survivors (rank >= proposal_count, discarded) fill it.
+            max_keep = nb * P
+            keep, _count = batched_nms_padded(
+                flat_boxes, flat_scores, img_ix, cfg.RPN_NMS_THRESHOLD,
+                max_keep)
+            valid = keep >= 0                          # [K]
+            keep_c = keep.clamp(min=0)
+            img = img_ix[keep_c]                       # [K], score-desc
             # Per-image rank of each kept box (global order preserves
-            # per-image score order).
+            # per-image score order); invalid rows rank out of range.
             onehot = (img.unsqueeze(1) ==
-                      torch.arange(nb, device=img.device).unsqueeze(0))
+                      torch.arange(nb, device=device).unsqueeze(0)) \
+                & valid.unsqueeze(1)
             rank = onehot.long().cumsum(dim=0)[
-                torch.arange(img.numel(), device=img.device), img] - 1
-            sel = rank < proposal_count
-            rois[g0 + img[sel], rank[sel]] = flat_boxes[keep[sel]]
-        return rois
+                torch.arange(img.numel(), device=device), img] - 1
+            sel = valid & (rank < proposal_count)
+            tgt_img = torch.where(sel, g0 + img, torch.full_like(img, B))
+            tgt_rank = torch.where(
+                sel, rank, torch.full_like(rank, proposal_count))
+            rois_buf.index_put_((tgt_img, tgt_rank), flat_boxes[keep_c])
+        return rois_buf[:B, :proposal_count]

@@ -62,6 +62,42 @@ def nms(boxes: torch.Tensor, scores: torch.Tensor, iou_threshold: float,
     return keep
 
 
+def nms_padded(boxes: torch.Tensor, scores: torch.Tensor,
+               iou_threshold: float, max_out: int):
+    """Fixed-shape NMS with ZERO host syncs.
+
+    Returns (keep [max_out] int64, count [1] int32-on-device): keep holds
+    original indices in score-descending order, -1-padded. The GPU path
+    never touches the host (capture-safe); CPU falls back to greedy NMS
+    + padding.
+    """
+    if boxes.is_cuda:
+        ext = require_ext()
+        return ext.nms_padded(boxes.contiguous().float(),
+                              scores.contiguous().float(),
+                              float(iou_threshold), int(max_out))
+    keep = _nms_torch(boxes.float(), scores.float(), iou_threshold)
+    keep = keep[:max_out]
+    out = torch.full((max_out,), -1, dtype=torch.long)
+    out[:keep.shape[0]] = keep
+    count = torch.tensor([keep.shape[0]], dtype=torch.int32)
+    return out, count
+
+
+def batched_nms_padded(boxes: torch.Tensor, scores: torch.Tensor,
+                       idxs: torch.Tensor, iou_threshold: float,
+                       max_out: int):
+    """Sync-free grouped NMS (coordinate-offset trick + nms_padded)."""
+    if boxes.numel() == 0:
+        return (torch.full((max_out,), -1, dtype=torch.long,
+                           device=boxes.device),
+                torch.zeros(1, dtype=torch.int32, device=boxes.device))
+    max_coord = boxes.max()
+    offsets = idxs.to(boxes) * (max_coord + 1)
+    return nms_padded(boxes + offsets[:, None], scores, iou_threshold,
+                      max_out)
+
+
 def batched_nms(boxes: torch.Tensor, scores: torch.Tensor,
                 idxs: torch.Tensor, iou_threshold: float) -> torch.Tensor:
     """NMS over multiple groups (e.g. classes) in one call.

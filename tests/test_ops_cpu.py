@@ -114,3 +114,18 @@ def test_batched_nms_classes_dont_suppress_each_other():
     assert len(keep) == 2
     keep_same = batched_nms(boxes, scores, torch.tensor([1, 1]), 0.5)
     assert len(keep_same) == 1
+
+
+def test_nms_padded_matches_nms():
+    from maskrcnn_amd.ops import nms, nms_padded
+    rng = np.random.RandomState(11)
+    boxes = rng.rand(80, 4).astype(np.float32)
+    boxes[:, 2:] = boxes[:, :2] + np.abs(boxes[:, 2:]) + 0.05
+    scores = rng.rand(80).astype(np.float32)
+    ref = nms(torch.from_numpy(boxes), torch.from_numpy(scores), 0.5)
+    keep, count = nms_padded(torch.from_numpy(boxes),
+                             torch.from_numpy(scores), 0.5, 100)
+    n = int(count.item())
+    assert n == ref.shape[0]
+    np.testing.assert_array_equal(keep[:n].numpy(), ref.numpy())
+    assert (keep[n:] == -1).all()

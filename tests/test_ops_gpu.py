@@ -163,3 +163,15 @@ def test_model_train_step_gpu(dev):
     torch.cuda.synchronize()
     for k, v in losses.items():
         assert torch.isfinite(v), (k, v)
+
+
+def test_nms_padded_gpu_matches_cpu(dev):
+    _require_ext()
+    from maskrcnn_amd.ops import nms_padded
+    rng = np.random.RandomState(12)
+    boxes = torch.from_numpy(_rand_boxes(700, rng))
+    scores = torch.from_numpy(rng.rand(700).astype(np.float32))
+    keep_c, count_c = nms_padded(boxes, scores, 0.6, 700)
+    keep_g, count_g = nms_padded(boxes.to(dev), scores.to(dev), 0.6, 700)
+    assert int(count_g.item()) == int(count_c.item())
+    np.testing.assert_array_equal(keep_g.cpu().numpy(), keep_c.numpy())
