@@ -323,9 +323,10 @@ class MaskRCNN:
         """Normalized anchors (cached per shape); numpy for serving parity."""
         return self.anchor_cache.get(image_shape).numpy()
 
-    def unmold_detections(self, detections, mrcnn_mask, original_image_shape,
+    @staticmethod
+    def unmold_detections(detections, mrcnn_mask, original_image_shape,
                           image_shape, window):
-        """Reference unmold_detections:~2500.
+        """Reference unmold_detections:~2500 (static: pure numpy).
 
         detections: [D, 6]; mrcnn_mask: [D, C, mh, mw] (torch order);
         Returns (boxes px [N,4], class_ids [N], scores [N], masks [H,W,N]).

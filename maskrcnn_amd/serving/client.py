@@ -58,11 +58,10 @@ class ServingClientBase:
         detections = outputs["mrcnn_detection"]
         masks_tf = outputs["mrcnn_mask"]  # [B, D, mh, mw, C]
         masks = np.transpose(masks_tf, (0, 1, 4, 2, 3))  # torch layout
-        unmolder = MaskRCNN.__new__(MaskRCNN)  # reuse unmold_detections
         for i, image in enumerate(images):
             boxes, class_ids, scores, full_masks = \
                 MaskRCNN.unmold_detections(
-                    unmolder, detections[i], masks[i], image.shape,
+                    detections[i], masks[i], image.shape,
                     molded_shape, windows[i])
             results.append({"rois": boxes, "class_ids": class_ids,
                             "scores": scores, "masks": full_masks})
