@@ -38,17 +38,36 @@ class FPNClassifierHead(nn.Module):
         self.num_classes = nc
 
     def forward(self, rois, feature_maps):
-        """rois [B,N,4] -> (logits [B,N,C], probs [B,N,C], bbox [B,N,C,4])."""
+        """rois [B,N,4] -> (logits [B,N,C], probs [B,N,C], bbox [B,N,C,4]).
+
+        The 7x7-VALID conv1 and 1x1 conv2 are mathematically dense
+        layers over the pooled tile — evaluated as F.linear (hipBLASLt
+        GEMMs on ROCm) rather than MIOpen convs; parameters stay the
+        Conv2d modules (state-dict parity with the reference naming).
+        """
         B, N, _ = rois.shape
         x = self.roi_align(rois, feature_maps)           # [B*N,256,7,7]
-        x = F.relu(self.mrcnn_class_bn1(self.mrcnn_class_conv1(x)))
-        x = F.relu(self.mrcnn_class_bn2(self.mrcnn_class_conv2(x)))
-        x = x.flatten(1)                                 # [B*N, fc]
+        w1 = self.mrcnn_class_conv1.weight
+        x = F.linear(
+            x.reshape(x.shape[0], -1),
+            w1.reshape(w1.shape[0], -1).to(x.dtype),
+            self.mrcnn_class_conv1.bias.to(x.dtype))     # [B*N, fc]
+        x = F.relu(self._bn_vec(self.mrcnn_class_bn1, x))
+        w2 = self.mrcnn_class_conv2.weight
+        x = F.linear(x, w2.reshape(w2.shape[0], -1).to(x.dtype),
+                     self.mrcnn_class_conv2.bias.to(x.dtype))
+        x = F.relu(self._bn_vec(self.mrcnn_class_bn2, x))
         logits = self.mrcnn_class_logits(x)
         probs = F.softmax(logits, dim=-1)
         bbox = self.mrcnn_bbox_fc(x).reshape(B * N, self.num_classes, 4)
         return (logits.reshape(B, N, -1), probs.reshape(B, N, -1),
                 bbox.reshape(B, N, self.num_classes, 4))
+
+    @staticmethod
+    def _bn_vec(bn, x):
+        """Apply a (Frozen)BatchNorm2d's affine/stats to a [M, C] vector
+        (the 4D module applied at spatial size 1x1)."""
+        return bn(x.unsqueeze(-1).unsqueeze(-1)).squeeze(-1).squeeze(-1)
 
 
 class FPNMaskHead(nn.Module):

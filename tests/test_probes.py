@@ -116,3 +116,28 @@ def test_pyramid_roi_align_level_routing(shapes_config):
     assert out.shape == (2, C, 7, 7)
     assert torch.allclose(out[0], torch.ones_like(out[0]))
     assert torch.allclose(out[1], 4 * torch.ones_like(out[1]))
+
+
+def test_classifier_head_linear_equals_conv(shapes_config):
+    """The head's F.linear evaluation must equal the reference conv
+    formulation (7x7 VALID conv == dense over the pooled tile)."""
+    import torch.nn.functional as F
+    from maskrcnn_amd.model.heads import FPNClassifierHead
+    torch.manual_seed(0)
+    head = FPNClassifierHead(shapes_config)
+    x = torch.randn(6, 256, shapes_config.POOL_SIZE,
+                    shapes_config.POOL_SIZE)
+    # Reference formulation.
+    ref = F.relu(head.mrcnn_class_bn1(head.mrcnn_class_conv1(x)))
+    ref = F.relu(head.mrcnn_class_bn2(head.mrcnn_class_conv2(ref)))
+    ref = ref.flatten(1)
+    # Linear formulation (what forward uses).
+    w1 = head.mrcnn_class_conv1.weight
+    y = F.linear(x.reshape(6, -1), w1.reshape(w1.shape[0], -1),
+                 head.mrcnn_class_conv1.bias)
+    y = F.relu(head._bn_vec(head.mrcnn_class_bn1, y))
+    w2 = head.mrcnn_class_conv2.weight
+    y = F.linear(y, w2.reshape(w2.shape[0], -1),
+                 head.mrcnn_class_conv2.bias)
+    y = F.relu(head._bn_vec(head.mrcnn_class_bn2, y))
+    torch.testing.assert_close(y, ref, atol=1e-4, rtol=1e-4)
