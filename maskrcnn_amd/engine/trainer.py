@@ -182,7 +182,11 @@ class Trainer:
         except ImportError:
             pass
 
-    def fit(self, learning_rate, epochs, layers="all", steps_per_epoch=None):
+    def fit(self, learning_rate, epochs, layers="all",
+            steps_per_epoch=None, custom_callbacks=None):
+        """custom_callbacks: list of callables(epoch, scalars_dict) run
+        per epoch on rank 0 (the reference's Keras custom_callbacks
+        parameter, model.py:~2310)."""
         cfg = self.config
         self.wrapper.set_trainable(layers)
         if self.distributed:
@@ -222,6 +226,9 @@ class Trainer:
             log.info("epoch %d: %.1fs %s", epoch, dt,
                      {k: round(v, 4) for k, v in means.items()})
             self._log_scalars(epoch, means, dt)
+            if custom_callbacks and self.rank == 0:
+                for cb in custom_callbacks:
+                    cb(epoch, dict(means))
             if self.rank == 0:
                 ckpt = self.wrapper.checkpoint_path.format(epoch=epoch)
                 self.wrapper.save_weights(ckpt)

@@ -469,4 +469,5 @@ class MaskRCNN:
         from ..engine.trainer import Trainer
         trainer = Trainer(self, train_dataset, val_dataset,
                           augmentation=augmentation)
-        trainer.fit(learning_rate, epochs, layers)
+        trainer.fit(learning_rate, epochs, layers,
+                    custom_callbacks=custom_callbacks)

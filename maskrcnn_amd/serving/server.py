@@ -257,6 +257,28 @@ class ModelServer:
         app = FastAPI(title="maskrcnn_amd model server")
         outer = self
 
+        @app.get("/v1/models/{model_name}/metadata")
+        def model_metadata(model_name: str):
+            if model_name != outer.state.name or \
+                    outer.state.engine is None:
+                return JSONResponse({"error": "model not found"},
+                                    status_code=404)
+            sig = build_signature(outer.state.engine.config)
+            def ti(t):
+                return {"dtype": "DT_FLOAT",
+                        "tensor_shape": {"dim": [{"size": str(d)}
+                                                 for d in t.shape]},
+                        "name": t.name}
+            return {"model_spec": {"name": outer.state.name,
+                                   "version": str(outer.state.version)},
+                    "metadata": {"signature_def": {"signature_def": {
+                        "serving_default": {
+                            "inputs": {k: ti(v)
+                                       for k, v in sig.inputs.items()},
+                            "outputs": {k: ti(v)
+                                        for k, v in sig.outputs.items()},
+                            "method_name": sig.method_name}}}}}
+
         @app.get("/v1/models/{model_name}")
         def model_status(model_name: str):
             ok = (model_name == outer.state.name and

@@ -329,3 +329,55 @@ def test_multi_replica_state(tmp_path, shapes_config):
                                out2["mrcnn_detection"], atol=1e-5)
     for b in state.batchers:
         b.close()
+
+
+def test_batcher_propagates_engine_errors():
+    from maskrcnn_amd.serving.batcher import DynamicBatcher
+
+    class Boom:
+        def predict(self, *a, **k):
+            raise ValueError("engine exploded")
+
+    b = DynamicBatcher(Boom(), max_batch_size=2,
+                       batch_timeout_micros=1000)
+    fut = b.submit(np.zeros((1, 4, 4, 3), np.float32),
+                   np.zeros((1, 5), np.float32))
+    with pytest.raises(ValueError, match="engine exploded"):
+        fut.result(timeout=10)
+    b.close()
+
+
+def test_trainer_custom_callbacks(shapes_config, shapes_dataset,
+                                  tmp_path):
+    from maskrcnn_amd.engine.trainer import Trainer
+    from maskrcnn_amd.model import MaskRCNN
+    cfg = shapes_config
+    m = MaskRCNN("training", cfg, model_dir=str(tmp_path))
+    tr = Trainer(m, shapes_dataset)
+    seen = []
+    tr.fit(cfg.LEARNING_RATE, epochs=1, layers="heads",
+           steps_per_epoch=1,
+           custom_callbacks=[lambda ep, sc: seen.append((ep, sc))])
+    assert len(seen) == 1
+    assert seen[0][0] == 0 and "total_loss" in seen[0][1]
+
+
+def test_data_loader_skips_bad_images(shapes_config, shapes_dataset):
+    """Reference parity: a failing load_image is logged and another
+    image drawn (data_generator error tolerance, model.py:~1700)."""
+    from maskrcnn_amd.data.loader import TorchMaskRCNNDataset
+    broken = shapes_dataset
+    orig = broken.load_image
+
+    def flaky(image_id):
+        if image_id == 0:
+            raise IOError("corrupt image")
+        return orig(image_id)
+
+    broken.load_image = flaky
+    try:
+        tds = TorchMaskRCNNDataset(broken, shapes_config)
+        sample = tds[0]  # index 0 fails -> a replacement is drawn
+        assert sample["image"].shape[0] == 3
+    finally:
+        broken.load_image = orig

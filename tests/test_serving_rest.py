@@ -79,3 +79,38 @@ def test_rest_unknown_model_404(rest_server):
     r = requests.post(f"http://127.0.0.1:{port}/v1/models/nope:predict",
                       json={"inputs": {}}, timeout=10)
     assert r.status_code == 404
+
+
+def test_rest_metadata_endpoint(rest_server):
+    import requests
+    server, cfg, port = rest_server
+    r = requests.get(
+        f"http://127.0.0.1:{port}/v1/models/mask_rcnn/metadata",
+        timeout=10)
+    assert r.status_code == 200
+    md = r.json()
+    sig = md["metadata"]["signature_def"]["signature_def"][
+        "serving_default"]
+    assert set(sig["inputs"]) == {"input_image", "input_image_meta",
+                                  "input_anchors"}
+    assert len(sig["outputs"]) == 7
+    assert md["model_spec"]["name"] == "mask_rcnn"
+
+
+def test_rest_predict_instances_format(rest_server):
+    """TF-Serving row ('instances') JSON schema."""
+    import requests
+    server, cfg, port = rest_server
+    H = cfg.IMAGE_SHAPE[0]
+    inst = {
+        "input_image": np.zeros((H, H, 3), np.float32).tolist(),
+        "input_image_meta": np.concatenate(
+            [[0], (H, H, 3), (H, H, 3), (0, 0, H, H), [1.0],
+             np.ones(cfg.NUM_CLASSES)]).tolist(),
+    }
+    r = requests.post(
+        f"http://127.0.0.1:{port}/v1/models/mask_rcnn:predict",
+        json={"instances": [inst]}, timeout=120)
+    assert r.status_code == 200, r.text[:300]
+    out = r.json()["outputs"]
+    assert "mrcnn_detection" in out
