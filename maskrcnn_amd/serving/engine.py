@@ -42,6 +42,17 @@ class InferenceEngine:
         self.anchor_cache = AnchorCache(config)
         self._lock = threading.Lock()  # one inference stream per engine
         self._pinned: Dict[tuple, "object"] = {}  # staging buffers
+        # hipGraph replay of the steady-state forward (N14). The
+        # inference path is sync-free by construction (padded NMS,
+        # masked scatter — no .item()/.nonzero() host syncs), which is
+        # what makes whole-forward capture legal. Off by default until
+        # GPU-validated (round-2 block); enable per-engine or via
+        # MASKRCNN_SERVING_HIPGRAPH=1.
+        import os
+        self.use_hip_graph = (
+            self.device.type == "cuda" and
+            os.environ.get("MASKRCNN_SERVING_HIPGRAPH", "0") == "1")
+        self._graphs: Dict[tuple, dict] = {}  # (B,H,W) -> capture state
 
     def _upload(self, arr: np.ndarray):
         """Host->device through a reused pinned buffer (a 32x1024² fp32
@@ -61,6 +72,60 @@ class InferenceEngine:
     #: the 7 signature outputs (S1 order)
     OUTPUT_NAMES = ("mrcnn_detection", "mrcnn_class", "mrcnn_bbox",
                     "mrcnn_mask", "rois", "rpn_class", "rpn_bbox")
+
+    def _autocast(self):
+        import torch
+        return torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                              enabled=self.use_bf16)
+
+    def _forward(self, images, metas, anchors):
+        """Eager forward, or hipGraph replay when enabled.
+
+        Replay path: one captured graph per (batch, image-shape, anchor
+        count); inputs are copied into the capture's static buffers and
+        the whole forward replays as a single graph launch (the serving
+        forward is ~300 small launches; replay removes the per-launch
+        CPU dispatch that dominates at batch 1)."""
+        import torch
+        if not (self.use_hip_graph and images.is_cuda):
+            with self._autocast():
+                return self.model(images, metas, anchors,
+                                  mode="inference")
+        key = (tuple(images.shape), tuple(anchors.shape))
+        st = self._graphs.get(key)
+        if st is None:
+            st = self._capture(images, metas, anchors)
+        st["img"].copy_(images)
+        st["meta"].copy_(metas)
+        st["anc"].copy_(anchors)
+        st["graph"].replay()
+        return st["out"]
+
+    def _capture(self, images, metas, anchors):
+        """Warm up (3 eager runs on a side stream — also drives MIOpen
+        Find for the shape) then capture one forward into a hipGraph
+        (torch.cuda.CUDAGraph IS hipGraph on ROCm)."""
+        import torch
+        static_img = images.clone()
+        static_meta = metas.clone()
+        static_anc = anchors.clone()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                with self._autocast():
+                    self.model(static_img, static_meta, static_anc,
+                               mode="inference")
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            with self._autocast():
+                out = self.model(static_img, static_meta, static_anc,
+                                 mode="inference")
+        st = {"graph": g, "img": static_img, "meta": static_meta,
+              "anc": static_anc, "out": out}
+        self._graphs[(tuple(images.shape), tuple(anchors.shape))] = st
+        return st
 
     def predict(self, input_image: np.ndarray,
                 input_image_meta: np.ndarray,
@@ -98,9 +163,7 @@ class InferenceEngine:
             else:
                 anchors = self.anchor_cache.get((H, W, 3),
                                                 device=self.device)
-            with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
-                                enabled=self.use_bf16):
-                out = self.model(images, metas, anchors, mode="inference")
+            out = self._forward(images, metas, anchors)
             # Trim the padded batch rows GPU-side BEFORE DtoH (a padded
             # batch-32 output set is ~1 GB) and transfer only the
             # requested outputs (a full set adds ~12 MB/image of rpn/roi
