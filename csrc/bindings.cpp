@@ -31,6 +31,8 @@ void conv3x3_v2_launch(const void*, const void*, const float*, const void*,
                        void*, int, int, int, int, int, bool, hipStream_t);
 void conv3x3_v3_launch(const void*, const void*, const float*, const void*,
                        void*, int, int, int, int, int, bool, hipStream_t);
+void conv3x3_wrw_launch(const void*, const void*, float*, int, int, int,
+                        int, int, hipStream_t);
 }
 
 namespace {
@@ -237,6 +239,30 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
   return out;
 }
 
+torch::Tensor conv3x3_wrw(torch::Tensor x, torch::Tensor dy) {
+  // x: [B,Cin,H,W], dy: [B,Cout,H,W] — both channels_last bf16.
+  // Returns dW fp32 as [9, Cout, Cin] (tap-major; the python wrapper
+  // reshapes to the torch [Cout, Cin, 3, 3] layout).
+  TORCH_CHECK(x.is_cuda() && dy.is_cuda(), "conv3x3_wrw: GPU tensors");
+  TORCH_CHECK(x.dim() == 4 && dy.dim() == 4, "x/dy must be 4-D");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+                  dy.scalar_type() == torch::kBFloat16,
+              "conv3x3_wrw: bf16 only");
+  TORCH_CHECK(is_channels_last(x) && is_channels_last(dy),
+              "conv3x3_wrw: channels_last required");
+  const int B = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  const int Cout = dy.size(1);
+  TORCH_CHECK(dy.size(0) == B && dy.size(2) == H && dy.size(3) == W,
+              "conv3x3_wrw: x/dy spatial mismatch");
+  TORCH_CHECK(Cin % 8 == 0 && Cout % 8 == 0,
+              "conv3x3_wrw: channels must be multiples of 8");
+  auto dw = torch::zeros({9, Cout, Cin},
+                         x.options().dtype(torch::kFloat));
+  conv3x3_wrw_launch(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(),
+                     B, H, W, Cin, Cout, cur_stream());
+  return dw;
+}
+
 }  // namespace
 
 void register_rle(pybind11::module_& m);
@@ -260,4 +286,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "3x3 conv s1p1 + bias + optional ReLU (CDNA4 MFMA implicit GEMM)",
         py::arg("x"), py::arg("w_packed"), py::arg("bias"),
         py::arg("relu") = false, py::arg("version") = 0);
+  m.def("conv3x3_wrw", &conv3x3_wrw,
+        "3x3 conv s1p1 weight gradient (CDNA4 MFMA, split-K atomics)",
+        py::arg("x"), py::arg("dy"));
 }

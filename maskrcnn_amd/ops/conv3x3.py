@@ -4,8 +4,9 @@ The SURVEY.md N1 op: frozen BatchNorm folds into the conv (w_eff = w *
 scale_per_outchannel; bias_eff = shift + scale * conv_bias), computed
 differentiably in torch so training still updates w/bias/BN-affine; the
 folded conv+ReLU then runs as one hand-written CDNA4 implicit-GEMM MFMA
-kernel (csrc/hip/conv3x3.hip). Backward uses torch.nn.grad conv grads
-(MIOpen-backed) — the forward fusion is the custom part.
+kernel (csrc/hip/conv3x3.hip). Backward: dgrad can reuse the forward
+kernel on rotated weights; wrw has its own split-K MFMA kernel
+(csrc/hip/conv3x3_wrw.hip, MASKRCNN_HIP_WRW=1); aten/MIOpen otherwise.
 """
 
 from __future__ import annotations
@@ -78,10 +79,27 @@ class _Conv3x3Fn(torch.autograd.Function):
             from .conv_policy import use_hip_conv3x3
             use_hip_dgrad = use_hip_conv3x3(
                 go_cl.size(0), w.size(0), go_cl.size(2), go_cl.size(3))
+        # Weight gradient: HIP split-K MFMA kernel (csrc/hip/
+        # conv3x3_wrw.hip — wrw was 47% of the original MIOpen step
+        # profile) when opted in; aten/MIOpen otherwise. Off by default
+        # until GPU-validated (round-2 block1; MASKRCNN_HIP_WRW=1).
+        import os
+        use_hip_wrw = (
+            ctx.needs_input_grad[1] and ext is not None and
+            grad_out.is_cuda and go_cl.dtype == torch.bfloat16 and
+            x.dtype == torch.bfloat16 and
+            x.is_contiguous(memory_format=torch.channels_last) and
+            x.size(1) % 8 == 0 and w.size(0) % 8 == 0 and
+            os.environ.get("MASKRCNN_HIP_WRW", "0") == "1")
         need_aten_gx = ctx.needs_input_grad[0] and not use_hip_dgrad
+        need_aten_gw = ctx.needs_input_grad[1] and not use_hip_wrw
         gx, gw, _ = torch.ops.aten.convolution_backward(
             go_cl, x, wd, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
-            [need_aten_gx, ctx.needs_input_grad[1], False])
+            [need_aten_gx, need_aten_gw, False])
+        if use_hip_wrw:
+            # [9, Cout, Cin] tap-major fp32 -> [Cout, Cin, 3, 3].
+            gw = ext.conv3x3_wrw(x, go_cl) \
+                .reshape(3, 3, w.size(0), w.size(1)).permute(2, 3, 0, 1)
         if use_hip_dgrad:
             wr = pack_weight(rotate_for_dgrad(w))
             zb = torch.zeros(w.size(1), device=w.device,

@@ -204,3 +204,55 @@ def test_gpu_dgrad_via_forward_kernel():
     ref.backward(g.float())
     torch.testing.assert_close(gx_hip.float(), x2.grad, atol=0.3,
                                rtol=0.05)
+
+
+@pytest.mark.gpu
+@pytest.mark.xfail(strict=False,
+                   reason="wrw kernel pending first GPU validation")
+@pytest.mark.parametrize("cin,cout,h,w,batch", [
+    (64, 64, 64, 48, 2),      # early resnet, ragged W segment
+    (256, 256, 32, 32, 2),    # FPN shape, exact segments
+    (256, 256, 14, 14, 8),    # mask-head shape (W < one K chunk)
+    (128, 192, 30, 50, 3),    # ragged channel tiles
+])
+def test_gpu_wrw_kernel_matches_reference(cin, cout, h, w, batch):
+    """Split-K MFMA weight gradient vs fp32 autograd reference."""
+    from maskrcnn_amd.ops._ext import require_ext
+    ext = require_ext()
+    dev = torch.device("cuda:0")
+    torch.manual_seed(0)
+    x = torch.randn(batch, cin, h, w, device=dev).bfloat16() \
+        .to(memory_format=torch.channels_last)
+    dy = (torch.randn(batch, cout, h, w, device=dev) * 0.1).bfloat16() \
+        .to(memory_format=torch.channels_last)
+    dw = ext.conv3x3_wrw(x, dy) \
+        .reshape(3, 3, cout, cin).permute(2, 3, 0, 1)
+    xf = x.float().requires_grad_(True)
+    wf = torch.zeros(cout, cin, 3, 3, device=dev, requires_grad=True)
+    F.conv2d(xf, wf, padding=1).backward(dy.float())
+    # K = batch*h*w bf16 products accumulated in fp32: tolerance scales
+    # with sqrt(K) * input scale.
+    torch.testing.assert_close(dw, wf.grad, atol=3.0 * (batch * h * w) **
+                               0.5 * 0.01, rtol=0.05)
+
+
+@pytest.mark.gpu
+@pytest.mark.xfail(strict=False,
+                   reason="wrw kernel pending first GPU validation")
+def test_gpu_autograd_uses_hip_wrw(monkeypatch):
+    """MASKRCNN_HIP_WRW=1 routes the Function's weight grad through the
+    kernel and matches the default aten path."""
+    monkeypatch.setenv("MASKRCNN_HIP_WRW", "1")
+    dev = torch.device("cuda:0")
+    torch.manual_seed(0)
+    x = torch.randn(2, 64, 32, 32, device=dev).bfloat16() \
+        .to(memory_format=torch.channels_last).requires_grad_(True)
+    wt = (torch.randn(64, 64, 3, 3, device=dev) * 0.05).requires_grad_()
+    b = torch.randn(64, device=dev, requires_grad=True)
+    out = conv3x3_bias_relu(x, wt, b, relu=False)
+    out.sum().backward()
+    g_hip = wt.grad.clone()
+    monkeypatch.setenv("MASKRCNN_HIP_WRW", "0")
+    wt.grad = None
+    conv3x3_bias_relu(x, wt, b, relu=False).sum().backward()
+    torch.testing.assert_close(g_hip, wt.grad, atol=2.0, rtol=0.05)

@@ -15,6 +15,11 @@ block1)  # v3 kernel validation + conv A/B (~4 min)
   tail -4 gpurun_out/r2_conv_tests.log
   timeout 400 python tools/conv_bench.py > gpurun_out/r2_conv_ab.log 2>&1
   grep -v libdrm gpurun_out/r2_conv_ab.log
+  # hipGraph serving replay (engine._forward capture path)
+  timeout 300 python -m pytest \
+      tests/test_gpu_integration.py::test_engine_hipgraph_matches_eager \
+      -m gpu -q --runxfail >> gpurun_out/r2_conv_tests.log 2>&1
+  tail -2 gpurun_out/r2_conv_tests.log
   ;;
 block2)  # training-path A/Bs (one box; ~12 min incl. one find phase)
   run() { timeout 500 python bench.py --steps 15 --warmup 5 \
@@ -26,6 +31,8 @@ block2)  # training-path A/Bs (one box; ~12 min incl. one find phase)
   echo "=== all three";        MASKRCNN_TRAIN_FUSED_FPN=1 \
       MASKRCNN_TRAIN_HIP_CONV=1 MASKRCNN_TRAIN_CL_ROI=1 run all
   echo "=== vectorized targets"; MASKRCNN_VECTORIZED_TARGETS=1 run vtgt
+  echo "=== hip wrw";          MASKRCNN_TRAIN_HIP_CONV=1 \
+      MASKRCNN_HIP_WRW=1 run wrw
   ;;
 block3)  # serving bench (~8 min incl. 265 s batch-32 find)
   timeout 900 python bench/serving_bench.py --requests 96 \
