@@ -133,17 +133,29 @@ class ModelState:
 
 
 class ModelServer:
-    def __init__(self, model_name: str, model_base_path: str,
+    def __init__(self, model_name: str = None, model_base_path: str = None,
                  grpc_port: int = 8500, rest_port: int = 8501,
                  metrics_port: Optional[int] = None, device=None,
                  devices=None, max_batch_size: int = 32,
                  batch_timeout_micros: int = 2000,
-                 poll_interval_s: float = 5.0, config=None):
-        self.state = ModelState(model_name, model_base_path, device=device,
-                                devices=devices,
-                                max_batch_size=max_batch_size,
-                                batch_timeout_micros=batch_timeout_micros,
-                                config=config)
+                 poll_interval_s: float = 5.0, config=None,
+                 model_configs=None):
+        """Single model (model_name + model_base_path, TF-Serving's
+        --model_name/--model_base_path mode) or several (model_configs =
+        parse_model_config(models.config), the --model_config_file
+        mode). All models share the batching parameters and devices."""
+        if model_configs is None:
+            if model_name is None or model_base_path is None:
+                raise ValueError("model_name+model_base_path or "
+                                 "model_configs required")
+            model_configs = [{"name": model_name,
+                              "base_path": model_base_path}]
+        self.models: Dict[str, ModelState] = {
+            mc["name"]: ModelState(
+                mc["name"], mc["base_path"], device=device,
+                devices=devices, max_batch_size=max_batch_size,
+                batch_timeout_micros=batch_timeout_micros, config=config)
+            for mc in model_configs}
         self.grpc_port = grpc_port
         self.rest_port = rest_port
         self.metrics_port = metrics_port
@@ -153,22 +165,34 @@ class ModelServer:
         self._rest_thread = None
         self._uvicorn_server = None
 
+    @property
+    def state(self) -> ModelState:
+        """The first (default) model — single-model back-compat."""
+        return next(iter(self.models.values()))
+
+    def _get_state(self, name: Optional[str]) -> ModelState:
+        if not name:
+            return self.state
+        st = self.models.get(name)
+        if st is None:
+            raise KeyError(f"unknown model {name!r}")
+        return st
+
     # ----------------------------------------------------------- predict
     def _handle_predict(self, request: PredictRequest) -> PredictResponse:
         t0 = time.monotonic()
         if _PROM:
             REQS.labels("grpc").inc()
-        name = request.model_spec.name or self.state.name
-        if name != self.state.name:
-            raise KeyError(f"unknown model {name!r}")
+        state = self._get_state(request.model_spec.name)
+        name = state.name
         sig = request.model_spec.signature_name or "serving_default"
         if sig != "serving_default":
             raise KeyError(f"unknown signature {sig!r}")
         want_version = request.model_spec.version
-        if want_version is not None and want_version != self.state.version:
+        if want_version is not None and want_version != state.version:
             raise KeyError(
                 f"model {name!r} version {want_version} not loaded "
-                f"(serving {self.state.version})")
+                f"(serving {state.version})")
         try:
             images = tensor_to_numpy(request.inputs["input_image"]) \
                 .astype(np.float32)
@@ -182,7 +206,7 @@ class ModelServer:
                 request.inputs["input_anchors"]).astype(np.float32)
         # The filter is pushed into the engine: unrequested outputs skip
         # the DtoH transfer entirely.
-        outputs, version = self.state.predict(
+        outputs, version = state.predict(
             images, metas, anchors,
             outputs=list(request.output_filter) or None)
         if request.output_filter:
@@ -190,7 +214,7 @@ class ModelServer:
                        if k in request.output_filter}
         resp = PredictResponse(
             outputs={k: numpy_to_tensor(v) for k, v in outputs.items()},
-            model_spec=ModelSpec(name=self.state.name, version=version,
+            model_spec=ModelSpec(name=name, version=version,
                                  signature_name="serving_default"))
         if _PROM:
             LAT.observe(time.monotonic() - t0)
@@ -218,16 +242,16 @@ class ModelServer:
                 context.abort(grpc.StatusCode.INTERNAL, str(e))
 
         def get_model_metadata(request: GetModelMetadataRequest, context):
-            name = request.model_spec.name or outer.state.name
-            if name != outer.state.name:
-                context.abort(grpc.StatusCode.NOT_FOUND,
-                              f"unknown model {name!r}")
-            eng = outer.state.engine
+            try:
+                state = outer._get_state(request.model_spec.name)
+            except KeyError as e:
+                context.abort(grpc.StatusCode.NOT_FOUND, str(e))
+            eng = state.engine
             sigs = {"serving_default": build_signature(eng.config)} \
                 if eng is not None else {}
             return GetModelMetadataResponse(
-                model_spec=ModelSpec(name=outer.state.name,
-                                     version=outer.state.version),
+                model_spec=ModelSpec(name=state.name,
+                                     version=state.version),
                 signature_defs=sigs)
 
         handler = grpc.method_handlers_generic_handler(
@@ -259,18 +283,18 @@ class ModelServer:
 
         @app.get("/v1/models/{model_name}/metadata")
         def model_metadata(model_name: str):
-            if model_name != outer.state.name or \
-                    outer.state.engine is None:
+            state = outer.models.get(model_name)
+            if state is None or state.engine is None:
                 return JSONResponse({"error": "model not found"},
                                     status_code=404)
-            sig = build_signature(outer.state.engine.config)
+            sig = build_signature(state.engine.config)
             def ti(t):
                 return {"dtype": "DT_FLOAT",
                         "tensor_shape": {"dim": [{"size": str(d)}
                                                  for d in t.shape]},
                         "name": t.name}
-            return {"model_spec": {"name": outer.state.name,
-                                   "version": str(outer.state.version)},
+            return {"model_spec": {"name": state.name,
+                                   "version": str(state.version)},
                     "metadata": {"signature_def": {"signature_def": {
                         "serving_default": {
                             "inputs": {k: ti(v)
@@ -281,10 +305,10 @@ class ModelServer:
 
         @app.get("/v1/models/{model_name}")
         def model_status(model_name: str):
-            ok = (model_name == outer.state.name and
-                  outer.state.version is not None)
+            state = outer.models.get(model_name)
+            ok = state is not None and state.version is not None
             return {"model_version_status": [{
-                "version": str(outer.state.version or 0),
+                "version": str((state.version if state else None) or 0),
                 "state": "AVAILABLE" if ok else "UNAVAILABLE",
                 "status": {"error_code": "OK" if ok else "NOT_FOUND",
                            "error_message": ""},
@@ -298,7 +322,8 @@ class ModelServer:
                 return JSONResponse({"error": "unsupported verb"},
                                     status_code=404)
             name = model_and_verb[:-len(":predict")]
-            if name != outer.state.name:
+            state = outer.models.get(name)
+            if state is None:
                 return JSONResponse(
                     {"error": f"model {name} not found"}, status_code=404)
             try:
@@ -328,7 +353,7 @@ class ModelServer:
                         {"error": "missing inputs/instances"},
                         status_code=400)
                 t0 = time.monotonic()
-                outputs, _ = outer.state.predict(images, metas, anchors)
+                outputs, _ = state.predict(images, metas, anchors)
                 if _PROM:
                     LAT.observe(time.monotonic() - t0)
                 return {"outputs": {k: v.tolist()
@@ -343,13 +368,15 @@ class ModelServer:
 
     # --------------------------------------------------------- lifecycle
     def start(self, wait_for_model: bool = True, warmup: bool = False):
-        self.state.maybe_reload()
-        if wait_for_model and self.state.version is None:
-            raise FileNotFoundError(
-                f"no servable version under {self.state.base_path}")
+        for state in self.models.values():
+            state.maybe_reload()
+            if wait_for_model and state.version is None:
+                raise FileNotFoundError(
+                    f"no servable version under {state.base_path}")
         if warmup:
             log.info("warmup predict at max batch size...")
-            self.state.warmup()
+            for state in self.models.values():
+                state.warmup()
         if self.metrics_port and _PROM:
             start_http_server(self.metrics_port)
         self._grpc_server = self._build_grpc()
@@ -369,16 +396,18 @@ class ModelServer:
         self._poll_thread = threading.Thread(target=self._poll_loop,
                                              daemon=True)
         self._poll_thread.start()
-        log.info("ModelServer up: grpc=:%d rest=:%s model=%s v%s",
-                 self.grpc_port, self.rest_port, self.state.name,
-                 self.state.version)
+        log.info("ModelServer up: grpc=:%d rest=:%s models=%s",
+                 self.grpc_port, self.rest_port,
+                 {n: s.version for n, s in self.models.items()})
 
     def _poll_loop(self):
         while not self._stop.wait(self.poll_interval_s):
-            try:
-                self.state.maybe_reload()
-            except Exception:  # noqa: BLE001
-                log.exception("version reload failed")
+            for state in self.models.values():
+                try:
+                    state.maybe_reload()
+                except Exception:  # noqa: BLE001
+                    log.exception("version reload failed (%s)",
+                                  state.name)
 
     def stop(self, grace: float = 1.0):
         self._stop.set()
@@ -386,8 +415,9 @@ class ModelServer:
             self._grpc_server.stop(grace).wait()
         if self._uvicorn_server is not None:
             self._uvicorn_server.should_exit = True
-        for b in self.state.batchers:
-            b.close()
+        for state in self.models.values():
+            for b in state.batchers:
+                b.close()
 
     def wait(self):
         self._grpc_server.wait_for_termination()
@@ -397,11 +427,16 @@ def main():  # pragma: no cover
     import argparse
     ap = argparse.ArgumentParser(description="maskrcnn_amd model server")
     ap.add_argument("--model_name", default="mask_rcnn")
-    ap.add_argument("--model_base_path", required=True)
+    ap.add_argument("--model_base_path", default=None)
+    ap.add_argument("--model_config_file", default=None,
+                    help="TF-Serving models.config text proto "
+                         "(multi-model mode)")
     ap.add_argument("--port", type=int, default=8500)
     ap.add_argument("--rest_api_port", type=int, default=8501)
     ap.add_argument("--metrics_port", type=int, default=None)
     ap.add_argument("--enable_batching", action="store_true")
+    ap.add_argument("--batching_parameters_file", default=None,
+                    help="TF-Serving batching_parameters.txt text proto")
     ap.add_argument("--num_replicas", type=int, default=1,
                     help="one engine per GPU: cuda:0..cuda:N-1")
     ap.add_argument("--max_batch_size", type=int, default=32)
@@ -411,12 +446,24 @@ def main():  # pragma: no cover
     devices = None
     if args.num_replicas > 1:
         devices = [f"cuda:{i}" for i in range(args.num_replicas)]
+    max_batch = args.max_batch_size if args.enable_batching else 1
+    timeout_us = args.batch_timeout_micros
+    if args.batching_parameters_file:
+        from .config_file import parse_batching_parameters
+        bp = parse_batching_parameters(args.batching_parameters_file)
+        max_batch = bp.get("max_batch_size", max_batch)
+        timeout_us = bp.get("batch_timeout_micros", timeout_us)
+    model_configs = None
+    if args.model_config_file:
+        from .config_file import parse_model_config
+        model_configs = parse_model_config(args.model_config_file)
+    elif args.model_base_path is None:
+        ap.error("--model_base_path or --model_config_file required")
     server = ModelServer(
         args.model_name, args.model_base_path, grpc_port=args.port,
         rest_port=args.rest_api_port, metrics_port=args.metrics_port,
-        devices=devices,
-        max_batch_size=args.max_batch_size if args.enable_batching else 1,
-        batch_timeout_micros=args.batch_timeout_micros)
+        devices=devices, model_configs=model_configs,
+        max_batch_size=max_batch, batch_timeout_micros=timeout_us)
     server.start()
     server.wait()
 

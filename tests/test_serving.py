@@ -381,3 +381,86 @@ def test_data_loader_skips_bad_images(shapes_config, shapes_dataset):
         assert sample["image"].shape[0] == 3
     finally:
         broken.load_image = orig
+
+
+def test_text_proto_parsers(tmp_path):
+    from maskrcnn_amd.serving.config_file import (
+        parse_batching_parameters, parse_model_config, parse_text_proto)
+    mc = tmp_path / "models.config"
+    mc.write_text("""
+# served models
+model_config_list {
+  config {
+    name: 'mask_rcnn'
+    base_path: '/models/mask_rcnn'
+    model_platform: 'tensorflow'
+  }
+  config { name: "second"  base_path: "/models/second" }
+}
+""")
+    cfgs = parse_model_config(str(mc))
+    assert cfgs == [
+        {"name": "mask_rcnn", "base_path": "/models/mask_rcnn",
+         "model_platform": "tensorflow"},
+        {"name": "second", "base_path": "/models/second",
+         "model_platform": "tensorflow"}]
+    bp = tmp_path / "batching_parameters.txt"
+    bp.write_text("""
+max_batch_size { value: 32 }
+batch_timeout_micros { value: 2000 }
+num_batch_threads { value: 8 }
+max_enqueued_batches { value: 1000000 }
+""")
+    assert parse_batching_parameters(str(bp)) == {
+        "max_batch_size": 32, "batch_timeout_micros": 2000,
+        "num_batch_threads": 8, "max_enqueued_batches": 1000000}
+    with pytest.raises(ValueError):
+        parse_text_proto("a { b: 1")
+
+
+def test_multi_model_server(tmp_path):
+    """--model_config_file mode: two models behind one front door,
+    routed by model_spec.name (grpc) and URL (rest status)."""
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+    from maskrcnn_amd.serving.server import ModelServer
+    from maskrcnn_amd.serving.wireproto import (ModelSpec, PredictRequest,
+                                                numpy_to_tensor)
+
+    class SmallConfig(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 200
+        POST_NMS_ROIS_INFERENCE = 20
+
+    cfg = SmallConfig()
+    configs = []
+    for name in ("alpha", "beta"):
+        m = MaskRCNN("inference", cfg, model_dir=str(tmp_path))
+        base = str(tmp_path / "models" / name)
+        export_saved_model(m, cfg, base, version=1)
+        configs.append({"name": name, "base_path": base})
+    server = ModelServer(model_configs=configs, grpc_port=0,
+                         rest_port=None, max_batch_size=1,
+                         batch_timeout_micros=100, config=cfg)
+    server.start()
+    try:
+        assert set(server.models) == {"alpha", "beta"}
+        H = cfg.IMAGE_SHAPE[0]
+        meta = np.zeros((1, cfg.IMAGE_META_SIZE), np.float32)
+        meta[:, 4:7] = (H, H, 3)
+        meta[:, 7:11] = (0, 0, H, H)
+        for name in ("alpha", "beta"):
+            req = PredictRequest(
+                model_spec=ModelSpec(name=name),
+                inputs={"input_image": numpy_to_tensor(
+                            np.zeros((1, H, H, 3), np.float32)),
+                        "input_image_meta": numpy_to_tensor(meta)})
+            resp = server._handle_predict(req)
+            assert resp.model_spec.name == name
+            assert "mrcnn_detection" in resp.outputs
+        with pytest.raises(KeyError):
+            server._handle_predict(PredictRequest(
+                model_spec=ModelSpec(name="gamma"), inputs={}))
+    finally:
+        server.stop()
