@@ -464,3 +464,33 @@ def test_multi_model_server(tmp_path):
                 model_spec=ModelSpec(name="gamma"), inputs={}))
     finally:
         server.stop()
+
+
+def test_demo_app_roundtrip(shapes_config, tmp_path):
+    """S5 demo app: upload PNG -> rendered detections PNG back."""
+    import importlib.util
+    import io as _io
+    spec = importlib.util.spec_from_file_location(
+        "demo_app", os.path.join(os.path.dirname(__file__), "..",
+                                 "samples", "demo_app.py"))
+    demo_app = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(demo_app)
+    from PIL import Image
+    from fastapi.testclient import TestClient
+    from maskrcnn_amd.model import MaskRCNN
+
+    m = MaskRCNN("inference", shapes_config, model_dir=str(tmp_path))
+    names = [f"c{i}" for i in range(shapes_config.NUM_CLASSES)]
+    app = demo_app.create_app(lambda img: m.detect([img])[0], names)
+    client = TestClient(app)
+    assert "Detect" in client.get("/").text
+    img = Image.fromarray(
+        (np.random.rand(96, 96, 3) * 255).astype(np.uint8))
+    buf = _io.BytesIO()
+    img.save(buf, format="PNG")
+    r = client.post("/detect", content=buf.getvalue())
+    assert r.status_code == 200
+    assert r.headers["content-type"] == "image/png"
+    assert Image.open(_io.BytesIO(r.content)).size[0] > 0
+    r = client.post("/detect", content=b"not an image")
+    assert r.status_code == 400
