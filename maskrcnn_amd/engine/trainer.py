@@ -192,6 +192,7 @@ class Trainer:
         if self.distributed:
             self._rebuild_ddp()
         optimizer = build_optimizer(self.model, cfg, learning_rate)
+        self._resumed_opt = self._maybe_resume_optimizer(optimizer)
         loader = make_data_loader(self.train_dataset, cfg,
                                   augmentation=self.augmentation,
                                   num_workers=self.num_workers,
@@ -232,5 +233,37 @@ class Trainer:
             if self.rank == 0:
                 ckpt = self.wrapper.checkpoint_path.format(epoch=epoch)
                 self.wrapper.save_weights(ckpt)
+                # Full train state beside the weights (the reference
+                # saved weights only; SGD momentum + epoch make resume
+                # exact). Latest-only: overwritten each epoch.
+                torch.save({"optimizer": optimizer.state_dict(),
+                            "epoch": epoch},
+                           os.path.join(self.wrapper.log_dir,
+                                        "train_state.pth"))
             self.wrapper.epoch += 1
         return self.history
+
+    def _maybe_resume_optimizer(self, optimizer) -> bool:
+        """Restore SGD momentum when resuming (find_last/set_log_dir set
+        wrapper.epoch > 0 and the matching train_state.pth is present).
+        Param-group mismatch (different frozen set) is non-fatal: log
+        and train with fresh momentum."""
+        if self.wrapper.epoch == 0:
+            return False
+        path = os.path.join(self.wrapper.log_dir, "train_state.pth")
+        if not os.path.exists(path):
+            return False
+        try:
+            state = torch.load(path, map_location="cpu",
+                               weights_only=True)
+            if state.get("epoch") != self.wrapper.epoch - 1:
+                log.warning("train_state.pth is for epoch %s, resuming "
+                            "at %d: skipping optimizer restore",
+                            state.get("epoch"), self.wrapper.epoch)
+                return False
+            optimizer.load_state_dict(state["optimizer"])
+            log.info("restored optimizer state from %s", path)
+            return True
+        except (ValueError, KeyError, RuntimeError) as e:
+            log.warning("could not restore optimizer state: %s", e)
+            return False

@@ -41,6 +41,8 @@ try:
     BATCH = Histogram("predict_batch_size", "executed batch size",
                       buckets=(1, 2, 4, 8, 16, 32, 64))
     VERSION = Gauge("model_version", "loaded model version")
+    HBM = Gauge("hbm_bytes_allocated",
+                "torch.cuda.memory_allocated across replicas")
 except ImportError:  # pragma: no cover
     _PROM = False
 
@@ -219,6 +221,10 @@ class ModelServer:
         if _PROM:
             LAT.observe(time.monotonic() - t0)
             BATCH.observe(images.shape[0])
+            import torch
+            if torch.cuda.is_available():
+                HBM.set(sum(torch.cuda.memory_allocated(d)
+                            for d in range(torch.cuda.device_count())))
         return resp
 
     # -------------------------------------------------------------- grpc

@@ -113,7 +113,7 @@ def test_serving_version_pinning():
     from maskrcnn_amd.serving.wireproto import (ModelSpec, PredictRequest,
                                                 numpy_to_tensor)
     ms = ModelServer.__new__(ModelServer)
-    ms.state = ModelState("m", "/nonexistent")
+    ms.models = {"m": ModelState("m", "/nonexistent")}
     ms.state.version = 3
     req = PredictRequest(model_spec=ModelSpec(name="m", version=2))
     req.inputs["input_image"] = numpy_to_tensor(

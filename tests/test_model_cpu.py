@@ -111,3 +111,30 @@ def test_load_weights_exclude_head_surgery(shapes_config, tmp_path):
     assert torch.equal(
         m2.model.backbone.conv1.weight,
         m.model.backbone.conv1.weight)
+
+
+def test_optimizer_state_resume(shapes_config, shapes_dataset, tmp_path):
+    """Epoch checkpoints save SGD momentum beside the weights; a fresh
+    MaskRCNN resumed via load_weights(find_last()) restores it."""
+    import os
+
+    from maskrcnn_amd.engine.trainer import Trainer
+    from maskrcnn_amd.model import MaskRCNN
+
+    cfg = shapes_config
+    m = MaskRCNN("training", cfg, model_dir=str(tmp_path))
+    tr = Trainer(m, shapes_dataset)
+    tr.fit(cfg.LEARNING_RATE, epochs=1, layers="heads",
+           steps_per_epoch=1)
+    state_path = os.path.join(m.log_dir, "train_state.pth")
+    assert os.path.exists(state_path)
+    assert not tr._resumed_opt
+
+    m2 = MaskRCNN("training", cfg, model_dir=str(tmp_path))
+    last = m2.find_last()
+    m2.load_weights(last, by_name=True)
+    assert m2.epoch == 1
+    tr2 = Trainer(m2, shapes_dataset)
+    tr2.fit(cfg.LEARNING_RATE, epochs=2, layers="heads",
+            steps_per_epoch=1)
+    assert tr2._resumed_opt
