@@ -8,6 +8,7 @@ from .boxes import (  # noqa: F401
     non_max_suppression, apply_box_deltas, box_refinement,
     norm_boxes, denorm_boxes, norm_boxes_t, denorm_boxes_t,
     apply_box_deltas_t, clip_boxes_t, box_refinement_t, overlaps_t,
+    batch_slice,
 )
 from .anchors import generate_anchors, generate_pyramid_anchors  # noqa: F401
 from .image import (  # noqa: F401

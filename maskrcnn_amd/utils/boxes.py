@@ -239,3 +239,26 @@ def denorm_boxes_t(boxes: torch.Tensor, shape) -> torch.Tensor:
     shift = torch.tensor([0., 0., 1., 1.],
                          dtype=boxes.dtype, device=boxes.device)
     return boxes * scale + shift
+
+
+def batch_slice(inputs, graph_fn, batch_size, names=None):
+    """Apply graph_fn per batch item and restack (reference
+    utils.batch_slice, mrcnn/utils.py:~820 — a TF workaround for ops
+    without batch support; kept for API parity, though the MI355X model
+    graph is natively batched and never calls it).
+
+    inputs: tensor or list of tensors with leading batch dim.
+    graph_fn: callable over per-item slices returning tensor(s).
+    """
+    if not isinstance(inputs, list):
+        inputs = [inputs]
+    outputs = []
+    for i in range(batch_size):
+        out = graph_fn(*[x[i] for x in inputs])
+        if not isinstance(out, (list, tuple)):
+            out = [out]
+        outputs.append(out)
+    # [item][output] -> [output][item], stacked.
+    outputs = list(zip(*outputs))
+    result = [torch.stack(list(o), dim=0) for o in outputs]
+    return result[0] if len(result) == 1 else result

@@ -75,3 +75,18 @@ def test_clip_boxes_t():
     window = torch.tensor([0., 0., 1., 1.])
     out = B.clip_boxes_t(boxes, window)
     assert out.min() >= 0 and out.max() <= 1
+
+
+def test_batch_slice_parity():
+    """utils.batch_slice applies fn per item and restacks (reference
+    mrcnn/utils.py:~820)."""
+    import torch
+    from maskrcnn_amd.utils import batch_slice
+    x = torch.arange(12.0).reshape(3, 4)
+    y = torch.arange(6.0).reshape(3, 2)
+    out = batch_slice([x, y], lambda a, b: a.sum() + b.sum(), 3)
+    assert out.shape == (3,)
+    assert torch.equal(out, x.sum(1) + y.sum(1))
+    a, b = batch_slice(x, lambda v: (v * 2, v + 1), 2)
+    assert a.shape == (2, 4) and torch.equal(a, x[:2] * 2)
+    assert torch.equal(b, x[:2] + 1)
