@@ -33,6 +33,9 @@ void conv3x3_v3_launch(const void*, const void*, const float*, const void*,
                        void*, int, int, int, int, int, bool, hipStream_t);
 void conv3x3_wrw_launch(const void*, const void*, float*, int, int, int,
                         int, int, hipStream_t);
+void conv3x3_fp8_launch(const void*, const void*, const float*,
+                        const float*, const float*, void*, int, int, int,
+                        int, int, bool, hipStream_t);
 }
 
 namespace {
@@ -239,6 +242,41 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
   return out;
 }
 
+torch::Tensor conv3x3_fp8(torch::Tensor x, torch::Tensor w_packed,
+                          torch::Tensor bias, torch::Tensor w_scale,
+                          torch::Tensor a_scale, bool relu) {
+  // x: [B,Cin,H,W] channels_last, e4m3 passed as uint8 view;
+  // w_packed: [9,Cout,Cin] uint8 (e4m3, tap-major); w_scale: [Cout]
+  // f32; a_scale: [1] f32 DEVICE tensor (no host sync). Out: bf16 NHWC.
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "x must be [B,C,H,W] cuda");
+  TORCH_CHECK(x.scalar_type() == torch::kUInt8,
+              "x must be a uint8 view of e4m3");
+  TORCH_CHECK(is_channels_last(x), "x must be channels_last (NHWC)");
+  TORCH_CHECK(w_packed.dim() == 3 && w_packed.size(0) == 9 &&
+                  w_packed.scalar_type() == torch::kUInt8 &&
+                  w_packed.is_contiguous(),
+              "w_packed must be contiguous uint8 [9, Cout, Cin]");
+  const int B = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  const int Cout = w_packed.size(1);
+  TORCH_CHECK(w_packed.size(2) == Cin, "weight Cin mismatch");
+  TORCH_CHECK(Cin % 32 == 0, "Cin must be a multiple of 32");
+  TORCH_CHECK(a_scale.is_cuda() && a_scale.numel() == 1 &&
+                  a_scale.scalar_type() == torch::kFloat,
+              "a_scale must be a 1-elem f32 device tensor");
+  auto bias_f = bias.to(torch::kFloat).contiguous();
+  auto wsc = w_scale.to(torch::kFloat).contiguous();
+  TORCH_CHECK(wsc.numel() == Cout, "w_scale must be [Cout]");
+  auto out = torch::empty(
+      {B, Cout, H, W},
+      x.options().dtype(torch::kBFloat16)
+          .memory_format(at::MemoryFormat::ChannelsLast));
+  conv3x3_fp8_launch(x.data_ptr(), w_packed.data_ptr(),
+                     bias_f.data_ptr<float>(), wsc.data_ptr<float>(),
+                     a_scale.data_ptr<float>(), out.data_ptr(),
+                     B, H, W, Cin, Cout, relu, cur_stream());
+  return out;
+}
+
 torch::Tensor conv3x3_wrw(torch::Tensor x, torch::Tensor dy) {
   // x: [B,Cin,H,W], dy: [B,Cout,H,W] — both channels_last bf16.
   // Returns dW fp32 as [9, Cout, Cin] (tap-major; the python wrapper
@@ -286,6 +324,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "3x3 conv s1p1 + bias + optional ReLU (CDNA4 MFMA implicit GEMM)",
         py::arg("x"), py::arg("w_packed"), py::arg("bias"),
         py::arg("relu") = false, py::arg("version") = 0);
+  m.def("conv3x3_fp8", &conv3x3_fp8,
+        "3x3 conv s1p1, e4m3 in / bf16 out (fp8 MFMA rate)",
+        py::arg("x"), py::arg("w_packed"), py::arg("bias"),
+        py::arg("w_scale"), py::arg("a_scale"),
+        py::arg("relu") = false);
   m.def("conv3x3_wrw", &conv3x3_wrw,
         "3x3 conv s1p1 weight gradient (CDNA4 MFMA, split-K atomics)",
         py::arg("x"), py::arg("dy"));

@@ -92,7 +92,13 @@ class FPNMaskHead(nn.Module):
     def _conv_stack(self, x):
         """The 4x (3x3 conv + frozen-BN + ReLU) stack; each layer runs as
         one HIP MFMA kernel on GPU (14x14 maps: measured 1.19x over
-        MIOpen before counting the fused BN/ReLU, profiles/r01_conv_ab.md)."""
+        MIOpen before counting the fused BN/ReLU, profiles/r01_conv_ab.md).
+        MASKRCNN_FP8_COMPUTE=1 (serving, BASELINE config #5) runs the
+        folded convs at the fp8 MFMA rate with cached e4m3 weights."""
+        import os
+        if (self._fused and x.is_cuda and not self.training and
+                os.environ.get("MASKRCNN_FP8_COMPUTE", "0") == "1"):
+            return self._conv_stack_fp8(x)
         from ..ops.conv_policy import use_hip_conv3x3
         if not (self._fused and x.is_cuda and not self.training and
                 use_hip_conv3x3(x.shape[0], x.shape[1], x.shape[2],
@@ -106,6 +112,35 @@ class FPNMaskHead(nn.Module):
                                 bn.bias, bn.running_mean, bn.running_var,
                                 eps=bn.eps, relu=True)
         return x
+
+    def _conv_stack_fp8(self, x):
+        """e4m3-compute stack: BN-folded weights quantized once and
+        cached (weights are static in serving; the cache is dropped by
+        train() via the _fp8_cache attribute check in forward)."""
+        from ..ops.conv3x3 import conv3x3_fp8, pack_weight_fp8
+        if getattr(self, "_fp8_cache", None) is None:
+            import torch as _t
+            cache = []
+            with _t.no_grad():
+                for i in range(4):
+                    conv = self.convs[3 * i]
+                    bn = self.convs[3 * i + 1]
+                    s = bn.weight * (bn.running_var + bn.eps).rsqrt()
+                    w_eff = conv.weight * s.view(-1, 1, 1, 1)
+                    b_eff = bn.bias + (conv.bias - bn.running_mean) * s
+                    qp, scale = pack_weight_fp8(w_eff)
+                    cache.append((qp.to(x.device),
+                                  scale.to(x.device),
+                                  b_eff.float().to(x.device)))
+            self._fp8_cache = cache
+        for qp, scale, b_eff in self._fp8_cache:
+            x = conv3x3_fp8(x, qp, scale, b_eff, relu=True)
+        return x
+
+    def train(self, mode: bool = True):
+        if mode:
+            self._fp8_cache = None  # weights may change: drop quant cache
+        return super().train(mode)
 
     def forward(self, rois, feature_maps):
         """rois [B,N,4] -> mask LOGITS [B,N,C,28,28].

@@ -24,6 +24,42 @@ def pack_weight(w: torch.Tensor) -> torch.Tensor:
         .contiguous().to(torch.bfloat16)
 
 
+def pack_weight_fp8(w: torch.Tensor):
+    """[Cout, Cin, 3, 3] -> ([9, Cout, Cin] uint8 e4m3, [Cout] f32
+    scale) for the fp8-compute kernel (absmax/448 per out-channel,
+    utils/fp8.py convention)."""
+    from ..utils.fp8 import quantize_per_channel
+    Cout, Cin, _, _ = w.shape
+    q, scale = quantize_per_channel(w, dim=0)
+    qp = q.view(torch.uint8).permute(2, 3, 0, 1) \
+        .reshape(9, Cout, Cin).contiguous()
+    return qp, scale.reshape(-1).float()
+
+
+def quantize_activations_fp8(x: torch.Tensor):
+    """Per-tensor e4m3 quant of a channels_last activation — all on
+    GPU, no host sync. Returns (uint8 view NHWC, [1] f32 scale)."""
+    from ..utils.fp8 import FP8_DTYPE, FP8_MAX
+    xf = x.float()
+    scale = (xf.abs().amax() / FP8_MAX).clamp(min=1e-12).reshape(1)
+    q = (xf / scale).clamp(-FP8_MAX, FP8_MAX).to(FP8_DTYPE) \
+        .contiguous(memory_format=torch.channels_last)
+    return q.view(torch.uint8), scale
+
+
+def conv3x3_fp8(x, w_packed_u8, w_scale, bias, relu=True):
+    """fp8-compute 3x3 conv: quantize activations per-tensor on the
+    fly, run e4m3 x e4m3 MFMA, dequant in the epilogue. Inference-only
+    (no autograd). x bf16/f32 channels_last; weights from
+    pack_weight_fp8."""
+    from ._ext import require_ext
+    ext = require_ext()
+    xq, a_scale = quantize_activations_fp8(x)
+    zb = bias.float() if bias is not None else \
+        torch.zeros(w_packed_u8.shape[1], device=x.device)
+    return ext.conv3x3_fp8(xq, w_packed_u8, zb, w_scale, a_scale, relu)
+
+
 def rotate_for_dgrad(w: torch.Tensor) -> torch.Tensor:
     """Weight transform making bwd-data a FORWARD conv:
     dgrad(x) = conv3x3(grad_out, w') with

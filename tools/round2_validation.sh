@@ -9,7 +9,7 @@ cd "${GRAFT_REPO_ROOT:-/root/repo}"
 mkdir -p gpurun_out
 
 case "${1:-block1}" in
-block1)  # v3 kernel validation + conv A/B (~4 min)
+block1)  # v3 + wrw + fp8 kernel validation + conv A/B (~5 min)
   timeout 400 python -m pytest tests/test_conv3x3.py -m gpu -q \
       --runxfail > gpurun_out/r2_conv_tests.log 2>&1
   tail -4 gpurun_out/r2_conv_tests.log
