@@ -38,7 +38,7 @@ def build_config(args):
 
     class BenchConfig(Config):
         NAME = "coco_bench"
-        BACKBONE = "resnet101"
+        BACKBONE = args.backbone
         NUM_CLASSES = 81
         IMAGES_PER_GPU = args.batch
         GPU_COUNT = 1  # one process per GPU; DP handled by DDP wrapper
@@ -186,6 +186,10 @@ def main():
     ap.add_argument("--batch", type=int, default=2)
     ap.add_argument("--image-size", type=int, default=1024)
     ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--backbone", default="resnet101",
+                    choices=["resnet101", "resnet50"],
+                    help="resnet50 only for the CPU/CI smoke path; the "
+                         "headline config is resnet101 (BASELINE.json)")
     ap.add_argument("--backend", default=None,
                     help="override dist backend (nccl|gloo)")
     ap.add_argument("--phase-timing", action="store_true",

@@ -65,3 +65,33 @@ def test_maskrcnn_ddp_step_two_ranks():
     assert res[0]["grad_sum"] == pytest.approx(res[1]["grad_sum"],
                                                rel=1e-5)
     assert res[0]["grad_sum"] > 0
+
+
+def test_bench_script_two_proc_gloo(tmp_path):
+    """End-to-end insurance for the driver's SCALE run: launch bench.py
+    exactly as the driver does (torch.distributed.run, one proc per
+    'GPU') on CPU/gloo with a shrunk config and validate the rank-0
+    JSON contract line."""
+    import json
+    import subprocess
+    import sys
+
+    port = _free_port()
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", str(port), "bench.py", "--gpus", "2",
+           "--steps", "1", "--warmup", "0", "--batch", "1",
+           "--image-size", "256", "--backbone", "resnet50"]
+    proc = subprocess.run(
+        cmd, capture_output=True, text=True, timeout=900,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    json_lines = [ln for ln in proc.stdout.splitlines()
+                  if ln.startswith("{")]
+    assert len(json_lines) == 1, proc.stdout[-1000:]
+    rec = json.loads(json_lines[0])
+    assert rec["n_gpus"] == 2
+    assert rec["metric"] == "train_images_per_sec"
+    assert rec["value"] > 0
+    assert rec["config"]["parallelism"] == "dp2"
+    assert rec["config"]["global_batch"] == 2
