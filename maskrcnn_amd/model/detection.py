@@ -17,7 +17,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ..ops import batched_nms
+from ..ops import batched_nms, batched_nms_padded
 from ..utils.boxes import apply_box_deltas_t, norm_boxes_t
 from ..utils.meta import parse_image_meta_t
 
@@ -75,6 +75,9 @@ class DetectionLayer(nn.Module):
         keep = class_ids > 0
         if cfg.DETECTION_MIN_CONFIDENCE:
             keep &= class_scores >= cfg.DETECTION_MIN_CONFIDENCE
+        if _use_syncfree(rois):
+            return self._forward_syncfree(refined, class_ids,
+                                          class_scores, keep, B, N, C)
         flat_ix = keep.reshape(-1).nonzero(as_tuple=True)[0]
         D = cfg.DETECTION_MAX_INSTANCES
         out = rois.new_zeros((B, D, 6), dtype=torch.float32)
@@ -101,3 +104,61 @@ class DetectionLayer(nn.Module):
                           scores_f[nms_keep, None]], dim=1)
         out[kimg[sel], rank[sel]] = rows[sel]
         return out
+
+    def _forward_syncfree(self, refined, class_ids, class_scores, keep,
+                          B, N, C):
+        """Fixed-shape, zero-host-sync detection post-process
+        (hipGraph-capture path; auto-selected while capturing or via
+        MASKRCNN_SYNCFREE_DETECTION=1).
+
+        Dropped boxes keep their slot with score 0 (a zero-score box can
+        never suppress a positive-score one, so NMS results match the
+        gathered eager path); padded NMS + masked index_put_ scatter
+        into a dummy-row buffer replaces nonzero()/boolean gathers —
+        the same pattern as ProposalLayer."""
+        cfg = self.config
+        device = refined.device
+        D = cfg.DETECTION_MAX_INSTANCES
+        scores_m = torch.where(keep, class_scores,
+                               torch.zeros_like(class_scores))
+        out_buf = refined.new_zeros((B + 1, D + 1, 6))
+        group_sz = max(1, 16384 // N)
+        for g0 in range(0, B, group_sz):
+            g1 = min(g0 + group_sz, B)
+            nb = g1 - g0
+            flat_boxes = refined[g0:g1].reshape(-1, 4)
+            flat_scores = scores_m[g0:g1].reshape(-1)
+            flat_cls = class_ids[g0:g1].reshape(-1)
+            img_ix = torch.arange(nb, device=device) \
+                .repeat_interleave(N)
+            grp = img_ix * (C + 1) + flat_cls
+            keep_idx, _count = batched_nms_padded(
+                flat_boxes, flat_scores, grp,
+                cfg.DETECTION_NMS_THRESHOLD, nb * N)
+            valid = keep_idx >= 0
+            kc = keep_idx.clamp(min=0)
+            valid &= flat_scores[kc] > 0      # drop zero-score slots
+            img = img_ix[kc]                  # score-desc global order
+            onehot = (img.unsqueeze(1) ==
+                      torch.arange(nb, device=device).unsqueeze(0)) \
+                & valid.unsqueeze(1)
+            rank = onehot.long().cumsum(dim=0)[
+                torch.arange(img.numel(), device=device), img] - 1
+            sel = valid & (rank < D)
+            tgt_img = torch.where(sel, g0 + img, torch.full_like(img, B))
+            tgt_rank = torch.where(sel, rank, torch.full_like(rank, D))
+            rows = torch.cat(
+                [flat_boxes[kc], flat_cls[kc, None].float(),
+                 flat_scores[kc, None]], dim=1)
+            out_buf.index_put_((tgt_img, tgt_rank), rows)
+        return out_buf[:B, :D]
+
+
+def _use_syncfree(t: torch.Tensor) -> bool:
+    import os
+    flag = os.environ.get("MASKRCNN_SYNCFREE_DETECTION", "")
+    if flag == "1":
+        return True
+    if flag == "0":
+        return False
+    return t.is_cuda and torch.cuda.is_current_stream_capturing()

@@ -141,3 +141,56 @@ def test_classifier_head_linear_equals_conv(shapes_config):
                  head.mrcnn_class_conv2.bias)
     y = F.relu(head._bn_vec(head.mrcnn_class_bn2, y))
     torch.testing.assert_close(y, ref, atol=1e-4, rtol=1e-4)
+
+
+def test_detection_syncfree_matches_eager(monkeypatch):
+    """The fixed-shape capture-safe DetectionLayer path must produce the
+    same detections as the nonzero/gather eager path."""
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model.detection import DetectionLayer
+    from maskrcnn_amd.utils.meta import compose_image_meta
+
+    class C(ShapesConfig):
+        DETECTION_MIN_CONFIDENCE = 0.3
+        DETECTION_MAX_INSTANCES = 8
+
+    cfg = C()
+    torch.manual_seed(3)
+    layer = DetectionLayer(cfg)
+    B, N, Ccls = 2, 60, cfg.NUM_CLASSES
+    rois = torch.rand(B, N, 4)
+    rois[..., 2:] = rois[..., :2] * 0.5 + 0.4
+    probs = torch.softmax(torch.randn(B, N, Ccls) * 2, dim=-1)
+    deltas = torch.randn(B, N, Ccls, 4) * 0.1
+    metas = torch.stack([
+        torch.from_numpy(compose_image_meta(
+            i, (128, 128, 3), (128, 128, 3), (0, 0, 128, 128), 1.0,
+            np.ones(Ccls))).float()
+        for i in range(B)])
+
+    monkeypatch.setenv("MASKRCNN_SYNCFREE_DETECTION", "0")
+    eager = layer(rois, probs, deltas, metas)
+    monkeypatch.setenv("MASKRCNN_SYNCFREE_DETECTION", "1")
+    syncfree = layer(rois, probs, deltas, metas)
+    for b in range(B):
+        ne = int((eager[b][:, 5] > 0).sum())
+        ns = int((syncfree[b][:, 5] > 0).sum())
+        assert ne == ns
+        torch.testing.assert_close(
+            syncfree[b][:ns].sort(dim=0).values,
+            eager[b][:ne].sort(dim=0).values, atol=1e-6, rtol=1e-6)
+
+
+def test_roialign_syncfree_matches_eager(monkeypatch):
+    from maskrcnn_amd.model.roialign import PyramidROIAlign
+    torch.manual_seed(0)
+    layer = PyramidROIAlign(7, [1024, 1024, 3])
+    feats = [torch.randn(2, 8, s, s) for s in (256, 128, 64, 32)]
+    rois = torch.rand(2, 12, 4)
+    rois[..., 2:] = rois[..., :2] + torch.rand(2, 12, 2) * 0.5
+    rois[:, -2:] = 0  # padding rows
+    monkeypatch.setenv("MASKRCNN_SYNCFREE_ROI", "0")
+    eager = layer(rois, feats)
+    monkeypatch.setenv("MASKRCNN_SYNCFREE_ROI", "1")
+    syncfree = layer(rois, feats)
+    torch.testing.assert_close(syncfree, eager, atol=1e-5, rtol=1e-5)
