@@ -109,17 +109,23 @@ class InferenceEngine:
         static_img = images.clone()
         static_meta = metas.clone()
         static_anc = anchors.clone()
+        # cache_enabled=False: autocast's weight-cast cache must not be
+        # captured (stale casts would be baked into the graph — the
+        # torch.cuda.graph + autocast documented requirement).
+        ac = lambda: torch.autocast(  # noqa: E731
+            device_type="cuda", dtype=torch.bfloat16,
+            enabled=self.use_bf16, cache_enabled=False)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(3):
-                with self._autocast():
+                with ac():
                     self.model(static_img, static_meta, static_anc,
                                mode="inference")
         torch.cuda.current_stream().wait_stream(s)
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
-            with self._autocast():
+            with ac():
                 out = self.model(static_img, static_meta, static_anc,
                                  mode="inference")
         st = {"graph": g, "img": static_img, "meta": static_meta,
