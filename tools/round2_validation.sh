@@ -40,6 +40,11 @@ block3)  # serving bench (~8 min incl. 265 s batch-32 find)
       > gpurun_out/r2_serving.json 2> gpurun_out/r2_serving.err
   cat gpurun_out/r2_serving.json
   grep -E "WARNING|batch of" gpurun_out/r2_serving.err | tail -5
+  # A/B: hipGraph replay of the engine forward (validated in block1)
+  MASKRCNN_SERVING_HIPGRAPH=1 timeout 700 python bench/serving_bench.py \
+      --requests 96 --client-procs 4 --client-threads 8 --warmup 16 \
+      > gpurun_out/r2_serving_graph.json 2>/dev/null
+  cat gpurun_out/r2_serving_graph.json
   ;;
 block4)  # steady-state rocprof of the tuned step (profile AFTER warmup
          # by profiling a second process on the warm box: find-db is
@@ -74,4 +79,10 @@ if [ "${1:-}" = "block6" ]; then
       --image-size 1344 --weight-dtype fp8 --max-batch 16 \
       > gpurun_out/r2_fp8_serving.json 2> gpurun_out/r2_fp8_serving.err
   cat gpurun_out/r2_fp8_serving.json
+  # A/B: fp8 COMPUTE in the mask head (kernel validated in block1)
+  MASKRCNN_FP8_COMPUTE=1 timeout 800 python bench/serving_bench.py \
+      --requests 48 --client-procs 4 --client-threads 8 --warmup 8 \
+      --image-size 1344 --weight-dtype fp8 --max-batch 16 \
+      > gpurun_out/r2_fp8c_serving.json 2>/dev/null
+  cat gpurun_out/r2_fp8c_serving.json
 fi
