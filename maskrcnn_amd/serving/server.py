@@ -152,11 +152,13 @@ class ModelServer:
                                  "model_configs required")
             model_configs = [{"name": model_name,
                               "base_path": model_base_path}]
+        self._model_kwargs = dict(
+            device=device, devices=devices,
+            max_batch_size=max_batch_size,
+            batch_timeout_micros=batch_timeout_micros, config=config)
         self.models: Dict[str, ModelState] = {
-            mc["name"]: ModelState(
-                mc["name"], mc["base_path"], device=device,
-                devices=devices, max_batch_size=max_batch_size,
-                batch_timeout_micros=batch_timeout_micros, config=config)
+            mc["name"]: ModelState(mc["name"], mc["base_path"],
+                                   **self._model_kwargs)
             for mc in model_configs}
         self.grpc_port = grpc_port
         self.rest_port = rest_port
@@ -179,6 +181,30 @@ class ModelServer:
         if st is None:
             raise KeyError(f"unknown model {name!r}")
         return st
+
+    def reload_config(self, model_configs) -> None:
+        """Apply a new served-model set (the ModelService
+        HandleReloadConfigRequest semantics): load additions, keep
+        matching entries, unload (close) removed models."""
+        new: Dict[str, ModelState] = {}
+        for mc in model_configs:
+            cur = self.models.get(mc["name"])
+            if cur is not None and cur.base_path == mc["base_path"]:
+                new[mc["name"]] = cur
+            else:
+                st = ModelState(mc["name"], mc["base_path"],
+                                **self._model_kwargs)
+                st.maybe_reload()
+                if st.version is None:
+                    raise FileNotFoundError(
+                        f"no servable version under {mc['base_path']}")
+                new[mc["name"]] = st
+        kept = set(map(id, new.values()))
+        removed = [s for s in self.models.values() if id(s) not in kept]
+        self.models = new
+        for st in removed:
+            for b in st.batchers:
+                b.close()
 
     # ----------------------------------------------------------- predict
     def _handle_predict(self, request: PredictRequest) -> PredictResponse:
@@ -260,6 +286,33 @@ class ModelServer:
                                      version=state.version),
                 signature_defs=sigs)
 
+        from .wireproto import (GetModelStatusRequest,
+                                GetModelStatusResponse, ModelVersionStatus,
+                                ReloadConfigRequest, ReloadConfigResponse,
+                                StatusProto, STATE_AVAILABLE)
+
+        def get_model_status(request: GetModelStatusRequest, context):
+            try:
+                state = outer._get_state(request.model_spec.name)
+            except KeyError as e:
+                context.abort(grpc.StatusCode.NOT_FOUND, str(e))
+            return GetModelStatusResponse(
+                model_version_status=[ModelVersionStatus(
+                    version=state.version or 0,
+                    state=STATE_AVAILABLE if state.version is not None
+                    else 0)])
+
+        def handle_reload_config(request: ReloadConfigRequest, context):
+            try:
+                outer.reload_config(
+                    [{"name": c.name, "base_path": c.base_path}
+                     for c in request.model_configs])
+                return ReloadConfigResponse(status=StatusProto(0, ""))
+            except Exception as e:  # noqa: BLE001
+                log.exception("reload config failed")
+                return ReloadConfigResponse(
+                    status=StatusProto(13, str(e)))  # 13 = INTERNAL
+
         handler = grpc.method_handlers_generic_handler(
             "tensorflow.serving.PredictionService",
             {"Predict": grpc.unary_unary_rpc_method_handler(
@@ -270,11 +323,22 @@ class ModelServer:
                 get_model_metadata,
                 request_deserializer=GetModelMetadataRequest.decode,
                 response_serializer=lambda r: r.encode())})
+        model_svc = grpc.method_handlers_generic_handler(
+            "tensorflow.serving.ModelService",
+            {"GetModelStatus": grpc.unary_unary_rpc_method_handler(
+                get_model_status,
+                request_deserializer=GetModelStatusRequest.decode,
+                response_serializer=lambda r: r.encode()),
+             "HandleReloadConfigRequest":
+                 grpc.unary_unary_rpc_method_handler(
+                     handle_reload_config,
+                     request_deserializer=ReloadConfigRequest.decode,
+                     response_serializer=lambda r: r.encode())})
         server = grpc.server(
             cf.ThreadPoolExecutor(max_workers=16),
             options=[("grpc.max_receive_message_length", 256 * 1024 * 1024),
                      ("grpc.max_send_message_length", 256 * 1024 * 1024)])
-        server.add_generic_rpc_handlers((handler,))
+        server.add_generic_rpc_handlers((handler, model_svc))
         bound = server.add_insecure_port(f"[::]:{self.grpc_port}")
         self.grpc_port = bound
         return server

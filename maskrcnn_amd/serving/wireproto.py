@@ -603,3 +603,201 @@ def decode_saved_model(buf: bytes):
                     k, sv = _decode_map_entry(v2)
                     sigs[k] = SignatureDef.decode(sv)
     return version, tags, sigs
+
+
+# ------------------------------------------- ModelService (model mgmt)
+# tensorflow_serving/apis/get_model_status.proto +
+# model_management.proto + config/model_server_config.proto field
+# numbers; the upstream gRPC service is
+# /tensorflow.serving.ModelService/{GetModelStatus,HandleReloadConfigRequest}.
+
+#: ModelVersionStatus.State enum values (get_model_status.proto).
+STATE_UNKNOWN, STATE_START, STATE_LOADING = 0, 10, 20
+STATE_AVAILABLE, STATE_UNLOADING, STATE_END = 30, 40, 50
+
+
+@dataclass
+class StatusProto:
+    """tensorflow.error-style status: 1 error_code, 2 error_message."""
+    error_code: int = 0
+    error_message: str = ""
+
+    def encode(self) -> bytes:
+        out = bytearray()
+        _w_varint_field(out, 1, self.error_code)
+        _w_string_field(out, 2, self.error_message)
+        return bytes(out)
+
+    @classmethod
+    def decode(cls, buf: bytes) -> "StatusProto":
+        m = cls()
+        pos = 0
+        while pos < len(buf):
+            fno, wt, v, pos = _r_field(buf, pos)
+            if fno == 1 and wt == _WT_VARINT:
+                m.error_code = v
+            elif fno == 2 and wt == _WT_LEN:
+                m.error_message = v.decode("utf-8")
+        return m
+
+
+@dataclass
+class ModelVersionStatus:
+    """1 version (int64), 2 state (enum), 3 status."""
+    version: int = 0
+    state: int = STATE_UNKNOWN
+    status: StatusProto = field(default_factory=StatusProto)
+
+    def encode(self) -> bytes:
+        out = bytearray()
+        _w_varint_field(out, 1, _zigzag_i64(self.version))
+        _w_varint_field(out, 2, self.state)
+        _w_len_field(out, 3, self.status.encode())
+        return bytes(out)
+
+    @classmethod
+    def decode(cls, buf: bytes) -> "ModelVersionStatus":
+        m = cls()
+        pos = 0
+        while pos < len(buf):
+            fno, wt, v, pos = _r_field(buf, pos)
+            if fno == 1 and wt == _WT_VARINT:
+                m.version = v if v < (1 << 63) else v - (1 << 64)
+            elif fno == 2 and wt == _WT_VARINT:
+                m.state = v
+            elif fno == 3 and wt == _WT_LEN:
+                m.status = StatusProto.decode(v)
+        return m
+
+
+@dataclass
+class GetModelStatusRequest:
+    """1 model_spec."""
+    model_spec: ModelSpec = field(default_factory=ModelSpec)
+
+    def encode(self) -> bytes:
+        out = bytearray()
+        _w_len_field(out, 1, self.model_spec.encode())
+        return bytes(out)
+
+    @classmethod
+    def decode(cls, buf: bytes) -> "GetModelStatusRequest":
+        m = cls()
+        pos = 0
+        while pos < len(buf):
+            fno, wt, v, pos = _r_field(buf, pos)
+            if fno == 1 and wt == _WT_LEN:
+                m.model_spec = ModelSpec.decode(v)
+        return m
+
+
+@dataclass
+class GetModelStatusResponse:
+    """1 repeated model_version_status."""
+    model_version_status: List[ModelVersionStatus] = \
+        field(default_factory=list)
+
+    def encode(self) -> bytes:
+        out = bytearray()
+        for s in self.model_version_status:
+            _w_len_field(out, 1, s.encode())
+        return bytes(out)
+
+    @classmethod
+    def decode(cls, buf: bytes) -> "GetModelStatusResponse":
+        m = cls()
+        pos = 0
+        while pos < len(buf):
+            fno, wt, v, pos = _r_field(buf, pos)
+            if fno == 1 and wt == _WT_LEN:
+                m.model_version_status.append(
+                    ModelVersionStatus.decode(v))
+        return m
+
+
+@dataclass
+class ModelConfigProto:
+    """model_server_config.proto ModelConfig: 1 name, 2 base_path,
+    4 model_platform."""
+    name: str = ""
+    base_path: str = ""
+    model_platform: str = "tensorflow"
+
+    def encode(self) -> bytes:
+        out = bytearray()
+        _w_string_field(out, 1, self.name)
+        _w_string_field(out, 2, self.base_path)
+        _w_string_field(out, 4, self.model_platform)
+        return bytes(out)
+
+    @classmethod
+    def decode(cls, buf: bytes) -> "ModelConfigProto":
+        m = cls()
+        pos = 0
+        while pos < len(buf):
+            fno, wt, v, pos = _r_field(buf, pos)
+            if fno == 1 and wt == _WT_LEN:
+                m.name = v.decode("utf-8")
+            elif fno == 2 and wt == _WT_LEN:
+                m.base_path = v.decode("utf-8")
+            elif fno == 4 and wt == _WT_LEN:
+                m.model_platform = v.decode("utf-8")
+        return m
+
+
+@dataclass
+class ReloadConfigRequest:
+    """model_management.proto ReloadConfigRequest: 1 config
+    (ModelServerConfig: 1 model_config_list (ModelConfigList:
+    1 repeated config))."""
+    model_configs: List[ModelConfigProto] = field(default_factory=list)
+
+    def encode(self) -> bytes:
+        lst = bytearray()
+        for c in self.model_configs:
+            _w_len_field(lst, 1, c.encode())
+        cfg = bytearray()
+        _w_len_field(cfg, 1, bytes(lst))
+        out = bytearray()
+        _w_len_field(out, 1, bytes(cfg))
+        return bytes(out)
+
+    @classmethod
+    def decode(cls, buf: bytes) -> "ReloadConfigRequest":
+        m = cls()
+        pos = 0
+        while pos < len(buf):
+            fno, wt, v, pos = _r_field(buf, pos)
+            if fno == 1 and wt == _WT_LEN:
+                p = 0
+                while p < len(v):
+                    f2, w2, v2, p = _r_field(v, p)
+                    if f2 == 1 and w2 == _WT_LEN:
+                        pp = 0
+                        while pp < len(v2):
+                            f3, w3, v3, pp = _r_field(v2, pp)
+                            if f3 == 1 and w3 == _WT_LEN:
+                                m.model_configs.append(
+                                    ModelConfigProto.decode(v3))
+        return m
+
+
+@dataclass
+class ReloadConfigResponse:
+    """1 status."""
+    status: StatusProto = field(default_factory=StatusProto)
+
+    def encode(self) -> bytes:
+        out = bytearray()
+        _w_len_field(out, 1, self.status.encode())
+        return bytes(out)
+
+    @classmethod
+    def decode(cls, buf: bytes) -> "ReloadConfigResponse":
+        m = cls()
+        pos = 0
+        while pos < len(buf):
+            fno, wt, v, pos = _r_field(buf, pos)
+            if fno == 1 and wt == _WT_LEN:
+                m.status = StatusProto.decode(v)
+        return m

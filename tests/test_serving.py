@@ -494,3 +494,64 @@ def test_demo_app_roundtrip(shapes_config, tmp_path):
     assert Image.open(_io.BytesIO(r.content)).size[0] > 0
     r = client.post("/detect", content=b"not an image")
     assert r.status_code == 400
+
+
+def test_grpc_model_service_status_and_reload(served_model, tmp_path):
+    """tensorflow.serving.ModelService: GetModelStatus over the wire
+    and HandleReloadConfigRequest adding a second model at runtime."""
+    import grpc
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+    from maskrcnn_amd.serving.wireproto import (
+        GetModelStatusRequest, GetModelStatusResponse, ModelConfigProto,
+        ModelSpec, ReloadConfigRequest, ReloadConfigResponse,
+        STATE_AVAILABLE)
+
+    server, cfg, m = served_model
+    channel = grpc.insecure_channel(f"localhost:{server.grpc_port}")
+    status_call = channel.unary_unary(
+        "/tensorflow.serving.ModelService/GetModelStatus",
+        request_serializer=lambda r: r.encode(),
+        response_deserializer=GetModelStatusResponse.decode)
+    st = status_call(GetModelStatusRequest(
+        model_spec=ModelSpec(name="mask_rcnn")), timeout=10)
+    assert st.model_version_status[0].version == server.state.version
+    assert st.model_version_status[0].state == STATE_AVAILABLE
+
+    with pytest.raises(grpc.RpcError) as e:
+        status_call(GetModelStatusRequest(
+            model_spec=ModelSpec(name="nope")), timeout=10)
+    assert e.value.code() == grpc.StatusCode.NOT_FOUND
+
+    # Reload: keep mask_rcnn, add second.
+    base2 = str(tmp_path / "models" / "second")
+    export_saved_model(m, cfg, base2, version=7)
+    reload_call = channel.unary_unary(
+        "/tensorflow.serving.ModelService/HandleReloadConfigRequest",
+        request_serializer=lambda r: r.encode(),
+        response_deserializer=ReloadConfigResponse.decode)
+    first_base = server.state.base_path
+    resp = reload_call(ReloadConfigRequest(model_configs=[
+        ModelConfigProto(name="mask_rcnn", base_path=first_base),
+        ModelConfigProto(name="second", base_path=base2)]), timeout=60)
+    assert resp.status.error_code == 0, resp.status.error_message
+    assert set(server.models) == {"mask_rcnn", "second"}
+    st2 = status_call(GetModelStatusRequest(
+        model_spec=ModelSpec(name="second")), timeout=10)
+    assert st2.model_version_status[0].version == 7
+
+    # Reload with a bad path -> INTERNAL status, serving set unchanged.
+    resp = reload_call(ReloadConfigRequest(model_configs=[
+        ModelConfigProto(name="mask_rcnn", base_path=first_base),
+        ModelConfigProto(name="ghost", base_path="/nonexistent")]),
+        timeout=60)
+    assert resp.status.error_code == 13
+    assert set(server.models) == {"mask_rcnn", "second"}
+    # Drop back to the single model (unloads 'second').
+    resp = reload_call(ReloadConfigRequest(model_configs=[
+        ModelConfigProto(name="mask_rcnn", base_path=first_base)]),
+        timeout=60)
+    assert resp.status.error_code == 0
+    assert set(server.models) == {"mask_rcnn"}
+    channel.close()
