@@ -135,3 +135,49 @@ class RestServingClient(ServingClientBase):
         out = r.json()["outputs"]
         return {k: np.asarray(v, dtype=np.float32)
                 for k, v in out.items()}
+
+
+class ModelServiceClient:
+    """tensorflow.serving.ModelService client: model status + runtime
+    config reload (the tensorflow_serving.apis model_service_pb2_grpc
+    surface, spoken over the hand-vendored wire protos)."""
+
+    def __init__(self, target: str = "localhost:8500",
+                 timeout: float = 30.0):
+        import grpc
+
+        from .wireproto import (GetModelStatusResponse,
+                                ReloadConfigResponse)
+        self.timeout = timeout
+        self.channel = grpc.insecure_channel(target)
+        self._status = self.channel.unary_unary(
+            "/tensorflow.serving.ModelService/GetModelStatus",
+            request_serializer=lambda r: r.encode(),
+            response_deserializer=GetModelStatusResponse.decode)
+        self._reload = self.channel.unary_unary(
+            "/tensorflow.serving.ModelService/HandleReloadConfigRequest",
+            request_serializer=lambda r: r.encode(),
+            response_deserializer=ReloadConfigResponse.decode)
+
+    def get_model_status(self, model_name: str):
+        """Returns a list of {'version', 'state'} dicts."""
+        from .wireproto import GetModelStatusRequest, ModelSpec
+        resp = self._status(GetModelStatusRequest(
+            model_spec=ModelSpec(name=model_name)), timeout=self.timeout)
+        return [{"version": s.version, "state": s.state}
+                for s in resp.model_version_status]
+
+    def reload_config(self, model_configs):
+        """model_configs: [{'name', 'base_path'}, ...]. Raises
+        RuntimeError on a rejected reload."""
+        from .wireproto import ModelConfigProto, ReloadConfigRequest
+        resp = self._reload(ReloadConfigRequest(model_configs=[
+            ModelConfigProto(name=c["name"], base_path=c["base_path"])
+            for c in model_configs]), timeout=self.timeout)
+        if resp.status.error_code != 0:
+            raise RuntimeError(
+                f"reload rejected ({resp.status.error_code}): "
+                f"{resp.status.error_message}")
+
+    def close(self):
+        self.channel.close()

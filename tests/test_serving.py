@@ -555,3 +555,15 @@ def test_grpc_model_service_status_and_reload(served_model, tmp_path):
     assert resp.status.error_code == 0
     assert set(server.models) == {"mask_rcnn"}
     channel.close()
+
+
+def test_model_service_client(served_model):
+    from maskrcnn_amd.serving.client import ModelServiceClient
+    from maskrcnn_amd.serving.wireproto import STATE_AVAILABLE
+    server, cfg, m = served_model
+    c = ModelServiceClient(f"localhost:{server.grpc_port}")
+    st = c.get_model_status("mask_rcnn")
+    assert st[0]["state"] == STATE_AVAILABLE
+    with pytest.raises(RuntimeError, match="reload rejected"):
+        c.reload_config([{"name": "x", "base_path": "/nonexistent"}])
+    c.close()

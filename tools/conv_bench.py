@@ -95,5 +95,60 @@ def main():
               f"{'':8} {ref / ours:6.2f}x")
 
 
+def bench_wrw():
+    """wrw kernel vs aten.convolution_backward weight grad."""
+    from maskrcnn_amd.ops._ext import require_ext
+    ext = require_ext()
+    dev = torch.device("cuda:0")
+    print("\nconv3x3 wrw (weight grad):")
+    for (B, Cin, H, W, Cout) in [(2, 64, 256, 256, 64),
+                                 (2, 128, 128, 128, 128),
+                                 (2, 256, 64, 64, 256),
+                                 (2, 512, 32, 32, 512),
+                                 (2, 256, 256, 256, 256),
+                                 (400, 256, 14, 14, 256)]:
+        x = torch.randn(B, Cin, H, W, device=dev).bfloat16() \
+            .to(memory_format=torch.channels_last)
+        dy = torch.randn(B, Cout, H, W, device=dev).bfloat16() \
+            .to(memory_format=torch.channels_last)
+        w = torch.randn(Cout, Cin, 3, 3, device=dev).bfloat16()
+        ours = timeit(lambda: ext.conv3x3_wrw(x, dy))
+        ref = timeit(lambda: torch.ops.aten.convolution_backward(
+            dy, x, w, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+            [False, True, False])[1])
+        flops = 2.0 * B * H * W * Cout * Cin * 9
+        print(f"{str((B, Cin, H, W, Cout)):>28} "
+              f"ours={ours:7.3f}ms({flops / (ours * 1e-3) / 1e12:5.0f}TF)"
+              f" aten={ref:7.3f}ms {ref / ours:5.2f}x")
+
+
+def bench_fp8():
+    """fp8-compute conv (incl. on-the-fly act quant) vs bf16 v2."""
+    from maskrcnn_amd.ops._ext import require_ext
+    from maskrcnn_amd.ops.conv3x3 import (conv3x3_fp8, pack_weight,
+                                          pack_weight_fp8)
+    ext = require_ext()
+    dev = torch.device("cuda:0")
+    print("\nconv3x3 fp8 e4m3 (incl. activation quant) vs bf16 v2:")
+    for (B, Cin, H, W, Cout) in [(2, 256, 64, 64, 256),
+                                 (2, 256, 256, 256, 256),
+                                 (400, 256, 14, 14, 256),
+                                 (16, 256, 84, 84, 256)]:  # 1344/16 maps
+        x = torch.randn(B, Cin, H, W, device=dev).bfloat16() \
+            .to(memory_format=torch.channels_last)
+        w = (torch.randn(Cout, Cin, 3, 3, device=dev) * 0.05)
+        b = torch.randn(Cout, device=dev)
+        qp, scale = pack_weight_fp8(w)
+        wp = pack_weight(w)
+        f8 = timeit(lambda: conv3x3_fp8(x, qp, scale, b, relu=True))
+        bf = timeit(lambda: ext.conv3x3_fwd(x, wp, b.float(), True, 2))
+        flops = 2.0 * B * H * W * Cout * Cin * 9
+        print(f"{str((B, Cin, H, W, Cout)):>28} "
+              f"fp8={f8:7.3f}ms({flops / (f8 * 1e-3) / 1e12:5.0f}TF) "
+              f"bf16v2={bf:7.3f}ms {bf / f8:5.2f}x")
+
+
 if __name__ == "__main__":
     main()
+    bench_wrw()
+    bench_fp8()
