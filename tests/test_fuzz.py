@@ -81,3 +81,26 @@ def test_nms_invariants(n, thresh, seed):
     ov = B.compute_overlaps(boxes[k], boxes[k])
     np.fill_diagonal(ov, 0)
     assert (ov <= thresh + 1e-5).all()
+
+
+@given(st.lists(st.tuples(st.text(min_size=1, max_size=20),
+                          st.text(min_size=1, max_size=40)),
+                min_size=0, max_size=5),
+       st.integers(min_value=0, max_value=2 ** 62),
+       st.sampled_from([0, 10, 20, 30, 40, 50]))
+@settings(max_examples=50, deadline=None)
+def test_model_service_protos_roundtrip(configs, version, state):
+    from maskrcnn_amd.serving.wireproto import (
+        GetModelStatusResponse, ModelConfigProto, ModelVersionStatus,
+        ReloadConfigRequest, StatusProto)
+    req = ReloadConfigRequest(model_configs=[
+        ModelConfigProto(name=n, base_path=p) for n, p in configs])
+    dec = ReloadConfigRequest.decode(req.encode())
+    assert [(c.name, c.base_path) for c in dec.model_configs] == configs
+    resp = GetModelStatusResponse(model_version_status=[
+        ModelVersionStatus(version=version, state=state,
+                           status=StatusProto(13, "err"))])
+    d2 = GetModelStatusResponse.decode(resp.encode())
+    assert d2.model_version_status[0].version == version
+    assert d2.model_version_status[0].state == state
+    assert d2.model_version_status[0].status.error_message == "err"
