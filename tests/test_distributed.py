@@ -95,3 +95,23 @@ def test_bench_script_two_proc_gloo(tmp_path):
     assert rec["value"] > 0
     assert rec["config"]["parallelism"] == "dp2"
     assert rec["config"]["global_batch"] == 2
+
+
+def test_train_cli_one_epoch(tmp_path):
+    """tools/train.py end to end on CPU (shapes, 2 steps, heads) —
+    writes an epoch checkpoint + metrics.csv + train_state.pth."""
+    import glob
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    proc = subprocess.run(
+        [sys.executable, "tools/train.py", "--dataset", "shapes",
+         "--epochs", "1", "--steps", "2", "--layers", "heads",
+         "--num-workers", "0", "--logs", str(tmp_path)],
+        capture_output=True, text=True, timeout=1200, cwd=root)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    ckpts = glob.glob(str(tmp_path / "*" / "mask_rcnn_*_0000.pth"))
+    assert len(ckpts) == 1, ckpts
+    assert glob.glob(str(tmp_path / "*" / "metrics.csv"))
+    assert glob.glob(str(tmp_path / "*" / "train_state.pth"))
