@@ -79,3 +79,49 @@ def fp8_bytes_saved(state: dict) -> int:
         if k.endswith("::fp8"):
             saved += v.numel() * 3  # fp32 -> 1 byte
     return saved
+
+
+# --------------------------------------------------------------- MX fp8
+# OCP MX block scaling for the gfx950 mfma_scale_*_f8f6f4 kernels
+# (round-2 compute path): 32 consecutive K elements share one e8m0
+# scale (a biased power-of-two exponent byte). The intrinsic probe
+# (NOTES_ROUND2.md) confirmed __builtin_amdgcn_mfma_scale_f32_
+# 16x16x128_f8f6f4 takes i32x8 operands (32 e4m3 bytes/lane) plus
+# per-lane scale ints.
+
+MX_BLOCK = 32
+_E4M3_EMAX = 8  # floor(log2(448))
+
+
+def mx_quantize(x: torch.Tensor, dim: int = -1):
+    """Block-32 MX quantization along `dim`.
+
+    Returns (q e4m3fn same shape, exp uint8 with dim shrunk by 32):
+    x ~= q.float() * 2^(exp - 127) broadcast over each block.
+    `dim`'s size must be a multiple of 32.
+    """
+    dim = dim % x.dim()
+    K = x.shape[dim]
+    assert K % MX_BLOCK == 0, f"dim size {K} not a multiple of 32"
+    xf = x.float().movedim(dim, -1)
+    shape = xf.shape
+    xb = xf.reshape(*shape[:-1], K // MX_BLOCK, MX_BLOCK)
+    amax = xb.abs().amax(dim=-1, keepdim=True).clamp(min=2.0 ** -126)
+    e = torch.floor(torch.log2(amax)) - _E4M3_EMAX
+    e = e.clamp(-127, 127)
+    scale = torch.pow(2.0, e)
+    q = (xb / scale).clamp(-FP8_MAX, FP8_MAX).to(FP8_DTYPE)
+    q = q.reshape(shape).movedim(-1, dim)
+    exp = (e.squeeze(-1) + 127).to(torch.uint8).movedim(-1, dim)
+    return q, exp
+
+
+def mx_dequantize(q: torch.Tensor, exp: torch.Tensor, dim: int = -1,
+                  dtype=torch.float32) -> torch.Tensor:
+    dim = dim % q.dim()
+    qf = q.float().movedim(dim, -1)
+    shape = qf.shape
+    qb = qf.reshape(*shape[:-1], shape[-1] // MX_BLOCK, MX_BLOCK)
+    e = exp.movedim(dim, -1).float().unsqueeze(-1) - 127.0
+    out = qb * torch.pow(2.0, e)
+    return out.reshape(shape).movedim(-1, dim).to(dtype)
