@@ -129,9 +129,11 @@ class _Conv3x3Fn(torch.autograd.Function):
             os.environ.get("MASKRCNN_HIP_WRW", "0") == "1")
         need_aten_gx = ctx.needs_input_grad[0] and not use_hip_dgrad
         need_aten_gw = ctx.needs_input_grad[1] and not use_hip_wrw
-        gx, gw, _ = torch.ops.aten.convolution_backward(
-            go_cl, x, wd, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
-            [need_aten_gx, need_aten_gw, False])
+        gx = gw = None
+        if need_aten_gx or need_aten_gw:
+            gx, gw, _ = torch.ops.aten.convolution_backward(
+                go_cl, x, wd, None, [1, 1], [1, 1], [1, 1], False,
+                [0, 0], 1, [need_aten_gx, need_aten_gw, False])
         if use_hip_wrw:
             # [9, Cout, Cin] tap-major fp32 -> [Cout, Cin, 3, 3].
             gw = ext.conv3x3_wrw(x, go_cl) \
