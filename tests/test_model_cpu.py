@@ -138,3 +138,30 @@ def test_optimizer_state_resume(shapes_config, shapes_dataset, tmp_path):
     tr2.fit(cfg.LEARNING_RATE, epochs=2, layers="heads",
             steps_per_epoch=1)
     assert tr2._resumed_opt
+
+
+def test_detect_pad64_nonsquare(shapes_config, tmp_path):
+    """End to end with IMAGE_RESIZE_MODE='pad64' (the nucleus-sample
+    mode): non-square input molds to a 64-multiple non-square shape;
+    anchors, meta and unmold must all follow."""
+    from maskrcnn_amd.model import MaskRCNN
+
+    class Pad64Config(type(shapes_config)):
+        IMAGE_RESIZE_MODE = "pad64"
+        IMAGE_MIN_DIM = 64
+        # IMAGE_SHAPE (anchors default / model assert) still derives
+        # from IMAGE_MAX_DIM, exactly like reference config.py; pad64
+        # molding itself ignores the max.
+        IMAGE_MAX_DIM = 256
+        IMAGE_MIN_SCALE = 0
+
+    cfg = Pad64Config()
+    m = MaskRCNN("inference", cfg, model_dir=str(tmp_path))
+    img = (np.random.rand(100, 160, 3) * 255).astype(np.uint8)
+    r = m.detect([img])[0]
+    assert r["masks"].shape[:2] == (100, 160)
+    assert r["rois"].shape[1] == 4
+    # Molded shape must be 64-multiples (192x256... wait 100->128, 160->192).
+    molded, _, windows = m.mold_inputs([img])
+    assert molded.shape[1] % 64 == 0 and molded.shape[2] % 64 == 0
+    assert molded.shape[1] != molded.shape[2]  # stayed non-square
