@@ -121,3 +121,30 @@ def test_engine_hipgraph_matches_eager(shapes_config, monkeypatch):
             np.testing.assert_allclose(o_g[k], o_e[k], atol=1e-3,
                                        rtol=1e-3)
     assert len(graphed._graphs) == 1
+
+
+@pytest.mark.gpu
+@pytest.mark.xfail(strict=False,
+                   reason="fp8 kernel pending first GPU validation")
+def test_mask_head_fp8_close_to_bf16(shapes_config, monkeypatch):
+    """MASKRCNN_FP8_COMPUTE routes the mask-head conv stack through the
+    e4m3 kernel; detections' masks must stay close to the bf16 path."""
+    import torch
+    from maskrcnn_amd.model.heads import FPNMaskHead
+
+    head = FPNMaskHead(shapes_config).cuda().eval() \
+        .to(memory_format=torch.channels_last)
+    rois = torch.rand(1, 8, 4, device="cuda")
+    rois[..., 2:] = rois[..., :2] * 0.5 + 0.45
+    feats = [torch.randn(1, 256, s, s, device="cuda").bfloat16()
+             .to(memory_format=torch.channels_last)
+             for s in (64, 32, 16, 8)]
+    with torch.no_grad(), torch.autocast("cuda", torch.bfloat16):
+        monkeypatch.setenv("MASKRCNN_FP8_COMPUTE", "0")
+        ref = head(rois, feats).float()
+        monkeypatch.setenv("MASKRCNN_FP8_COMPUTE", "1")
+        out = head(rois, feats).float()
+    assert head._fp8_cache is not None and len(head._fp8_cache) == 4
+    # Logit-space agreement: fp8 quant noise through 4 layers.
+    err = (out - ref).abs().mean()
+    assert float(err) < 0.15, float(err)
