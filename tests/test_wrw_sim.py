@@ -112,3 +112,20 @@ def test_wrw_sim_channel_tile_guards():
     got = torch.from_numpy(dw).reshape(3, 3, cout, cin) \
         .permute(2, 3, 0, 1)
     torch.testing.assert_close(got, wt.grad, atol=1e-3, rtol=1e-4)
+
+
+def test_xcd_swizzle_bijective():
+    """The v3 kernel's XCD-aware blockIdx remap (conv3x3_v3.hip) must be
+    a bijection for EVERY workgroup count — the naive (orig%8)*ceil+...
+    form silently drops/duplicates tiles when nwg%8 != 0 (guide §5)."""
+    NXCD = 8
+    for nwg in list(range(1, 100)) + [256, 768, 1000, 4095, 4096]:
+        seen = set()
+        for bid in range(nwg):
+            q, r = divmod(nwg, NXCD)
+            xcd, pos = bid % NXCD, bid // NXCD
+            out = (xcd * (q + 1) if xcd < r
+                   else r * (q + 1) + (xcd - r) * q) + pos
+            assert 0 <= out < nwg, (nwg, bid, out)
+            seen.add(out)
+        assert len(seen) == nwg, f"not bijective at nwg={nwg}"
