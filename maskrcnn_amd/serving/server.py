@@ -396,6 +396,11 @@ class ModelServer:
             if state is None:
                 return JSONResponse(
                     {"error": f"model {name} not found"}, status_code=404)
+            sig = body.get("signature_name", "serving_default")
+            if sig != "serving_default":
+                return JSONResponse(
+                    {"error": f"unknown signature {sig!r}"},
+                    status_code=400)
             try:
                 if "inputs" in body:       # columnar format
                     ins = body["inputs"]

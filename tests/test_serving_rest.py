@@ -114,3 +114,13 @@ def test_rest_predict_instances_format(rest_server):
     assert r.status_code == 200, r.text[:300]
     out = r.json()["outputs"]
     assert "mrcnn_detection" in out
+
+
+def test_rest_signature_name_validation(rest_server):
+    import requests
+    server, cfg, port = rest_server
+    r = requests.post(
+        f"http://127.0.0.1:{port}/v1/models/mask_rcnn:predict",
+        json={"signature_name": "bogus", "inputs": {}}, timeout=10)
+    assert r.status_code == 400
+    assert "signature" in r.json()["error"]
