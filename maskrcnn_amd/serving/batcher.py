@@ -55,22 +55,36 @@ class DynamicBatcher:
         `outputs` restricts which signature outputs are computed back to
         the host (None = all 7). Returns a Future resolving to the
         outputs dict (numpy)."""
+        if self._stop:
+            raise RuntimeError("batcher closed")
         fut: Future = Future()
         self.queue.put((input_image, input_image_meta, input_anchors,
                         outputs, fut))
         return fut
 
     def close(self):
+        """Graceful: stop accepting, drain what's queued (workers keep
+        processing until empty — TF-Serving unloads wait for in-flight
+        work), then fail any straggler enqueued during the race."""
         self._stop = True
         for t in self.threads:
-            t.join(timeout=2.0)
+            t.join(timeout=60.0)
+        while True:
+            try:
+                item = self.queue.get_nowait()
+            except queue.Empty:
+                break
+            if not item[-1].done():
+                item[-1].set_exception(RuntimeError("batcher closed"))
 
     # ------------------------------------------------------------ worker
     def _worker(self):
-        while not self._stop:
+        while True:
             try:
                 first = self.queue.get(timeout=0.1)
             except queue.Empty:
+                if self._stop:
+                    break
                 continue
             batch = [first]
             n = first[0].shape[0]

@@ -105,20 +105,33 @@ class ModelState:
         return True
 
     def predict(self, images, metas, anchors=None, outputs=None):
-        with self._lock:
-            if not self.batchers:
-                raise RuntimeError(
-                    f"model {self.name} has no loaded version")
-            # Shortest-queue replica (round-robin tiebreak).
-            qsizes = [b.queue.qsize() for b in self.batchers]
-            best = min(range(len(qsizes)),
-                       key=lambda i: (qsizes[i], (i - self._rr) %
-                                      len(qsizes)))
-            self._rr = (best + 1) % len(self.batchers)
-            batcher = self.batchers[best]
-            version = self.version
-        fut = batcher.submit(images, metas, anchors, outputs=outputs)
-        return fut.result(timeout=300.0), version
+        # Retry on the submit-vs-hot-swap race: maybe_reload() may close
+        # the batcher we just picked; re-pick from the fresh set.
+        for _attempt in range(3):
+            with self._lock:
+                if not self.batchers:
+                    raise RuntimeError(
+                        f"model {self.name} has no loaded version")
+                # Shortest-queue replica (round-robin tiebreak).
+                qsizes = [b.queue.qsize() for b in self.batchers]
+                best = min(range(len(qsizes)),
+                           key=lambda i: (qsizes[i], (i - self._rr) %
+                                          len(qsizes)))
+                self._rr = (best + 1) % len(self.batchers)
+                batcher = self.batchers[best]
+                version = self.version
+            try:
+                fut = batcher.submit(images, metas, anchors,
+                                     outputs=outputs)
+            except RuntimeError:
+                continue
+            try:
+                return fut.result(timeout=300.0), version
+            except RuntimeError as e:
+                if "batcher closed" in str(e):
+                    continue  # drained away by a reload: resubmit
+                raise
+        raise RuntimeError("predict failed: batcher kept closing")
 
     def warmup(self, image_shape=None):
         """Run one dummy max-batch predict per replica so MIOpen Find /
