@@ -104,3 +104,50 @@ def test_model_service_protos_roundtrip(configs, version, state):
     assert d2.model_version_status[0].version == version
     assert d2.model_version_status[0].state == state
     assert d2.model_version_status[0].status.error_message == "err"
+
+
+@given(st.integers(min_value=1, max_value=3),
+       st.integers(min_value=5, max_value=40),
+       st.integers(min_value=0, max_value=10 ** 6),
+       st.sampled_from([0.0, 0.3, 0.7]))
+@settings(max_examples=20, deadline=None)
+def test_detection_syncfree_equivalence_fuzz(B, N, seed, min_conf):
+    """Sync-free DetectionLayer == eager across random shapes/confs."""
+    import os
+
+    import torch
+
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model.detection import DetectionLayer
+    from maskrcnn_amd.utils.meta import compose_image_meta
+
+    class C(ShapesConfig):
+        DETECTION_MIN_CONFIDENCE = min_conf
+        DETECTION_MAX_INSTANCES = 6
+
+    cfg = C()
+    torch.manual_seed(seed)
+    layer = DetectionLayer(cfg)
+    Ccls = cfg.NUM_CLASSES
+    rois = torch.rand(B, N, 4)
+    rois[..., 2:] = rois[..., :2] * 0.5 + 0.4
+    probs = torch.softmax(torch.randn(B, N, Ccls) * 3, dim=-1)
+    deltas = torch.randn(B, N, Ccls, 4) * 0.1
+    metas = torch.stack([
+        torch.from_numpy(compose_image_meta(
+            i, (128, 128, 3), (128, 128, 3), (0, 0, 128, 128), 1.0,
+            np.ones(Ccls))).float() for i in range(B)])
+    try:
+        os.environ["MASKRCNN_SYNCFREE_DETECTION"] = "0"
+        eager = layer(rois, probs, deltas, metas)
+        os.environ["MASKRCNN_SYNCFREE_DETECTION"] = "1"
+        syncfree = layer(rois, probs, deltas, metas)
+    finally:
+        os.environ.pop("MASKRCNN_SYNCFREE_DETECTION", None)
+    for b in range(B):
+        ne = int((eager[b][:, 5] > 0).sum())
+        ns = int((syncfree[b][:, 5] > 0).sum())
+        assert ne == ns
+        torch.testing.assert_close(
+            syncfree[b][:ns].sort(dim=0).values,
+            eager[b][:ne].sort(dim=0).values, atol=1e-6, rtol=1e-6)
