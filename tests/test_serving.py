@@ -627,3 +627,69 @@ def test_concurrent_predicts_with_reload(tmp_path):
         time.sleep(0.25)
     assert server.state.version == 2
     server.stop()
+
+
+def test_server_cli_with_config_files(tmp_path):
+    """python -m maskrcnn_amd.serving.server with --model_config_file +
+    --batching_parameters_file: boots, serves REST status, dies clean."""
+    import signal
+    import socket
+    import subprocess
+    import sys
+
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+
+    class TinyConfig(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 100
+        POST_NMS_ROIS_INFERENCE = 10
+
+    cfg = TinyConfig()
+    m = MaskRCNN("inference", cfg, model_dir=str(tmp_path))
+    base = str(tmp_path / "models" / "cli_model")
+    export_saved_model(m, cfg, base, version=1)
+    mc = tmp_path / "models.config"
+    mc.write_text("model_config_list { config { name: 'cli_model' "
+                  f"base_path: '{base}' }} }}\n".replace("}}", "}"))
+    bp = tmp_path / "batching.txt"
+    bp.write_text("max_batch_size { value: 2 }\n"
+                  "batch_timeout_micros { value: 1000 }\n")
+
+    def free_port():
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        p = s.getsockname()[1]
+        s.close()
+        return p
+
+    gport, rport = free_port(), free_port()
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "maskrcnn_amd.serving.server",
+         "--model_config_file", str(mc),
+         "--batching_parameters_file", str(bp),
+         "--port", str(gport), "--rest_api_port", str(rport)],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+    try:
+        import requests
+        st = None
+        for _ in range(300):
+            if proc.poll() is not None:
+                raise AssertionError(proc.stdout.read()[-2000:])
+            try:
+                st = requests.get(
+                    f"http://127.0.0.1:{rport}/v1/models/cli_model",
+                    timeout=1).json()
+                break
+            except Exception:
+                time.sleep(0.2)
+        assert st is not None, "server never came up"
+        assert st["model_version_status"][0]["state"] == "AVAILABLE"
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+            proc.wait(timeout=10)
