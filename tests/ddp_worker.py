@@ -86,9 +86,55 @@ def run_maskrcnn(rank, world):
     return result
 
 
+def run_staged(rank, world):
+    """Matterport's canonical staged schedule under DDP: fit('heads')
+    then fit('all') in one process — exercises _rebuild_ddp()
+    (remove_hooks + re-bucket + re-broadcast). Double-registered hooks
+    would double-count bucket.pending and hang/raise in
+    finalize_backward."""
+    import tempfile
+
+    import torch
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from maskrcnn_amd.data.synthetic import ShapesConfig, ShapesDataset
+    from maskrcnn_amd.engine.trainer import Trainer
+    from maskrcnn_amd.model import MaskRCNN
+
+    class C(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 200
+        POST_NMS_ROIS_TRAINING = 50
+        TRAIN_ROIS_PER_IMAGE = 16
+        BACKBONE = "resnet50"
+
+    cfg = C()
+    ds = ShapesDataset()
+    ds.load_shapes(8, 128, 128, seed=7)
+    ds.prepare()
+    logs = tempfile.mkdtemp(prefix=f"staged_r{rank}_")
+    m = MaskRCNN("training", cfg, model_dir=logs)
+    tr = Trainer(m, ds)
+    tr.fit(cfg.LEARNING_RATE, epochs=1, layers="heads",
+           steps_per_epoch=2)
+    tr.fit(cfg.LEARNING_RATE / 10, epochs=2, layers="all",
+           steps_per_epoch=2)
+    # Weights must be identical across ranks after synced training.
+    w = torch.cat([p.detach().flatten()
+                   for p in m.model.parameters()])[:8]
+    result = {"rank": rank, "epochs": m.epoch,
+              "w_head": [round(float(v), 6) for v in w],
+              "losses_finite": all(
+                  all(v == v for v in h.values()) for h in tr.history)}
+    dist.destroy_process_group()
+    return result
+
+
 if __name__ == "__main__":
     mode = sys.argv[1]
     rank = int(sys.argv[2])
     world = int(sys.argv[3])
-    fn = {"linear": run_linear, "maskrcnn": run_maskrcnn}[mode]
+    fn = {"linear": run_linear, "maskrcnn": run_maskrcnn,
+          "staged": run_staged}[mode]
     print(json.dumps(fn(rank, world)), flush=True)

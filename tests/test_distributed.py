@@ -115,3 +115,14 @@ def test_train_cli_one_epoch(tmp_path):
     assert len(ckpts) == 1, ckpts
     assert glob.glob(str(tmp_path / "*" / "metrics.csv"))
     assert glob.glob(str(tmp_path / "*" / "train_state.pth"))
+
+
+@pytest.mark.timeout(900)
+def test_staged_training_ddp_two_ranks():
+    """heads -> all staged fits with DDP rebuild between stages: ranks
+    stay weight-synchronized and no bucket double-counts."""
+    res = _run_world("staged", timeout=800)
+    assert res[0]["epochs"] == res[1]["epochs"] == 2
+    assert res[0]["losses_finite"] and res[1]["losses_finite"]
+    assert res[0]["w_head"] == pytest.approx(res[1]["w_head"],
+                                             abs=1e-6)
