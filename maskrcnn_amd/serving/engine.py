@@ -64,6 +64,11 @@ class InferenceEngine:
         key = (t.shape, t.dtype)
         buf = self._pinned.get(key)
         if buf is None:
+            if len(self._pinned) >= 8:
+                # Bounded cache: shape-bucketed batching keeps this at
+                # 2-3 entries in practice; evict oldest under churn
+                # (pinned pages are a scarce host resource).
+                self._pinned.pop(next(iter(self._pinned)))
             buf = torch.empty_like(t, pin_memory=True)
             self._pinned[key] = buf
         buf.copy_(t)
