@@ -222,7 +222,16 @@ class MaskRCNN:
 
     def load_weights(self, path: str, by_name: bool = False, exclude=None):
         """Load weights; `exclude` drops head layers for class-count surgery
-        (reference load_weights:~2080). by_name accepted for parity."""
+        (reference load_weights:~2080). by_name accepted for parity.
+
+        `.h5` paths load matterport Keras checkpoints directly via the
+        name/transpose map in utils/keras_import.py (requires h5py);
+        exclude then takes the reference's KERAS layer names."""
+        if path.endswith(".h5"):
+            from ..utils.keras_import import load_keras_h5
+            load_keras_h5(self.model, path, exclude=exclude)
+            self.set_log_dir(path)
+            return []
         state = torch.load(path, map_location="cpu", weights_only=False)
         sd = state["model"] if isinstance(state, dict) and "model" in state \
             else state
