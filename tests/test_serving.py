@@ -693,3 +693,17 @@ def test_server_cli_with_config_files(tmp_path):
         except subprocess.TimeoutExpired:
             proc.kill()
             proc.wait(timeout=10)
+
+
+def test_batcher_multiple_worker_threads():
+    from maskrcnn_amd.serving.batcher import DynamicBatcher
+    eng = _FakeEngine()
+    b = DynamicBatcher(eng, max_batch_size=2, batch_timeout_micros=500,
+                       num_batch_threads=3)
+    futs = [b.submit(np.full((1, 4, 4, 3), float(i), np.float32),
+                     np.zeros((1, 5), np.float32)) for i in range(12)]
+    for i, f in enumerate(futs):
+        out = f.result(timeout=10)
+        assert out["out"][0] == pytest.approx(i * 4 * 4 * 3)
+    b.close()
+    assert sum(eng.calls) >= 12
