@@ -277,8 +277,7 @@ def main():
                       use_bf16, n_iters=5)
 
     if args.torch_profile and rank == 0:
-        from torch.profiler import (ProfilerActivity, profile,
-                                    tensorboard_trace_handler)
+        from torch.profiler import ProfilerActivity, profile
         activities = [ProfilerActivity.CPU]
         if use_cuda:
             activities.append(ProfilerActivity.CUDA)

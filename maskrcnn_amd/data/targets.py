@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import numpy as np
 
-from ..utils.boxes import (box_refinement, compute_overlaps, extract_bboxes)
+from ..utils.boxes import compute_overlaps, extract_bboxes
 from ..utils.image import resize_image
 from ..utils.masks import minimize_mask, resize_mask
 from ..utils.meta import compose_image_meta

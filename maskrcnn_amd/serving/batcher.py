@@ -15,7 +15,7 @@ import queue
 import threading
 import time
 from concurrent.futures import Future
-from typing import Dict, Optional
+from typing import Optional
 
 import numpy as np
 

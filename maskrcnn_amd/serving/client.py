@@ -10,7 +10,7 @@ wire protos (no TF needed).
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import numpy as np
 

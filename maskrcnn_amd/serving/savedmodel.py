@@ -22,7 +22,6 @@ from __future__ import annotations
 import json
 import os
 
-import numpy as np
 
 from .wireproto import (DT_FLOAT, SignatureDef, TensorInfo,
                         decode_saved_model, encode_saved_model)

@@ -10,7 +10,6 @@ engines behind the dynamic batcher, and Prometheus metrics.
 
 from __future__ import annotations
 
-import json
 import logging
 import os
 import threading

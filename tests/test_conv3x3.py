@@ -1,6 +1,5 @@
 """conv3x3 fused op: CPU fallback + GPU MFMA kernel numerics."""
 
-import numpy as np
 import pytest
 import torch
 import torch.nn.functional as F

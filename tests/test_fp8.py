@@ -1,7 +1,6 @@
 """fp8 (e4m3fn) weight path tests."""
 
 import numpy as np
-import pytest
 import torch
 
 from maskrcnn_amd.utils.fp8 import (dequantize, dequantize_state_dict,

@@ -1,6 +1,5 @@
 """fpn_fuse op tests: CPU fallback correctness + GPU kernel numerics."""
 
-import numpy as np
 import pytest
 import torch
 import torch.nn.functional as F

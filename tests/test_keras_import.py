@@ -2,7 +2,6 @@
 
 import numpy as np
 import pytest
-import torch
 
 from maskrcnn_amd.data.synthetic import ShapesConfig
 from maskrcnn_amd.model.mask_rcnn import MaskRCNNModel

@@ -1,7 +1,6 @@
 """End-to-end CPU model tests (BASELINE config #1 class: plumbing)."""
 
 import numpy as np
-import pytest
 import torch
 
 from maskrcnn_amd.model import MaskRCNN

@@ -5,7 +5,6 @@ tested against in tests/test_ops_gpu.py.
 """
 
 import numpy as np
-import pytest
 import torch
 
 from maskrcnn_amd.ops import batched_nms, crop_and_resize, nms, roi_align

@@ -1,7 +1,6 @@
 """Vectorized (sync-free) DetectionTargetLayer: invariants + equivalence
 of the selected-sample properties with the loop variant."""
 
-import numpy as np
 import pytest
 import torch
 
