@@ -105,10 +105,15 @@ def parse_model_config(path: str) -> List[Dict]:
     for c in cfgs:
         if "name" not in c or "base_path" not in c:
             raise ValueError(f"model config missing name/base_path: {c}")
-        out.append({"name": str(c["name"]),
-                    "base_path": str(c["base_path"]),
-                    "model_platform": str(c.get("model_platform",
-                                                "tensorflow"))})
+        entry = {"name": str(c["name"]),
+                 "base_path": str(c["base_path"]),
+                 "model_platform": str(c.get("model_platform",
+                                             "tensorflow"))}
+        # model_version_policy { all {} } / { latest { num_versions: 1 } }
+        mvp = c.get("model_version_policy")
+        if isinstance(mvp, dict) and "all" in mvp:
+            entry["version_policy"] = "all"
+        out.append(entry)
     return out
 
 

@@ -57,18 +57,35 @@ class ModelState:
 
     def __init__(self, name: str, base_path: str, device=None,
                  devices=None, max_batch_size: int = 32,
-                 batch_timeout_micros: int = 2000, config=None):
+                 batch_timeout_micros: int = 2000, config=None,
+                 version_policy: str = "latest"):
+        """version_policy: 'latest' (TF-Serving default — new version
+        replaces old) or 'all' (every discovered version stays
+        servable; requests pin via model_spec.version)."""
+        assert version_policy in ("latest", "all"), version_policy
         self.name = name
         self.base_path = base_path
         self.devices = list(devices) if devices else [device]
         self.max_batch_size = max_batch_size
         self.batch_timeout_micros = batch_timeout_micros
         self.config_override = config
+        self.version_policy = version_policy
         self.version: Optional[int] = None
-        self.engines: list = []
-        self.batchers: list = []
+        #: version -> {"engines": [...], "batchers": [...]}
+        self.versions: Dict[int, dict] = {}
         self._rr = 0
         self._lock = threading.Lock()
+
+    # Latest-version accessors (single-version compat).
+    @property
+    def engines(self):
+        v = self.versions.get(self.version)
+        return v["engines"] if v else []
+
+    @property
+    def batchers(self):
+        v = self.versions.get(self.version)
+        return v["batchers"] if v else []
 
     # Single-replica compat accessors.
     @property
@@ -94,31 +111,44 @@ class ModelState:
             batch_timeout_micros=self.batch_timeout_micros)
             for eng in engines]
         with self._lock:
-            old = self.batchers
-            self.engines, self.batchers, self.version = \
-                engines, batchers, latest
-        for b in old:
-            b.close()
+            old = (self.versions.pop(self.version, None)
+                   if self.version_policy == "latest" else None)
+            self.versions[latest] = {"engines": engines,
+                                     "batchers": batchers}
+            self.version = latest
+        if old:
+            for b in old["batchers"]:
+                b.close()
         if _PROM:
             VERSION.set(latest)
         return True
 
-    def predict(self, images, metas, anchors=None, outputs=None):
+    def predict(self, images, metas, anchors=None, outputs=None,
+                version=None):
         # Retry on the submit-vs-hot-swap race: maybe_reload() may close
-        # the batcher we just picked; re-pick from the fresh set.
+        # the batcher we just picked; re-pick from the fresh set (the
+        # CURRENT latest for unpinned requests).
+        explicit = version is not None
         for _attempt in range(3):
             with self._lock:
-                if not self.batchers:
+                want = version if explicit else self.version
+                slot = self.versions.get(want)
+                if slot is None:
+                    if explicit:
+                        raise KeyError(
+                            f"model {self.name!r} version {version} not "
+                            f"loaded (serving {sorted(self.versions)})")
                     raise RuntimeError(
                         f"model {self.name} has no loaded version")
+                bl = slot["batchers"]
                 # Shortest-queue replica (round-robin tiebreak).
-                qsizes = [b.queue.qsize() for b in self.batchers]
+                qsizes = [b.queue.qsize() for b in bl]
                 best = min(range(len(qsizes)),
                            key=lambda i: (qsizes[i], (i - self._rr) %
                                           len(qsizes)))
-                self._rr = (best + 1) % len(self.batchers)
-                batcher = self.batchers[best]
-                version = self.version
+                self._rr = (best + 1) % len(bl)
+                batcher = bl[best]
+                version = want
             try:
                 fut = batcher.submit(images, metas, anchors,
                                      outputs=outputs)
@@ -153,7 +183,7 @@ class ModelServer:
                  devices=None, max_batch_size: int = 32,
                  batch_timeout_micros: int = 2000,
                  poll_interval_s: float = 5.0, config=None,
-                 model_configs=None):
+                 model_configs=None, version_policy: str = "latest"):
         """Single model (model_name + model_base_path, TF-Serving's
         --model_name/--model_base_path mode) or several (model_configs =
         parse_model_config(models.config), the --model_config_file
@@ -167,10 +197,14 @@ class ModelServer:
         self._model_kwargs = dict(
             device=device, devices=devices,
             max_batch_size=max_batch_size,
-            batch_timeout_micros=batch_timeout_micros, config=config)
+            batch_timeout_micros=batch_timeout_micros, config=config,
+            version_policy=version_policy)
         self.models: Dict[str, ModelState] = {
-            mc["name"]: ModelState(mc["name"], mc["base_path"],
-                                   **self._model_kwargs)
+            mc["name"]: ModelState(
+                mc["name"], mc["base_path"],
+                **{**self._model_kwargs,
+                   **({"version_policy": mc["version_policy"]}
+                      if "version_policy" in mc else {})})
             for mc in model_configs}
         self.grpc_port = grpc_port
         self.rest_port = rest_port
@@ -229,10 +263,10 @@ class ModelServer:
         if sig != "serving_default":
             raise KeyError(f"unknown signature {sig!r}")
         want_version = request.model_spec.version
-        if want_version is not None and want_version != state.version:
+        if want_version is not None and want_version not in state.versions:
             raise KeyError(
                 f"model {name!r} version {want_version} not loaded "
-                f"(serving {state.version})")
+                f"(serving {sorted(state.versions) or state.version})")
         try:
             images = tensor_to_numpy(request.inputs["input_image"]) \
                 .astype(np.float32)
@@ -248,7 +282,8 @@ class ModelServer:
         # the DtoH transfer entirely.
         outputs, version = state.predict(
             images, metas, anchors,
-            outputs=list(request.output_filter) or None)
+            outputs=list(request.output_filter) or None,
+            version=want_version)
         if request.output_filter:
             outputs = {k: v for k, v in outputs.items()
                        if k in request.output_filter}
@@ -308,11 +343,11 @@ class ModelServer:
                 state = outer._get_state(request.model_spec.name)
             except KeyError as e:
                 context.abort(grpc.StatusCode.NOT_FOUND, str(e))
-            return GetModelStatusResponse(
-                model_version_status=[ModelVersionStatus(
-                    version=state.version or 0,
-                    state=STATE_AVAILABLE if state.version is not None
-                    else 0)])
+            statuses = [ModelVersionStatus(version=v,
+                                           state=STATE_AVAILABLE)
+                        for v in sorted(state.versions)] or \
+                [ModelVersionStatus(version=0, state=0)]
+            return GetModelStatusResponse(model_version_status=statuses)
 
         def handle_reload_config(request: ReloadConfigRequest, context):
             try:
@@ -389,12 +424,15 @@ class ModelServer:
         def model_status(model_name: str):
             state = outer.models.get(model_name)
             ok = state is not None and state.version is not None
+            if not ok:
+                return {"model_version_status": [{
+                    "version": "0", "state": "UNAVAILABLE",
+                    "status": {"error_code": "NOT_FOUND",
+                               "error_message": ""}}]}
             return {"model_version_status": [{
-                "version": str((state.version if state else None) or 0),
-                "state": "AVAILABLE" if ok else "UNAVAILABLE",
-                "status": {"error_code": "OK" if ok else "NOT_FOUND",
-                           "error_message": ""},
-            }]}
+                "version": str(v), "state": "AVAILABLE",
+                "status": {"error_code": "OK", "error_message": ""},
+            } for v in sorted(state.versions)]}
 
         @app.post("/v1/models/{model_and_verb:path}")
         async def predict(model_and_verb: str, body: dict = Body(...)):
@@ -503,8 +541,9 @@ class ModelServer:
         if self._uvicorn_server is not None:
             self._uvicorn_server.should_exit = True
         for state in self.models.values():
-            for b in state.batchers:
-                b.close()
+            for slot in state.versions.values():
+                for b in slot["batchers"]:
+                    b.close()
 
     def wait(self):
         self._grpc_server.wait_for_termination()

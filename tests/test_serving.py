@@ -707,3 +707,55 @@ def test_batcher_multiple_worker_threads():
         assert out["out"][0] == pytest.approx(i * 4 * 4 * 3)
     b.close()
     assert sum(eng.calls) >= 12
+
+
+def test_version_policy_all(tmp_path):
+    """version_policy='all': every version stays servable; requests pin
+    via model_spec.version (TF-Serving model_version_policy {all {}})."""
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+    from maskrcnn_amd.serving.server import ModelState
+
+    class TinyConfig(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 100
+        POST_NMS_ROIS_INFERENCE = 10
+
+    cfg = TinyConfig()
+    m = MaskRCNN("inference", cfg, model_dir=str(tmp_path))
+    base = str(tmp_path / "models" / "vp")
+    export_saved_model(m, cfg, base, version=1)
+    st = ModelState("vp", base, max_batch_size=1,
+                    batch_timeout_micros=100, config=cfg,
+                    version_policy="all")
+    st.maybe_reload()
+    assert st.version == 1
+    export_saved_model(m, cfg, base, version=2)
+    st.maybe_reload()
+    assert st.version == 2
+    assert sorted(st.versions) == [1, 2]
+
+    H = cfg.IMAGE_SHAPE[0]
+    meta = np.zeros((1, cfg.IMAGE_META_SIZE), np.float32)
+    meta[:, 4:7] = (H, H, 3)
+    meta[:, 7:11] = (0, 0, H, H)
+    img = np.zeros((1, H, H, 3), np.float32)
+    _, v = st.predict(img, meta)                 # unpinned -> latest
+    assert v == 2
+    _, v = st.predict(img, meta, version=1)      # pinned -> old version
+    assert v == 1
+    with pytest.raises(KeyError):
+        st.predict(img, meta, version=9)
+    for slot in st.versions.values():
+        for b in slot["batchers"]:
+            b.close()
+
+    # 'latest' policy drops the old version on reload.
+    st2 = ModelState("vp", base, max_batch_size=1,
+                     batch_timeout_micros=100, config=cfg)
+    st2.maybe_reload()
+    assert sorted(st2.versions) == [2]
+    for slot in st2.versions.values():
+        for b in slot["batchers"]:
+            b.close()
