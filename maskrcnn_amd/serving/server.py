@@ -442,10 +442,23 @@ class ModelServer:
                 return JSONResponse({"error": "unsupported verb"},
                                     status_code=404)
             name = model_and_verb[:-len(":predict")]
+            want_version = None
+            if "/versions/" in name:   # /v1/models/{m}/versions/{v}:predict
+                name, _, v = name.partition("/versions/")
+                try:
+                    want_version = int(v)
+                except ValueError:
+                    return JSONResponse(
+                        {"error": f"bad version {v!r}"}, status_code=400)
             state = outer.models.get(name)
             if state is None:
                 return JSONResponse(
                     {"error": f"model {name} not found"}, status_code=404)
+            if want_version is not None and \
+                    want_version not in state.versions:
+                return JSONResponse(
+                    {"error": f"version {want_version} of {name} not "
+                              f"loaded"}, status_code=404)
             sig = body.get("signature_name", "serving_default")
             if sig != "serving_default":
                 return JSONResponse(
@@ -478,7 +491,8 @@ class ModelServer:
                         {"error": "missing inputs/instances"},
                         status_code=400)
                 t0 = time.monotonic()
-                outputs, _ = state.predict(images, metas, anchors)
+                outputs, _ = state.predict(images, metas, anchors,
+                                           version=want_version)
                 if _PROM:
                     LAT.observe(time.monotonic() - t0)
                 return {"outputs": {k: v.tolist()

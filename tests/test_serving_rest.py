@@ -124,3 +124,25 @@ def test_rest_signature_name_validation(rest_server):
         json={"signature_name": "bogus", "inputs": {}}, timeout=10)
     assert r.status_code == 400
     assert "signature" in r.json()["error"]
+
+
+def test_rest_versioned_predict_url(rest_server):
+    """TF-Serving /v1/models/{m}/versions/{v}:predict routing."""
+    import requests
+    server, cfg, port = rest_server
+    cur = server.state.version
+    H = cfg.IMAGE_SHAPE[0]
+    meta = np.concatenate(
+        [[[0] + [H, H, 3] + [H, H, 3] + [0, 0, H, H] + [1.0]],
+         np.ones((1, cfg.NUM_CLASSES))], axis=1)
+    body = {"inputs": {
+        "input_image": np.zeros((1, H, H, 3), np.float32).tolist(),
+        "input_image_meta": meta.tolist()}}
+    r = requests.post(
+        f"http://127.0.0.1:{port}/v1/models/mask_rcnn/versions/"
+        f"{cur}:predict", json=body, timeout=120)
+    assert r.status_code == 200, r.text[:200]
+    r = requests.post(
+        f"http://127.0.0.1:{port}/v1/models/mask_rcnn/versions/"
+        f"999:predict", json=body, timeout=10)
+    assert r.status_code == 404
