@@ -420,6 +420,16 @@ class ModelServer:
                                         for k, v in sig.outputs.items()},
                             "method_name": sig.method_name}}}}}
 
+        @app.get("/v1/models/{model_name}/versions/{version}")
+        def version_status(model_name: str, version: int):
+            state = outer.models.get(model_name)
+            ok = state is not None and version in state.versions
+            return {"model_version_status": [{
+                "version": str(version),
+                "state": "AVAILABLE" if ok else "UNAVAILABLE",
+                "status": {"error_code": "OK" if ok else "NOT_FOUND",
+                           "error_message": ""}}]}
+
         @app.get("/v1/models/{model_name}")
         def model_status(model_name: str):
             state = outer.models.get(model_name)

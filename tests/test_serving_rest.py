@@ -146,3 +146,15 @@ def test_rest_versioned_predict_url(rest_server):
         f"http://127.0.0.1:{port}/v1/models/mask_rcnn/versions/"
         f"999:predict", json=body, timeout=10)
     assert r.status_code == 404
+
+
+def test_rest_version_status_url(rest_server):
+    import requests
+    server, cfg, port = rest_server
+    cur = server.state.version
+    st = requests.get(f"http://127.0.0.1:{port}/v1/models/mask_rcnn/"
+                      f"versions/{cur}", timeout=10).json()
+    assert st["model_version_status"][0]["state"] == "AVAILABLE"
+    st = requests.get(f"http://127.0.0.1:{port}/v1/models/mask_rcnn/"
+                      f"versions/999", timeout=10).json()
+    assert st["model_version_status"][0]["state"] == "UNAVAILABLE"
