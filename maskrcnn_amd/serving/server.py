@@ -445,7 +445,11 @@ class ModelServer:
             } for v in sorted(state.versions)]}
 
         @app.post("/v1/models/{model_and_verb:path}")
-        async def predict(model_and_verb: str, body: dict = Body(...)):
+        def predict(model_and_verb: str, body: dict = Body(...)):
+            # Sync handler ON PURPOSE: FastAPI runs it in its
+            # threadpool, so one slow batch (fut.result blocks up to
+            # minutes on a cold Find) cannot stall the event loop for
+            # every other REST request.
             if _PROM:
                 REQS.labels("rest").inc()
             if not model_and_verb.endswith(":predict"):

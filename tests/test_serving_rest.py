@@ -158,3 +158,35 @@ def test_rest_version_status_url(rest_server):
     st = requests.get(f"http://127.0.0.1:{port}/v1/models/mask_rcnn/"
                       f"versions/999", timeout=10).json()
     assert st["model_version_status"][0]["state"] == "UNAVAILABLE"
+
+
+def test_rest_concurrent_requests_not_serialized(rest_server):
+    """Sync (threadpool) predict handler: concurrent REST predicts must
+    overlap rather than serialize behind one blocked event loop."""
+    import threading
+
+    import requests
+    server, cfg, port = rest_server
+    H = cfg.IMAGE_SHAPE[0]
+    meta = np.concatenate(
+        [[[0] + [H, H, 3] + [H, H, 3] + [0, 0, H, H] + [1.0]],
+         np.ones((1, cfg.NUM_CLASSES))], axis=1)
+    body = {"inputs": {
+        "input_image": np.zeros((1, H, H, 3), np.float32).tolist(),
+        "input_image_meta": meta.tolist()}}
+    codes = []
+
+    def hit():
+        r = requests.post(
+            f"http://127.0.0.1:{port}/v1/models/mask_rcnn:predict",
+            json=body, timeout=120)
+        codes.append(r.status_code)
+
+    ts = [threading.Thread(target=hit) for _ in range(4)]
+    t0 = time.time()
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=120)
+    assert codes == [200] * 4, codes
+    assert time.time() - t0 < 100
