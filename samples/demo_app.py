@@ -72,11 +72,19 @@ def create_app(detector, class_names):
         except Exception as e:  # noqa: BLE001
             return JSONResponse({"error": f"bad image: {e}"},
                                 status_code=400)
-        r = detector(image)
-        buf = io.BytesIO()
-        visualize.display_instances(
-            image, r["rois"], r["masks"], r["class_ids"], class_names,
-            r.get("scores"), save_path=buf)
+        # Inference + rendering are blocking: run off the event loop so
+        # concurrent uploads and the index page stay responsive.
+        import anyio
+
+        def run():
+            r = detector(image)
+            buf = io.BytesIO()
+            visualize.display_instances(
+                image, r["rois"], r["masks"], r["class_ids"],
+                class_names, r.get("scores"), save_path=buf)
+            return r, buf
+
+        r, buf = await anyio.to_thread.run_sync(run)
         return Response(buf.getvalue(), media_type="image/png",
                         headers={"X-Detections":
                                  str(len(r["class_ids"]))})
