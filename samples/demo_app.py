@@ -50,10 +50,14 @@ async function go() {
 def create_app(detector, class_names):
     """detector: callable(image [H,W,3] uint8) -> result dict with
     rois/masks/class_ids/scores (the facade's detect()[0] contract)."""
+    import threading
+
     from fastapi import FastAPI, Request
     from fastapi.responses import HTMLResponse, JSONResponse, Response
 
     app = FastAPI(title="maskrcnn_amd demo")
+    # pyplot keeps GLOBAL figure state — serialize render jobs.
+    render_lock = threading.Lock()
 
     @app.get("/", response_class=HTMLResponse)
     def index():
@@ -77,11 +81,12 @@ def create_app(detector, class_names):
         import anyio
 
         def run():
-            r = detector(image)
-            buf = io.BytesIO()
-            visualize.display_instances(
-                image, r["rois"], r["masks"], r["class_ids"],
-                class_names, r.get("scores"), save_path=buf)
+            with render_lock:
+                r = detector(image)
+                buf = io.BytesIO()
+                visualize.display_instances(
+                    image, r["rois"], r["masks"], r["class_ids"],
+                    class_names, r.get("scores"), save_path=buf)
             return r, buf
 
         r, buf = await anyio.to_thread.run_sync(run)
