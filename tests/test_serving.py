@@ -759,3 +759,37 @@ def test_version_policy_all(tmp_path):
     for slot in st2.versions.values():
         for b in slot["batchers"]:
             b.close()
+
+
+def test_convert_checkpoint_roundtrip(tmp_path):
+    """tools/convert_checkpoint.py: pth -> SavedModel -> pth preserves
+    every tensor."""
+    import subprocess
+    import sys
+
+    import torch
+
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model.mask_rcnn import MaskRCNNModel
+
+    class C(ShapesConfig):
+        BACKBONE = "resnet50"
+
+    cfg = C()
+    m = MaskRCNNModel(cfg)
+    p1 = str(tmp_path / "a.pth")
+    torch.save({"model": m.state_dict()}, p1)
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for src, dst in ((p1, str(tmp_path / "sm")),
+                     (str(tmp_path / "sm"), str(tmp_path / "b.pth"))):
+        r = subprocess.run(
+            [sys.executable, "tools/convert_checkpoint.py", "--src", src,
+             "--dst", dst, "--num-classes", str(cfg.NUM_CLASSES),
+             "--backbone", "resnet50"],
+            capture_output=True, text=True, timeout=300, cwd=root)
+        assert r.returncode == 0, r.stderr[-1500:]
+    sd2 = torch.load(str(tmp_path / "b.pth"), weights_only=False)["model"]
+    sd1 = m.state_dict()
+    for k in sd1:
+        torch.testing.assert_close(sd1[k].float(), sd2[k].float(),
+                                   atol=1e-6, rtol=0)
