@@ -129,3 +129,77 @@ def test_xcd_swizzle_bijective():
             assert 0 <= out < nwg, (nwg, bid, out)
             seen.add(out)
         assert len(seen) == nwg, f"not bijective at nwg={nwg}"
+
+
+def _sim_conv3x3_fp8(xq, wq, bias, w_scale, a_scale, relu=True):
+    """Transcription of conv3x3_fp8.hip: implicit GEMM over 9 taps x
+    Cin/32 chunks with the kernel's exact staging maps (thread ->
+    (pixel row, 8-byte channel octet)) and the dequant epilogue
+    acc * a_scale * w_scale[co] + bias.
+
+    xq: [B,H,W,Cin] int-coded e4m3 values (float here), wq: [9,Cout,
+    Cin] tap-major. Returns [B,H,W,Cout] float."""
+    B, H, W, Cin = xq.shape
+    _, Cout, _ = wq.shape
+    BM, BN, BK = 64, 64, 32
+    M = B * H * W
+    out = np.zeros((M, Cout), np.float32)
+    for m0 in range(0, M, BM):
+        for n0 in range(0, Cout, BN):
+            acc = np.zeros((BM, BN), np.float32)
+            for tap in range(9):
+                dy, dx = tap // 3 - 1, tap % 3 - 1
+                for k0 in range(0, Cin, BK):
+                    a_tile = np.zeros((BM, BK), np.float32)
+                    b_tile = np.zeros((BN, BK), np.float32)
+                    for tid in range(256):
+                        lrow, lcol = tid >> 2, (tid & 3) * 8
+                        m = m0 + lrow
+                        if m < M:
+                            px, rest = m % W, m // W
+                            py, pb = rest % H, rest // H
+                            sy, sx = py + dy, px + dx
+                            if 0 <= sy < H and 0 <= sx < W:
+                                a_tile[lrow, lcol:lcol + 8] = \
+                                    xq[pb, sy, sx, k0 + lcol:k0 + lcol + 8]
+                        n = n0 + lrow
+                        if n < Cout:
+                            b_tile[lrow, lcol:lcol + 8] = \
+                                wq[tap, n, k0 + lcol:k0 + lcol + 8]
+                    acc += a_tile @ b_tile.T
+            for r in range(min(BM, M - m0)):
+                for c in range(min(BN, Cout - n0)):
+                    v = acc[r, c] * a_scale * w_scale[n0 + c] + bias[n0 + c]
+                    out[m0 + r, n0 + c] = max(v, 0.0) if relu else v
+    return out.reshape(B, H, W, Cout)
+
+
+def test_fp8_kernel_simulation_matches_reference():
+    """Design check of conv3x3_fp8.hip against the quantized-math
+    reference (exact: both sides use the same e4m3-coded operands)."""
+    torch.manual_seed(0)
+    B, H, W, Cin, Cout = 1, 5, 6, 32, 16
+    from maskrcnn_amd.ops.conv3x3 import pack_weight_fp8
+    from maskrcnn_amd.utils.fp8 import FP8_DTYPE, FP8_MAX
+
+    x = torch.randn(B, Cin, H, W)
+    w = torch.randn(Cout, Cin, 3, 3) * 0.1
+    bias = torch.randn(Cout)
+    qp, w_scale = pack_weight_fp8(w)
+    amax = x.abs().amax().clamp(min=1e-12)
+    a_scale = float(amax / FP8_MAX)
+    xq = (x / a_scale).clamp(-FP8_MAX, FP8_MAX).to(FP8_DTYPE)
+
+    sim = _sim_conv3x3_fp8(
+        xq.float().permute(0, 2, 3, 1).contiguous().numpy(),
+        qp.view(torch.float8_e4m3fn).float().numpy(),
+        bias.numpy(), w_scale.numpy(), a_scale, relu=True)
+
+    # Reference: same dequantized operands through torch conv.
+    x_deq = xq.float() * a_scale
+    w_deq = (qp.view(torch.float8_e4m3fn).float() *
+             w_scale.view(1, -1, 1)) \
+        .reshape(3, 3, Cout, Cin).permute(2, 3, 0, 1)
+    ref = F.relu(F.conv2d(x_deq, w_deq, bias, padding=1))
+    got = torch.from_numpy(sim).permute(0, 3, 1, 2)
+    torch.testing.assert_close(got, ref, atol=1e-3, rtol=1e-3)
