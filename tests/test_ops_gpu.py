@@ -175,3 +175,45 @@ def test_nms_padded_gpu_matches_cpu(dev):
     keep_g, count_g = nms_padded(boxes.to(dev), scores.to(dev), 0.6, 700)
     assert int(count_g.item()) == int(count_c.item())
     np.testing.assert_array_equal(keep_g.cpu().numpy(), keep_c.numpy())
+
+
+@pytest.mark.gpu
+def test_detection_syncfree_gpu_matches_eager(dev):
+    """Capture-safe DetectionLayer path on GPU (HIP nms_padded) equals
+    the eager nonzero/gather path."""
+    import os
+
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model.detection import DetectionLayer
+    from maskrcnn_amd.utils.meta import compose_image_meta
+
+    class C(ShapesConfig):
+        DETECTION_MIN_CONFIDENCE = 0.3
+        DETECTION_MAX_INSTANCES = 10
+
+    cfg = C()
+    torch.manual_seed(0)
+    layer = DetectionLayer(cfg).to(dev)
+    B, N, Ccls = 3, 80, cfg.NUM_CLASSES
+    rois = torch.rand(B, N, 4, device=dev)
+    rois[..., 2:] = rois[..., :2] * 0.5 + 0.4
+    probs = torch.softmax(torch.randn(B, N, Ccls, device=dev) * 2, -1)
+    deltas = torch.randn(B, N, Ccls, 4, device=dev) * 0.1
+    metas = torch.stack([
+        torch.from_numpy(compose_image_meta(
+            i, (128, 128, 3), (128, 128, 3), (0, 0, 128, 128), 1.0,
+            np.ones(Ccls))).float() for i in range(B)]).to(dev)
+    try:
+        os.environ["MASKRCNN_SYNCFREE_DETECTION"] = "0"
+        eager = layer(rois, probs, deltas, metas)
+        os.environ["MASKRCNN_SYNCFREE_DETECTION"] = "1"
+        syncfree = layer(rois, probs, deltas, metas)
+    finally:
+        os.environ.pop("MASKRCNN_SYNCFREE_DETECTION", None)
+    for b in range(B):
+        ne = int((eager[b][:, 5] > 0).sum())
+        ns = int((syncfree[b][:, 5] > 0).sum())
+        assert ne == ns
+        torch.testing.assert_close(
+            syncfree[b][:ns].sort(dim=0).values,
+            eager[b][:ne].sort(dim=0).values, atol=1e-5, rtol=1e-5)
