@@ -139,7 +139,7 @@ def test_model_train_step_gpu(dev):
     from bench import build_config, make_synthetic_batch
     import argparse
     args = argparse.Namespace(batch=1, image_size=512, steps=1, warmup=0,
-                              dtype="bf16")
+                              dtype="bf16", backbone="resnet101")
     cfg = build_config(args)
     from maskrcnn_amd.model.mask_rcnn import MaskRCNNModel
     from maskrcnn_amd.model.losses import compute_losses
