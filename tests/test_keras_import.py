@@ -89,3 +89,15 @@ def test_unknown_names_reported_not_fatal():
                                                  np.float32)})
     assert unexpected == ["totally_new_layer/kernel"]
     assert not loaded and len(missing) > 100
+
+
+def test_facade_h5_path_without_h5py(tmp_path):
+    """load_weights('.h5') routes to the Keras importer; without h5py
+    the error must say exactly what to do."""
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    m = MaskRCNN("inference", ShapesConfig(), model_dir=str(tmp_path))
+    f = tmp_path / "w.h5"
+    f.write_bytes(b"\x89HDF\r\n")
+    with pytest.raises(ImportError, match="h5py"):
+        m.load_weights(str(f), by_name=True)
