@@ -21,6 +21,19 @@ import numpy as np
 
 log = logging.getLogger(__name__)
 
+try:
+    from prometheus_client import Histogram
+    QUEUE_WAIT = Histogram(
+        "batch_queue_wait_seconds",
+        "request time spent queued before batch assembly",
+        buckets=(.0005, .001, .0025, .005, .01, .025, .05, .1, .5, 2))
+    INFER_TIME = Histogram(
+        "batch_inference_seconds", "engine.predict time per batch",
+        buckets=(.005, .01, .025, .05, .1, .25, .5, 1, 2.5, 10))
+    _PROM = True
+except ImportError:  # pragma: no cover
+    _PROM = False
+
 
 class DynamicBatcher:
     def __init__(self, engine, max_batch_size: int = 32,
@@ -58,6 +71,7 @@ class DynamicBatcher:
         if self._stop:
             raise RuntimeError("batcher closed")
         fut: Future = Future()
+        fut._enqueued_at = time.monotonic()
         self.queue.put((input_image, input_image_meta, input_anchors,
                         outputs, fut))
         return fut
@@ -132,10 +146,17 @@ class DynamicBatcher:
                 metas = np.concatenate(
                     [metas, np.repeat(metas[:1], pad, axis=0)], axis=0)
             t0 = time.monotonic()
+            if _PROM:
+                for *_r, fut in batch:
+                    QUEUE_WAIT.observe(
+                        t0 - getattr(fut, "_enqueued_at", t0))
             out = self.engine.predict(images, metas, anchors,
                                       trim_to=real_n, outputs=outputs)
+            dt = time.monotonic() - t0
+            if _PROM:
+                INFER_TIME.observe(dt)
             log.info("batch of %d (padded %d) in %.3fs", real_n,
-                     images.shape[0], time.monotonic() - t0)
+                     images.shape[0], dt)
             self.stats["batches"] += 1
             self.stats["requests"] += len(batch)
             self.stats["batch_size_sum"] += images.shape[0]
