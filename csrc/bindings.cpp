@@ -36,6 +36,10 @@ void conv3x3_wrw_launch(const void*, const void*, float*, int, int, int,
 void conv3x3_fp8_launch(const void*, const void*, const float*,
                         const float*, const float*, void*, int, int, int,
                         int, int, bool, hipStream_t);
+void detection_prep_launch(const void*, const void*, const void*,
+                           const float*, const float*, float*, int*,
+                           float*, int, int, int, float, bool,
+                           hipStream_t);
 }
 
 namespace {
@@ -301,6 +305,41 @@ torch::Tensor conv3x3_wrw(torch::Tensor x, torch::Tensor dy) {
   return dw;
 }
 
+std::vector<torch::Tensor> detection_prep(
+    torch::Tensor rois, torch::Tensor probs, torch::Tensor deltas,
+    torch::Tensor windows, torch::Tensor std_dev, double min_conf) {
+  // rois [B,N,4], probs [B,N,C], deltas [B,N,C,4] (fp32 or bf16,
+  // same dtype); windows [B,4] f32; std_dev [4] f32.
+  TORCH_CHECK(rois.is_cuda() && probs.is_cuda() && deltas.is_cuda(),
+              "detection_prep: GPU tensors");
+  TORCH_CHECK(rois.dim() == 3 && probs.dim() == 3 && deltas.dim() == 4,
+              "detection_prep: bad ranks");
+  const int B = rois.size(0), N = rois.size(1), C = probs.size(2);
+  TORCH_CHECK(deltas.size(2) == C && deltas.size(3) == 4,
+              "deltas must be [B,N,C,4]");
+  auto dtype = probs.scalar_type();
+  TORCH_CHECK(dtype == torch::kFloat || dtype == torch::kBFloat16,
+              "detection_prep: fp32/bf16 only");
+  auto r = rois.to(dtype).contiguous();
+  auto p = probs.contiguous();
+  auto d = deltas.to(dtype).contiguous();
+  auto w = windows.to(torch::kFloat).contiguous();
+  auto sd = std_dev.to(torch::kFloat).contiguous();
+  TORCH_CHECK(w.is_cuda() && w.numel() == (long)B * 4, "windows [B,4]");
+  TORCH_CHECK(sd.is_cuda() && sd.numel() == 4, "std_dev [4]");
+  auto refined = torch::empty({B, N, 4},
+                              r.options().dtype(torch::kFloat));
+  auto cls = torch::empty({B, N}, r.options().dtype(torch::kInt));
+  auto scores = torch::empty({B, N}, r.options().dtype(torch::kFloat));
+  detection_prep_launch(r.data_ptr(), p.data_ptr(), d.data_ptr(),
+                        w.data_ptr<float>(), sd.data_ptr<float>(),
+                        refined.data_ptr<float>(), cls.data_ptr<int>(),
+                        scores.data_ptr<float>(), B, N, C,
+                        (float)min_conf, dtype == torch::kBFloat16,
+                        cur_stream());
+  return {refined, cls, scores};
+}
+
 }  // namespace
 
 void register_rle(pybind11::module_& m);
@@ -329,6 +368,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("w_packed"), py::arg("bias"),
         py::arg("w_scale"), py::arg("a_scale"),
         py::arg("relu") = false);
+  m.def("detection_prep", &detection_prep,
+        "fused argmax/class-delta/clip detection prep (N9)",
+        py::arg("rois"), py::arg("probs"), py::arg("deltas"),
+        py::arg("windows"), py::arg("std_dev"), py::arg("min_conf"));
   m.def("conv3x3_wrw", &conv3x3_wrw,
         "3x3 conv s1p1 weight gradient (CDNA4 MFMA, split-K atomics)",
         py::arg("x"), py::arg("dy"));

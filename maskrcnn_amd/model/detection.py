@@ -54,6 +54,22 @@ class DetectionLayer(nn.Module):
                                (float(image_shape[0]),
                                 float(image_shape[1])))  # [B, 4]
 
+        if _use_syncfree(rois):
+            # Fused HIP prep (one detection_prep launch replaces the
+            # ~8 argmax/gather/refine/clip launches below) when the
+            # extension is present; torch prep otherwise.
+            from ..ops._ext import load_ext
+            ext = load_ext()
+            if rois.is_cuda and ext is not None:
+                dt = probs.dtype if probs.dtype in (
+                    torch.float32, torch.bfloat16) else torch.float32
+                refined, class_ids, scores_m = ext.detection_prep(
+                    rois.to(dt), probs.to(dt), deltas.to(dt),
+                    windows, self.bbox_std_dev,
+                    float(cfg.DETECTION_MIN_CONFIDENCE or 0.0))
+                return self._nms_scatter(refined, class_ids.long(),
+                                         scores_m, B, N, C)
+
         probs = probs.float()
         class_ids = probs.argmax(dim=2)                        # [B, N]
         class_scores = probs.gather(
@@ -116,11 +132,17 @@ class DetectionLayer(nn.Module):
         gathered eager path); padded NMS + masked index_put_ scatter
         into a dummy-row buffer replaces nonzero()/boolean gathers —
         the same pattern as ProposalLayer."""
+        scores_m = torch.where(keep, class_scores,
+                               torch.zeros_like(class_scores))
+        return self._nms_scatter(refined, class_ids, scores_m, B, N, C)
+
+    def _nms_scatter(self, refined, class_ids, scores_m, B, N, C):
+        """Chunked grouped padded-NMS + masked scatter (shared by the
+        torch and HIP-prep sync-free paths). scores_m: dropped rois
+        already zeroed."""
         cfg = self.config
         device = refined.device
         D = cfg.DETECTION_MAX_INSTANCES
-        scores_m = torch.where(keep, class_scores,
-                               torch.zeros_like(class_scores))
         out_buf = refined.new_zeros((B + 1, D + 1, 6))
         group_sz = max(1, 16384 // N)
         for g0 in range(0, B, group_sz):

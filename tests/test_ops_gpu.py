@@ -217,3 +217,36 @@ def test_detection_syncfree_gpu_matches_eager(dev):
         torch.testing.assert_close(
             syncfree[b][:ns].sort(dim=0).values,
             eager[b][:ne].sort(dim=0).values, atol=1e-5, rtol=1e-5)
+
+
+@pytest.mark.gpu
+@pytest.mark.xfail(strict=False,
+                   reason="detection_prep pending first GPU validation")
+def test_detection_prep_kernel_matches_torch(dev):
+    """Fused HIP prep vs the layer's torch prep block."""
+    _require_ext()
+    from maskrcnn_amd.ops._ext import require_ext
+    from maskrcnn_amd.utils.boxes import apply_box_deltas_t
+    ext = require_ext()
+    torch.manual_seed(1)
+    B, N, C = 3, 50, 5
+    rois = torch.rand(B, N, 4, device=dev)
+    rois[..., 2:] = rois[..., :2] * 0.5 + 0.4
+    probs = torch.softmax(torch.randn(B, N, C, device=dev) * 2, -1)
+    deltas = torch.randn(B, N, C, 4, device=dev) * 0.1
+    windows = torch.tensor([[0., 0., 1., 1.]] * B, device=dev)
+    std = torch.tensor([0.1, 0.1, 0.2, 0.2], device=dev)
+    refined, cls, scores = ext.detection_prep(rois, probs, deltas,
+                                              windows, std, 0.3)
+    class_ids = probs.argmax(2)
+    class_scores = probs.gather(2, class_ids[..., None]).squeeze(-1)
+    ds = deltas.gather(2, class_ids[..., None, None]
+                       .expand(-1, -1, 1, 4)).squeeze(2)
+    ref = apply_box_deltas_t(rois, ds * std).clamp(0, 1)
+    keep = (class_ids > 0) & (class_scores >= 0.3)
+    torch.testing.assert_close(refined, ref, atol=1e-5, rtol=1e-5)
+    assert torch.equal(cls.long(), class_ids)
+    torch.testing.assert_close(
+        scores, torch.where(keep, class_scores,
+                            torch.zeros_like(class_scores)),
+        atol=1e-6, rtol=1e-6)

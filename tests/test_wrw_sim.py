@@ -203,3 +203,73 @@ def test_fp8_kernel_simulation_matches_reference():
     ref = F.relu(F.conv2d(x_deq, w_deq, bias, padding=1))
     got = torch.from_numpy(sim).permute(0, 3, 1, 2)
     torch.testing.assert_close(got, ref, atol=1e-3, rtol=1e-3)
+
+
+def test_detection_prep_simulation_matches_torch():
+    """Transcription of detection_prep.hip's per-roi math vs the
+    DetectionLayer torch prep (argmax, class delta, apply, clip,
+    zero-score masking)."""
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.utils.boxes import apply_box_deltas_t
+
+    class C(ShapesConfig):
+        DETECTION_MIN_CONFIDENCE = 0.4
+
+    cfg = C()
+    torch.manual_seed(0)
+    B, N, Ccls = 2, 30, cfg.NUM_CLASSES
+    rois = torch.rand(B, N, 4)
+    rois[..., 2:] = rois[..., :2] * 0.5 + 0.4
+    probs = torch.softmax(torch.randn(B, N, Ccls) * 2, -1)
+    deltas = torch.randn(B, N, Ccls, 4) * 0.1
+    windows = torch.tensor([[0.0, 0.0, 1.0, 1.0],
+                            [0.1, 0.05, 0.9, 0.95]])
+    std = torch.tensor(cfg.BBOX_STD_DEV, dtype=torch.float32)
+
+    # --- kernel math, one "thread" per (b, n)
+    ref_boxes = torch.zeros(B, N, 4)
+    ref_cls = torch.zeros(B, N, dtype=torch.long)
+    ref_scores = torch.zeros(B, N)
+    import math
+    for b in range(B):
+        for n in range(N):
+            p = probs[b, n]
+            best = int(p.argmax())
+            best_v = float(p[best])
+            d = deltas[b, n, best] * std
+            y1, x1, y2, x2 = rois[b, n].tolist()
+            h, w = y2 - y1, x2 - x1
+            cy = y1 + 0.5 * h + float(d[0]) * h
+            cx = x1 + 0.5 * w + float(d[1]) * w
+            h *= math.exp(float(d[2]))
+            w *= math.exp(float(d[3]))
+            y1, x1 = cy - 0.5 * h, cx - 0.5 * w
+            y2, x2 = y1 + h, x1 + w
+            wy1, wx1, wy2, wx2 = windows[b].tolist()
+            box = [min(max(y1, wy1), wy2), min(max(x1, wx1), wx2),
+                   min(max(y2, wy1), wy2), min(max(x2, wx1), wx2)]
+            ref_boxes[b, n] = torch.tensor(box)
+            ref_cls[b, n] = best
+            keep = best > 0 and best_v >= cfg.DETECTION_MIN_CONFIDENCE
+            ref_scores[b, n] = best_v if keep else 0.0
+
+    # --- the layer's torch prep (the eager path's block)
+    class_ids = probs.argmax(2)
+    class_scores = probs.gather(2, class_ids[..., None]).squeeze(-1)
+    ds = deltas.gather(2, class_ids[..., None, None]
+                       .expand(-1, -1, 1, 4)).squeeze(2)
+    refined = apply_box_deltas_t(rois, ds * std)
+    wy1, wx1, wy2, wx2 = windows.unsqueeze(2).unbind(dim=1)
+    refined = torch.stack(
+        [refined[..., 0].clamp(wy1, wy2), refined[..., 1].clamp(wx1, wx2),
+         refined[..., 2].clamp(wy1, wy2), refined[..., 3].clamp(wx1, wx2)],
+        dim=-1)
+    keep = (class_ids > 0) & (class_scores >=
+                              cfg.DETECTION_MIN_CONFIDENCE)
+    scores_m = torch.where(keep, class_scores,
+                           torch.zeros_like(class_scores))
+
+    torch.testing.assert_close(ref_boxes, refined, atol=1e-5, rtol=1e-5)
+    assert torch.equal(ref_cls, class_ids)
+    torch.testing.assert_close(ref_scores, scores_m, atol=1e-6,
+                               rtol=1e-6)
