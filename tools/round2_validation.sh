@@ -11,8 +11,8 @@ mkdir -p gpurun_out
 case "${1:-block1}" in
 block1)  # v3 + wrw + fp8 kernel validation + conv A/B (~5 min)
   timeout 300 python tools/check_gpu_env.py || exit 1
-  timeout 400 python -m pytest tests/test_conv3x3.py -m gpu -q \
-      --runxfail > gpurun_out/r2_conv_tests.log 2>&1
+  timeout 500 python -m pytest tests/test_conv3x3.py tests/test_ops_gpu.py \
+      -m gpu -q --runxfail > gpurun_out/r2_conv_tests.log 2>&1
   tail -4 gpurun_out/r2_conv_tests.log
   timeout 400 python tools/conv_bench.py > gpurun_out/r2_conv_ab.log 2>&1
   grep -v libdrm gpurun_out/r2_conv_ab.log
