@@ -793,3 +793,24 @@ def test_convert_checkpoint_roundtrip(tmp_path):
     for k in sd1:
         torch.testing.assert_close(sd1[k].float(), sd2[k].float(),
                                    atol=1e-6, rtol=0)
+
+
+def test_serving_bench_cpu_smoke():
+    """bench/serving_bench.py end to end on CPU (tiny config): the
+    round-2 block3 command path must not rot under server refactors."""
+    import json
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "bench/serving_bench.py", "--requests", "8",
+         "--client-procs", "2", "--client-threads", "2", "--warmup", "2",
+         "--image-size", "128", "--backbone", "resnet50",
+         "--max-batch", "2"],
+        capture_output=True, text=True, timeout=600, cwd=root)
+    assert r.returncode == 0, r.stderr[-2000:]
+    rec = json.loads(r.stdout.strip().splitlines()[-1])
+    assert rec["metric"] == "serving_images_per_sec"
+    assert rec["value"] > 0
+    assert rec["steps"] == 8
