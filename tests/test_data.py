@@ -107,3 +107,30 @@ def test_dataset_prepare_class_mapping(shapes_dataset):
     assert ds.num_classes == 4
     assert ds.class_names[0] == "BG"
     assert ds.map_source_class_id("shapes.1") == 1
+
+
+def test_loader_crop_mode_end_to_end():
+    """IMAGE_RESIZE_MODE='crop' (nucleus-style training): samples crop
+    to IMAGE_MIN_DIM squares with consistent targets."""
+    import torch
+
+    from maskrcnn_amd.data.loader import TorchMaskRCNNDataset
+    from maskrcnn_amd.data.synthetic import ShapesConfig, ShapesDataset
+
+    class CropConfig(ShapesConfig):
+        IMAGE_RESIZE_MODE = "crop"
+        IMAGE_MIN_DIM = 64
+        IMAGE_MAX_DIM = None
+        IMAGE_MIN_SCALE = 0
+        USE_MINI_MASK = True
+
+    cfg = CropConfig()
+    ds = ShapesDataset()
+    ds.load_shapes(4, 128, 128, seed=3)
+    ds.prepare()
+    tds = TorchMaskRCNNDataset(ds, cfg)
+    s = tds[0]
+    assert s["image"].shape == (3, 64, 64)
+    assert s["gt_masks"].shape == (*cfg.MINI_MASK_SHAPE,
+                                   cfg.MAX_GT_INSTANCES)
+    assert torch.isfinite(s["rpn_bbox"]).all()
