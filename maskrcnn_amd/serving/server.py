@@ -420,6 +420,18 @@ class ModelServer:
                                         for k, v in sig.outputs.items()},
                             "method_name": sig.method_name}}}}}
 
+        @app.get("/monitoring/prometheus/metrics")
+        def prometheus_metrics():
+            # TF-Serving's --monitoring_config_file path on :8501.
+            if not _PROM:
+                return JSONResponse({"error": "prometheus_client "
+                                     "not installed"}, status_code=501)
+            from prometheus_client import (CONTENT_TYPE_LATEST,
+                                           generate_latest)
+            from fastapi.responses import Response as _Resp
+            return _Resp(generate_latest(),
+                         media_type=CONTENT_TYPE_LATEST)
+
         @app.get("/v1/models/{model_name}/versions/{version}")
         def version_status(model_name: str, version: int):
             state = outer.models.get(model_name)

@@ -190,3 +190,15 @@ def test_rest_concurrent_requests_not_serialized(rest_server):
         t.join(timeout=120)
     assert codes == [200] * 4, codes
     assert time.time() - t0 < 100
+
+
+def test_rest_prometheus_metrics_path(rest_server):
+    """TF-Serving monitoring path on the REST port."""
+    import requests
+    server, cfg, port = rest_server
+    r = requests.get(
+        f"http://127.0.0.1:{port}/monitoring/prometheus/metrics",
+        timeout=10)
+    assert r.status_code == 200
+    assert b"predict_requests_total" in r.content or \
+        b"batch_queue_wait" in r.content
