@@ -194,3 +194,26 @@ def test_roialign_syncfree_matches_eager(monkeypatch):
     monkeypatch.setenv("MASKRCNN_SYNCFREE_ROI", "1")
     syncfree = layer(rois, feats)
     torch.testing.assert_close(syncfree, eager, atol=1e-5, rtol=1e-5)
+
+
+def test_full_inference_syncfree_matches_default(shapes_config,
+                                                 monkeypatch):
+    """Whole-model inference with BOTH capture-safe paths forced equals
+    the default eager forward (the hipGraph-captured graph computes
+    exactly this composition)."""
+    import numpy as np
+
+    m = MaskRCNN("inference", shapes_config, model_dir="/tmp/sf_logs")
+    img = (np.random.RandomState(5).rand(120, 100, 3) * 255) \
+        .astype(np.uint8)
+    monkeypatch.setenv("MASKRCNN_SYNCFREE_DETECTION", "0")
+    monkeypatch.setenv("MASKRCNN_SYNCFREE_ROI", "0")
+    base = m.detect([img])[0]
+    monkeypatch.setenv("MASKRCNN_SYNCFREE_DETECTION", "1")
+    monkeypatch.setenv("MASKRCNN_SYNCFREE_ROI", "1")
+    sf = m.detect([img])[0]
+    assert base["rois"].shape == sf["rois"].shape
+    np.testing.assert_allclose(sf["rois"], base["rois"], atol=1)
+    np.testing.assert_array_equal(sf["class_ids"], base["class_ids"])
+    np.testing.assert_allclose(sf["scores"], base["scores"], atol=1e-5)
+    np.testing.assert_array_equal(sf["masks"], base["masks"])
