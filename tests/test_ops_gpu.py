@@ -178,9 +178,12 @@ def test_nms_padded_gpu_matches_cpu(dev):
 
 
 @pytest.mark.gpu
+@pytest.mark.xfail(strict=False,
+                   reason="routes through detection_prep — pending "
+                          "first GPU validation")
 def test_detection_syncfree_gpu_matches_eager(dev):
-    """Capture-safe DetectionLayer path on GPU (HIP nms_padded) equals
-    the eager nonzero/gather path."""
+    """Capture-safe DetectionLayer path on GPU (detection_prep + HIP
+    nms_padded) equals the eager nonzero/gather path."""
     import os
 
     from maskrcnn_amd.data.synthetic import ShapesConfig
