@@ -204,3 +204,47 @@ def test_native_rle_iou_matrix():
             inter = np.logical_and(dm, gm).sum()
             denom = dm.sum() if j == 1 else np.logical_or(dm, gm).sum()
             assert got[i, j] == pytest.approx(inter / max(denom, 1e-10))
+
+
+def test_visualize_suite_smoke(tmp_path, shapes_dataset):
+    """Every public visualize function renders without error (the
+    reference's notebook helpers are part of the capability surface)."""
+    import matplotlib
+    matplotlib.use("Agg")
+    import numpy as np
+
+    from maskrcnn_amd import visualize
+
+    ds = shapes_dataset
+    image = ds.load_image(0)
+    masks, class_ids = ds.load_mask(0)
+    from maskrcnn_amd.utils.boxes import extract_bboxes
+    boxes = extract_bboxes(masks)
+    scores = np.linspace(0.9, 0.5, len(class_ids))
+
+    visualize.display_images([image, image], titles=["a", "b"],
+                             save_path=str(tmp_path / "imgs.png"))
+    visualize.display_top_masks(image, masks, class_ids, ds.class_names,
+                                save_path=str(tmp_path / "top.png"))
+    visualize.draw_boxes(image, boxes=boxes, masks=masks,
+                         save_path=str(tmp_path / "boxes.png"))
+    visualize.draw_box(image.copy(), boxes[0], (255, 0, 0))
+    visualize.plot_precision_recall(
+        0.5, np.linspace(1, 0, 11), np.linspace(0, 1, 11),
+        save_path=str(tmp_path / "pr.png"))
+    visualize.plot_overlaps(
+        class_ids, class_ids, scores,
+        np.eye(len(class_ids)), ds.class_names, threshold=0.5,
+        save_path=str(tmp_path / "ov.png"))
+    visualize.display_differences(
+        image, boxes, class_ids, masks, boxes, class_ids, scores, masks,
+        ds.class_names, save_path=str(tmp_path / "diff.png"))
+    visualize.display_table([["a", "b"], ["1", "2"]])
+    colors = visualize.random_colors(5)
+    assert len(colors) == 5
+    out = visualize.apply_mask(image.astype(np.float32),
+                               masks[:, :, 0], colors[0])
+    assert out.shape == image.shape
+    for f in ("imgs.png", "top.png", "boxes.png", "pr.png", "ov.png",
+              "diff.png"):
+        assert (tmp_path / f).exists()
