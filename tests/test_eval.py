@@ -240,6 +240,10 @@ def test_visualize_suite_smoke(tmp_path, shapes_dataset):
         image, boxes, class_ids, masks, boxes, class_ids, scores, masks,
         ds.class_names, save_path=str(tmp_path / "diff.png"))
     visualize.display_table([["a", "b"], ["1", "2"]])
+    visualize.draw_rois(image, boxes.astype(np.float32),
+                        boxes.astype(np.float32), masks, class_ids,
+                        ds.class_names, limit=2,
+                        save_path=str(tmp_path / "rois.png"))
     colors = visualize.random_colors(5)
     assert len(colors) == 5
     out = visualize.apply_mask(image.astype(np.float32),
