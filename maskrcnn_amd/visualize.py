@@ -402,9 +402,13 @@ def display_table(table):
 
 
 def display_weight_stats(model):
-    """Weight shape/min/max/std table (reference display_weight_stats)."""
+    """Weight shape/min/max/std table (reference display_weight_stats).
+
+    Accepts the MaskRCNN facade (reference signature) or a raw
+    nn.Module."""
+    net = getattr(model, "model", model)
     table = [["WEIGHT NAME", "SHAPE", "MIN", "MAX", "STD"]]
-    for name, p in model.named_parameters():
+    for name, p in net.named_parameters():
         weight = p.detach().cpu().numpy()
         alert = ""
         if weight.min() == weight.max():

@@ -252,3 +252,12 @@ def test_visualize_suite_smoke(tmp_path, shapes_dataset):
     for f in ("imgs.png", "top.png", "boxes.png", "pr.png", "ov.png",
               "diff.png"):
         assert (tmp_path / f).exists()
+
+
+def test_display_weight_stats_accepts_facade(shapes_config, tmp_path,
+                                             capsys):
+    from maskrcnn_amd import visualize
+    from maskrcnn_amd.model import MaskRCNN
+    m = MaskRCNN("inference", shapes_config, model_dir=str(tmp_path))
+    table = visualize.display_weight_stats(m)
+    assert len(table) > 100 and table[0][0] == "WEIGHT NAME"
