@@ -320,7 +320,7 @@ std::vector<torch::Tensor> detection_prep(
   auto dtype = probs.scalar_type();
   TORCH_CHECK(dtype == torch::kFloat || dtype == torch::kBFloat16,
               "detection_prep: fp32/bf16 only");
-  auto r = rois.to(dtype).contiguous();
+  auto r = rois.to(torch::kFloat).contiguous();  // never downcast boxes
   auto p = probs.contiguous();
   auto d = deltas.to(dtype).contiguous();
   auto w = windows.to(torch::kFloat).contiguous();

@@ -16,7 +16,11 @@ namespace {
 
 template <typename T>
 __global__ void detection_prep_kernel(
-    const T* __restrict__ rois,        // [B,N,4] normalized
+    const float* __restrict__ rois,    // [B,N,4] normalized (ALWAYS
+                                       // fp32: bf16 coords lose ~2 px
+                                       // at 1024² — probs/deltas come
+                                       // from the bf16 head, rois from
+                                       // the fp32 proposal layer)
     const T* __restrict__ probs,       // [B,N,C]
     const T* __restrict__ deltas,      // [B,N,C,4]
     const float* __restrict__ windows, // [B,4] normalized y1x1y2x2
@@ -44,8 +48,8 @@ __global__ void detection_prep_kernel(
   const float dx = ld(d + 1) * std_dev[1];
   const float dh = ld(d + 2) * std_dev[2];
   const float dw = ld(d + 3) * std_dev[3];
-  const T* r = rois + i * 4;
-  float y1 = ld(r + 0), x1 = ld(r + 1), y2 = ld(r + 2), x2 = ld(r + 3);
+  const float* r = rois + i * 4;
+  float y1 = r[0], x1 = r[1], y2 = r[2], x2 = r[3];
   float h = y2 - y1, w = x2 - x1;
   float cy = y1 + 0.5f * h + dy * h;
   float cx = x1 + 0.5f * w + dx * w;
@@ -85,8 +89,7 @@ void detection_prep_launch(const void* rois, const void* probs,
   dim3 grid((unsigned)ceil_div((int)total, 256));
   if (bf16) {
     hipLaunchKernelGGL(detection_prep_kernel<__hip_bfloat16>, grid,
-                       dim3(256), 0, stream,
-                       (const __hip_bfloat16*)rois,
+                       dim3(256), 0, stream, (const float*)rois,
                        (const __hip_bfloat16*)probs,
                        (const __hip_bfloat16*)deltas, windows, std_dev,
                        refined, class_ids, scores, B, N, C, min_conf);

@@ -64,7 +64,7 @@ class DetectionLayer(nn.Module):
                 dt = probs.dtype if probs.dtype in (
                     torch.float32, torch.bfloat16) else torch.float32
                 refined, class_ids, scores_m = ext.detection_prep(
-                    rois.to(dt), probs.to(dt), deltas.to(dt),
+                    rois.float(), probs.to(dt), deltas.to(dt),
                     windows, self.bbox_std_dev,
                     float(cfg.DETECTION_MIN_CONFIDENCE or 0.0))
                 return self._nms_scatter(refined, class_ids.long(),
