@@ -133,11 +133,17 @@ def load_saved_model(export_dir: str):
     return state, index.get("config", {}), sigs
 
 
+def list_versions(base_dir: str):
+    """All numeric version subdirs holding a saved_model.pb, sorted."""
+    if not os.path.isdir(base_dir):
+        return []
+    return sorted(
+        int(d) for d in os.listdir(base_dir)
+        if d.isdigit() and
+        os.path.isfile(os.path.join(base_dir, d, "saved_model.pb")))
+
+
 def find_latest_version(base_dir: str):
     """Highest numeric version subdir (TF-Serving convention), or None."""
-    if not os.path.isdir(base_dir):
-        return None
-    versions = [int(d) for d in os.listdir(base_dir)
-                if d.isdigit() and
-                os.path.isfile(os.path.join(base_dir, d, "saved_model.pb"))]
-    return max(versions) if versions else None
+    versions = list_versions(base_dir)
+    return versions[-1] if versions else None

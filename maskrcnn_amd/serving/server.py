@@ -96,13 +96,10 @@ class ModelState:
     def batcher(self):
         return self.batchers[0] if self.batchers else None
 
-    def maybe_reload(self) -> bool:
-        latest = find_latest_version(self.base_path)
-        if latest is None or latest == self.version:
-            return False
-        export_dir = os.path.join(self.base_path, str(latest))
+    def _load_version(self, version: int):
+        export_dir = os.path.join(self.base_path, str(version))
         log.info("loading %s version %d from %s on %s", self.name,
-                 latest, export_dir, self.devices)
+                 version, export_dir, self.devices)
         engines = [InferenceEngine.from_saved_model(
             export_dir, device=dev, config=self.config_override)
             for dev in self.devices]
@@ -110,11 +107,28 @@ class ModelState:
             eng, max_batch_size=self.max_batch_size,
             batch_timeout_micros=self.batch_timeout_micros)
             for eng in engines]
+        return {"engines": engines, "batchers": batchers}
+
+    def maybe_reload(self) -> bool:
+        from .savedmodel import list_versions
+        # 'all': every on-disk version is servable (TF-Serving
+        # model_version_policy {all{}} — including at startup);
+        # 'latest': only the newest.
+        on_disk = list_versions(self.base_path)
+        if not on_disk:
+            return False
+        want = on_disk if self.version_policy == "all" else [on_disk[-1]]
+        new = {v: self._load_version(v) for v in want
+               if v not in self.versions}
+        latest = on_disk[-1]
+        if not new and latest == self.version:
+            return False
         with self._lock:
-            old = (self.versions.pop(self.version, None)
-                   if self.version_policy == "latest" else None)
-            self.versions[latest] = {"engines": engines,
-                                     "batchers": batchers}
+            self.versions.update(new)
+            old = None
+            if self.version_policy == "latest" and \
+                    self.version is not None and self.version != latest:
+                old = self.versions.pop(self.version, None)
             self.version = latest
         if old:
             for b in old["batchers"]:

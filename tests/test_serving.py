@@ -726,12 +726,12 @@ def test_version_policy_all(tmp_path):
     m = MaskRCNN("inference", cfg, model_dir=str(tmp_path))
     base = str(tmp_path / "models" / "vp")
     export_saved_model(m, cfg, base, version=1)
+    # Startup with BOTH versions already on disk: policy 'all' must
+    # load every one (TF-Serving {all{}} semantics), not just latest.
+    export_saved_model(m, cfg, base, version=2)
     st = ModelState("vp", base, max_batch_size=1,
                     batch_timeout_micros=100, config=cfg,
                     version_policy="all")
-    st.maybe_reload()
-    assert st.version == 1
-    export_saved_model(m, cfg, base, version=2)
     st.maybe_reload()
     assert st.version == 2
     assert sorted(st.versions) == [1, 2]
