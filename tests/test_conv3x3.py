@@ -297,3 +297,15 @@ def test_gpu_fp8_kernel_matches_reference(cin, cout, h, w):
     denom = ref.abs().clamp(min=1.0)
     assert float((err / denom).mean()) < 0.03
     assert float((err / denom).max()) < 0.5
+
+
+def test_conv_policy_matches_measured_table():
+    """The shape dispatch must follow the A/B table in
+    ops/conv_policy.py (profiles/r01_conv_ab.md)."""
+    from maskrcnn_amd.ops.conv_policy import use_hip_conv3x3
+    assert use_hip_conv3x3(2, 64, 256, 256)      # 1.26x ours
+    assert use_hip_conv3x3(2, 128, 128, 128)
+    assert use_hip_conv3x3(2, 256, 64, 64)
+    assert use_hip_conv3x3(400, 256, 14, 14)     # head shape
+    assert not use_hip_conv3x3(2, 512, 32, 32)   # MIOpen wins
+    assert not use_hip_conv3x3(2, 256, 256, 256)
