@@ -126,3 +126,27 @@ def test_staged_training_ddp_two_ranks():
     assert res[0]["losses_finite"] and res[1]["losses_finite"]
     assert res[0]["w_head"] == pytest.approx(res[1]["w_head"],
                                              abs=1e-6)
+
+
+def test_train_cli_resume_from_last(tmp_path):
+    """tools/train.py --model last resumes epoch numbering and restores
+    optimizer state (train_state.pth round trip through the CLI)."""
+    import glob
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+    def run(extra):
+        return subprocess.run(
+            [sys.executable, "tools/train.py", "--dataset", "shapes",
+             "--steps", "1", "--layers", "heads", "--num-workers", "0",
+             "--logs", str(tmp_path)] + extra,
+            capture_output=True, text=True, timeout=1200, cwd=root)
+
+    r = run(["--epochs", "1"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    r = run(["--epochs", "2", "--model", "last"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    ckpts = sorted(glob.glob(str(tmp_path / "*" / "mask_rcnn_*.pth")))
+    assert any(c.endswith("0001.pth") for c in ckpts), ckpts
