@@ -150,3 +150,24 @@ def test_train_cli_resume_from_last(tmp_path):
     assert r.returncode == 0, r.stderr[-2000:]
     ckpts = sorted(glob.glob(str(tmp_path / "*" / "mask_rcnn_*.pth")))
     assert any(c.endswith("0001.pth") for c in ckpts), ckpts
+
+
+def test_shapes_sample_cli(tmp_path):
+    """samples/shapes.py train -> detect end to end (the reference's
+    train_shapes notebook flow as a script)."""
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "samples/shapes.py", "train", "--epochs", "1",
+         "--steps", "1", "--layers", "heads", "--logs", str(tmp_path)],
+        capture_output=True, text=True, timeout=1200, cwd=root)
+    assert r.returncode == 0, r.stderr[-2000:]
+    out_png = str(tmp_path / "det.png")
+    r = subprocess.run(
+        [sys.executable, "samples/shapes.py", "detect", "--model",
+         "last", "--logs", str(tmp_path), "--out", out_png],
+        capture_output=True, text=True, timeout=1200, cwd=root)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert os.path.exists(out_png)
