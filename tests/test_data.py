@@ -134,3 +134,50 @@ def test_loader_crop_mode_end_to_end():
     assert s["gt_masks"].shape == (*cfg.MINI_MASK_SHAPE,
                                    cfg.MAX_GT_INSTANCES)
     assert torch.isfinite(s["rpn_bbox"]).all()
+
+
+def test_balloon_color_splash_and_via_loading(tmp_path):
+    """balloon sample: color_splash math + VIA-json dataset loading
+    (reference samples/balloon parity)."""
+    import importlib.util
+    import json as _json
+    import os
+
+    import numpy as np
+
+    spec = importlib.util.spec_from_file_location(
+        "balloon", os.path.join(os.path.dirname(__file__), "..",
+                                "samples", "balloon.py"))
+    balloon = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(balloon)
+
+    img = (np.random.rand(20, 20, 3) * 255).astype(np.uint8)
+    mask = np.zeros((20, 20, 1), bool)
+    mask[5:10, 5:10, 0] = True
+    out = balloon.color_splash(img, mask)
+    assert out.shape == img.shape
+    np.testing.assert_array_equal(out[6, 6], img[6, 6])   # kept color
+    assert not np.array_equal(out[0, 0], img[0, 0]) or \
+        img[0, 0, 0] == img[0, 0, 1] == img[0, 0, 2]      # grayed
+    # No-mask case -> fully gray.
+    out2 = balloon.color_splash(img, np.zeros((20, 20, 0), bool))
+    assert (out2[..., 0] == out2[..., 1]).all()
+
+    # Minimal VIA dataset: one image with one polygon region.
+    sub = tmp_path / "train"
+    sub.mkdir(parents=True)
+    from PIL import Image
+    Image.fromarray(img).save(sub / "img0.png")
+    via = {"x": {"filename": "img0.png",
+                 "regions": {"0": {"shape_attributes": {
+                     "name": "polygon",
+                     "all_points_x": [5, 15, 15, 5],
+                     "all_points_y": [5, 5, 15, 15]}}}}}
+    (sub / "via_region_data.json").write_text(_json.dumps(via))
+    ds = balloon.BalloonDataset()
+    ds.load_balloon(str(tmp_path), "train")
+    ds.prepare()
+    assert ds.num_images == 1
+    masks, cids = ds.load_mask(0)
+    assert masks.shape == (20, 20, 1) and cids.tolist() == [1]
+    assert masks[:, :, 0].sum() > 50  # the polygon filled
