@@ -181,3 +181,38 @@ def test_balloon_color_splash_and_via_loading(tmp_path):
     masks, cids = ds.load_mask(0)
     assert masks.shape == (20, 20, 1) and cids.tolist() == [1]
     assert masks[:, :, 0].sum() > 50  # the polygon filled
+
+
+def test_nucleus_dataset_loading(tmp_path):
+    """nucleus sample: DSB2018 per-instance mask-dir layout."""
+    import importlib.util
+    import os
+
+    import numpy as np
+    from PIL import Image
+
+    spec = importlib.util.spec_from_file_location(
+        "nucleus", os.path.join(os.path.dirname(__file__), "..",
+                                "samples", "nucleus.py"))
+    nucleus = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(nucleus)
+
+    iid = "img_a"
+    (tmp_path / iid / "images").mkdir(parents=True)
+    (tmp_path / iid / "masks").mkdir()
+    img = (np.random.rand(32, 32, 3) * 255).astype(np.uint8)
+    Image.fromarray(img).save(tmp_path / iid / "images" / f"{iid}.png")
+    for k in range(2):
+        m = np.zeros((32, 32), np.uint8)
+        m[4 + 10 * k:9 + 10 * k, 4:9] = 255
+        Image.fromarray(m).save(tmp_path / iid / "masks" / f"m{k}.png")
+
+    ds = nucleus.NucleusDataset()
+    ds.load_nucleus(str(tmp_path))
+    ds.prepare()
+    assert ds.num_images == 1
+    assert ds.load_image(0).shape == (32, 32, 3)
+    masks, cids = ds.load_mask(0)
+    assert masks.shape == (32, 32, 2)
+    assert cids.tolist() == [1, 1]
+    assert int(masks[:, :, 0].sum()) == 25
