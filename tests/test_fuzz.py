@@ -151,3 +151,22 @@ def test_detection_syncfree_equivalence_fuzz(B, N, seed, min_conf):
         torch.testing.assert_close(
             syncfree[b][:ns].sort(dim=0).values,
             eager[b][:ne].sort(dim=0).values, atol=1e-6, rtol=1e-6)
+
+
+@given(st.integers(min_value=1, max_value=6),
+       st.integers(min_value=1, max_value=4),
+       st.integers(min_value=0, max_value=10 ** 6),
+       st.floats(min_value=-12, max_value=12))
+@settings(max_examples=40, deadline=None)
+def test_mx_quantize_roundtrip_fuzz(rows, blocks, seed, log_scale):
+    """MX block quant: bounded relative error at ANY magnitude (the
+    e8m0 exponent covers 2^-127..2^127)."""
+    import torch
+
+    from maskrcnn_amd.utils.fp8 import mx_dequantize, mx_quantize
+    torch.manual_seed(seed)
+    x = torch.randn(rows, blocks * 32) * (10.0 ** log_scale)
+    q, exp = mx_quantize(x, dim=-1)
+    deq = mx_dequantize(q, exp, dim=-1)
+    rel = (deq - x).abs() / x.abs().clamp(min=1e-30)
+    assert float(rel.mean()) < 0.04
