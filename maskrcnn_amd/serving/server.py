@@ -263,8 +263,9 @@ class ModelServer:
         removed = [s for s in self.models.values() if id(s) not in kept]
         self.models = new
         for st in removed:
-            for b in st.batchers:
-                b.close()
+            for slot in st.versions.values():  # every loaded version
+                for b in slot["batchers"]:
+                    b.close()
 
     # ----------------------------------------------------------- predict
     def _handle_predict(self, request: PredictRequest) -> PredictResponse:
