@@ -97,13 +97,14 @@ class InferenceEngine:
                 return self.model(images, metas, anchors,
                                   mode="inference")
         key = (tuple(images.shape), tuple(anchors.shape))
-        st = self._graphs.get(key)
-        if st is None:
-            st = self._capture(images, metas, anchors)
-        st["img"].copy_(images)
-        st["meta"].copy_(metas)
-        st["anc"].copy_(anchors)
-        st["graph"].replay()
+        with torch.cuda.device(self.device):  # replicas: cuda:N != current
+            st = self._graphs.get(key)
+            if st is None:
+                st = self._capture(images, metas, anchors)
+            st["img"].copy_(images)
+            st["meta"].copy_(metas)
+            st["anc"].copy_(anchors)
+            st["graph"].replay()
         return st["out"]
 
     def _capture(self, images, metas, anchors):
@@ -120,7 +121,7 @@ class InferenceEngine:
         ac = lambda: torch.autocast(  # noqa: E731
             device_type="cuda", dtype=torch.bfloat16,
             enabled=self.use_bf16, cache_enabled=False)
-        s = torch.cuda.Stream()
+        s = torch.cuda.Stream(device=self.device)
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(3):
