@@ -31,13 +31,21 @@ class TorchMaskRCNNDataset(_TorchDataset):
         self.anchor_cache = AnchorCache(config)
         self.anchors_px = self.anchor_cache.get_pixel(config.IMAGE_SHAPE)
         self.seed = seed
+        self.epoch = 0
+
+    def set_epoch(self, epoch: int):
+        """Vary augmentation/sampling randomness across epochs (the
+        reference's generator kept RNG state across passes; a pure
+        (seed, idx) stream would repeat the same flips every epoch)."""
+        self.epoch = int(epoch)
 
     def __len__(self):
         return len(self.dataset.image_ids)
 
     def __getitem__(self, idx):
         rng = np.random.RandomState(
-            (self.seed * 1_000_003 + idx) % (2 ** 31))
+            (self.seed * 1_000_003 + self.epoch * 7_919 + idx)
+            % (2 ** 31))
         for attempt in range(5):
             image_id = self.dataset.image_ids[idx]
             try:

@@ -206,6 +206,8 @@ class Trainer:
             if self.distributed and loader.sampler is not None and \
                     hasattr(loader.sampler, "set_epoch"):
                 loader.sampler.set_epoch(epoch)
+            if hasattr(loader.dataset, "set_epoch"):
+                loader.dataset.set_epoch(epoch)
             t0 = time.time()
             step = 0
             epoch_losses = {}

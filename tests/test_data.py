@@ -216,3 +216,31 @@ def test_nucleus_dataset_loading(tmp_path):
     assert masks.shape == (32, 32, 2)
     assert cids.tolist() == [1, 1]
     assert int(masks[:, :, 0].sum()) == 25
+
+
+def test_loader_augmentation_varies_across_epochs(tmp_path):
+    """set_epoch must change the augmentation stream (same index,
+    different epoch -> flips can differ)."""
+    import numpy as np
+
+    from maskrcnn_amd.data.loader import TorchMaskRCNNDataset
+    from maskrcnn_amd.data.synthetic import ShapesConfig, ShapesDataset
+
+    ds = ShapesDataset()
+    ds.load_shapes(2, 128, 128, seed=1)
+    ds.prepare()
+    tds = TorchMaskRCNNDataset(ds, ShapesConfig(),
+                               augmentation="fliplr")
+    base = tds[0]["image"].numpy()
+    changed = False
+    for e in range(1, 6):
+        tds.set_epoch(e)
+        if not np.array_equal(tds[0]["image"].numpy(), base):
+            changed = True
+            break
+    assert changed, "augmentation identical across 6 epochs"
+    # Reproducible: same epoch -> same sample.
+    tds.set_epoch(3)
+    a = tds[0]["image"].numpy()
+    tds.set_epoch(3)
+    assert np.array_equal(tds[0]["image"].numpy(), a)
