@@ -412,10 +412,27 @@ class MaskRCNN:
 
     @torch.no_grad()
     def detect_molded(self, molded_images, image_metas, verbose: int = 0):
-        """Detection on already-molded inputs (reference detect_molded)."""
-        anchors = self.anchor_cache.get(molded_images[0].shape,
-                                        device=self.device)
-        return self.predict_molded(molded_images, image_metas, anchors)
+        """Detection on already-molded inputs -> list of result dicts
+        (reference detect_molded:~2480 returns UNMOLDED results, same
+        contract as detect()). Use predict_molded for raw tensors."""
+        from ..utils.meta import parse_image_meta
+        outputs = self.predict_molded(
+            molded_images, image_metas,
+            self.anchor_cache.get(molded_images[0].shape,
+                                  device=self.device))
+        detections = outputs["mrcnn_detection"].cpu().numpy()
+        mrcnn_mask = outputs["mrcnn_mask"].cpu().numpy()
+        results = []
+        for i in range(molded_images.shape[0]):
+            meta = parse_image_meta(image_metas[i:i + 1])
+            boxes, class_ids, scores, masks = self.unmold_detections(
+                detections[i], mrcnn_mask[i],
+                tuple(int(v) for v in meta["original_image_shape"][0]),
+                tuple(int(v) for v in meta["image_shape"][0]),
+                tuple(int(v) for v in meta["window"][0]))
+            results.append({"rois": boxes, "class_ids": class_ids,
+                            "scores": scores, "masks": masks})
+        return results
 
     @torch.no_grad()
     def predict_molded(self, molded_images, image_metas, anchors):

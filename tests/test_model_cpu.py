@@ -26,7 +26,9 @@ def test_inference_output_signature(shapes_config):
     m = MaskRCNN("inference", cfg, model_dir="/tmp/mrcnn_test_logs")
     img = (np.random.rand(128, 128, 3) * 255).astype(np.uint8)
     molded, metas, _ = m.mold_inputs([img])
-    out = m.detect_molded(molded, metas)
+    out = m.predict_molded(
+        molded, metas,
+        m.anchor_cache.get(molded[0].shape, device=m.device))
     D = cfg.DETECTION_MAX_INSTANCES
     C = cfg.NUM_CLASSES
     N = cfg.POST_NMS_ROIS_INFERENCE
@@ -164,3 +166,19 @@ def test_detect_pad64_nonsquare(shapes_config, tmp_path):
     molded, _, windows = m.mold_inputs([img])
     assert molded.shape[1] % 64 == 0 and molded.shape[2] % 64 == 0
     assert molded.shape[1] != molded.shape[2]  # stayed non-square
+
+
+def test_detect_molded_returns_unmolded_results(shapes_config):
+    """detect_molded matches the reference contract: a list of result
+    dicts (rois/class_ids/scores/masks), same as detect()."""
+    m = MaskRCNN("inference", shapes_config, model_dir="/tmp/dm_logs")
+    img = (np.random.RandomState(2).rand(100, 90, 3) * 255) \
+        .astype(np.uint8)
+    molded, metas, _ = m.mold_inputs([img])
+    res = m.detect_molded(molded, metas)
+    via_detect = m.detect([img])
+    assert isinstance(res, list) and len(res) == 1
+    r = res[0]
+    assert set(r) == {"rois", "class_ids", "scores", "masks"}
+    assert r["masks"].shape[:2] == img.shape[:2]
+    np.testing.assert_allclose(r["rois"], via_detect[0]["rois"])

@@ -169,7 +169,9 @@ def test_engine_from_saved_model(shapes_config, tmp_path):
     assert out["mrcnn_mask"].shape == (1, D, 28, 28, C)
 
     # Engine outputs must match the facade's predict path.
-    ref = m.detect_molded(molded, metas)
+    ref = m.predict_molded(
+        molded, metas,
+        m.anchor_cache.get(molded[0].shape, device=m.device))
     np.testing.assert_allclose(
         out["mrcnn_detection"], ref["mrcnn_detection"].cpu().numpy(),
         atol=1e-4)
