@@ -394,7 +394,16 @@ class MaskRCNN:
         assert self.mode == "inference", "Create model in inference mode."
         assert len(images) == self.config.BATCH_SIZE, \
             "len(images) must equal BATCH_SIZE"
+        if verbose:
+            from ..utils.meta import log as _log
+            _log(f"Processing {len(images)} images")
+            for image in images:
+                _log("image", image)
         molded_images, image_metas, windows = self.mold_inputs(images)
+        if verbose:
+            from ..utils.meta import log as _log
+            _log("molded_images", molded_images)
+            _log("image_metas", image_metas)
         anchors = self.anchor_cache.get(molded_images[0].shape,
                                         device=self.device)
         outputs = self.predict_molded(molded_images, image_metas, anchors)

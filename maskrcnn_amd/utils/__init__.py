@@ -18,7 +18,7 @@ from .masks import (  # noqa: F401
     resize_mask, minimize_mask, expand_mask, mold_mask, unmold_mask,
 )
 from .meta import (  # noqa: F401
-    compose_image_meta, parse_image_meta, parse_image_meta_t,
+    compose_image_meta, log, parse_image_meta, parse_image_meta_t,
 )
 from .metrics import (  # noqa: F401
     compute_matches, compute_ap, compute_ap_range, compute_recall,

@@ -55,3 +55,17 @@ def parse_image_meta_t(meta: torch.Tensor) -> dict:
         "scale": meta[:, 11],
         "active_class_ids": meta[:, 12:],
     }
+
+
+def log(text, array=None):
+    """Print a text message + optional array stats (reference
+    mrcnn/model.py log():~23 — the notebook debugging helper)."""
+    if array is not None:
+        import numpy as np
+        a = np.asarray(array)
+        text = text.ljust(25)
+        text += (f"shape: {str(a.shape):20}  "
+                 f"min: {a.min():10.5f}  max: {a.max():10.5f}"
+                 if a.size else f"shape: {str(a.shape):20}  (empty)")
+        text += f"  {a.dtype}"
+    print(text)
