@@ -170,3 +170,47 @@ def test_mx_quantize_roundtrip_fuzz(rows, blocks, seed, log_scale):
     deq = mx_dequantize(q, exp, dim=-1)
     rel = (deq - x).abs() / x.abs().clamp(min=1e-30)
     assert float(rel.mean()) < 0.04
+
+
+@given(st.integers(min_value=1, max_value=3),
+       st.integers(min_value=20, max_value=120),
+       st.integers(min_value=5, max_value=40),
+       st.integers(min_value=0, max_value=10 ** 6))
+@settings(max_examples=15, deadline=None)
+def test_proposal_padded_scatter_fuzz(B, A, count, seed):
+    """ProposalLayer (padded NMS + masked scatter) == reference
+    per-image loop across random shapes/counts."""
+    import torch
+
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model.proposal import ProposalLayer
+    from maskrcnn_amd.ops import nms
+    from maskrcnn_amd.utils.boxes import apply_box_deltas_t, clip_boxes_t
+
+    class C(ShapesConfig):
+        PRE_NMS_LIMIT = 50
+        RPN_NMS_THRESHOLD = 0.7
+
+    cfg = C()
+    torch.manual_seed(seed)
+    layer = ProposalLayer(cfg)
+    probs = torch.softmax(torch.randn(B, A, 2), dim=-1)
+    deltas = torch.randn(B, A, 4) * 0.2
+    anchors = torch.rand(A, 4)
+    anchors[:, 2:] = anchors[:, :2] + 0.1
+    rois = layer(probs, deltas, anchors, count)
+    assert rois.shape == (B, count, 4)
+
+    scores = probs[:, :, 1]
+    d = deltas * torch.tensor(cfg.RPN_BBOX_STD_DEV)
+    lim = min(cfg.PRE_NMS_LIMIT, A)
+    for b in range(B):
+        top_s, ix = torch.topk(scores[b], lim, sorted=True)
+        boxes = apply_box_deltas_t(anchors[ix], d[b, ix])
+        boxes = clip_boxes_t(boxes, torch.tensor([0., 0., 1., 1.]))
+        keep = nms(boxes, top_s, cfg.RPN_NMS_THRESHOLD, max_out=count)
+        expect = boxes[keep]
+        got = rois[b, :keep.shape[0]]
+        torch.testing.assert_close(got, expect, atol=1e-6, rtol=1e-6)
+        # Padding rows zeroed.
+        assert float(rois[b, keep.shape[0]:].abs().sum()) == 0.0
