@@ -171,3 +171,22 @@ def test_shapes_sample_cli(tmp_path):
         capture_output=True, text=True, timeout=1200, cwd=root)
     assert r.returncode == 0, r.stderr[-2000:]
     assert os.path.exists(out_png)
+
+
+def test_bench_script_single_process(tmp_path):
+    """The driver's N=1 invocation shape (plain python, no torchrun)."""
+    import json
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    proc = subprocess.run(
+        [sys.executable, "bench.py", "--gpus", "1", "--steps", "1",
+         "--warmup", "0", "--batch", "1", "--image-size", "256",
+         "--backbone", "resnet50"],
+        capture_output=True, text=True, timeout=900, cwd=root)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    rec = json.loads([ln for ln in proc.stdout.splitlines()
+                      if ln.startswith("{")][0])
+    assert rec["n_gpus"] == 1 and rec["value"] > 0
+    assert rec["data"] == "synthetic"
