@@ -29,6 +29,8 @@ def resize(image: np.ndarray, output_shape, order=1, preserve_range=True,
     if image.shape[0] == oh and image.shape[1] == ow:
         return image.copy()
     arr = np.asarray(image)
+    if not arr.flags.writeable:  # e.g. PIL-backed arrays
+        arr = arr.copy()
     squeeze = False
     if arr.ndim == 2:
         arr = arr[:, :, None]
