@@ -18,7 +18,7 @@ block1)  # v3 + wrw + fp8 kernel validation + conv A/B (~5 min)
   grep -v libdrm gpurun_out/r2_conv_ab.log
   # hipGraph serving replay (engine._forward capture path)
   timeout 300 python -m pytest \
-      tests/test_gpu_integration.py::test_engine_hipgraph_matches_eager \
+      tests/test_zz_hipgraph_gpu.py::test_engine_hipgraph_matches_eager \
       -m gpu -q --runxfail >> gpurun_out/r2_conv_tests.log 2>&1
   tail -2 gpurun_out/r2_conv_tests.log
   ;;
