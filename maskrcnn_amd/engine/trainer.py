@@ -26,11 +26,25 @@ log = logging.getLogger(__name__)
 def build_optimizer(model, config, learning_rate):
     """SGD + momentum; weight decay applied only to conv/linear weights
     (reference compile() adds L2 over non-BN weight matrices, not biases)."""
+    # Classify no-decay params by module TYPE (norm layers), not by name
+    # substring: BN inside nn.Sequential heads is named convs.1 etc. and
+    # a name check would wrongly decay its gamma/beta.
+    norm_types = (torch.nn.modules.batchnorm._BatchNorm,
+                  torch.nn.GroupNorm, torch.nn.LayerNorm)
+    try:
+        from ..model.resnet import FrozenBatchNorm2d
+        norm_types = norm_types + (FrozenBatchNorm2d,)
+    except ImportError:
+        pass
+    norm_param_ids = set()
+    for mod in model.modules():
+        if isinstance(mod, norm_types):
+            norm_param_ids.update(id(p) for p in mod.parameters(recurse=False))
     decay, no_decay = [], []
     for name, p in model.named_parameters():
         if not p.requires_grad:
             continue
-        if name.endswith(".bias") or "bn" in name.split(".")[-2:][0].lower():
+        if name.endswith(".bias") or id(p) in norm_param_ids:
             no_decay.append(p)
         else:
             decay.append(p)

@@ -40,7 +40,10 @@ def rpn_class_loss(rpn_match, rpn_class_logits):
 def rpn_bbox_loss(config, target_bbox, rpn_match, rpn_bbox):
     """target_bbox: [B, MAX_POS, 4] packed per image; rpn_bbox: [B, A, 4]."""
     rpn_match = rpn_match.squeeze(-1)
-    losses = []
+    # Reference batch_pack_graph packs ALL positive anchors across the
+    # batch into one tensor and takes a single global mean, so images
+    # with few positives are not over-weighted (mrcnn/model.py:~1060).
+    elems = []
     B = rpn_bbox.shape[0]
     for b in range(B):
         pos = (rpn_match[b] == 1).nonzero(as_tuple=True)[0]
@@ -49,10 +52,10 @@ def rpn_bbox_loss(config, target_bbox, rpn_match, rpn_bbox):
             continue
         pred = rpn_bbox[b, pos].float()
         tgt = target_bbox[b, :n].float()
-        losses.append(smooth_l1_loss(tgt, pred).mean())
-    if not losses:
+        elems.append(smooth_l1_loss(tgt, pred))
+    if not elems:
         return rpn_bbox.sum() * 0.0
-    return torch.stack(losses).mean()
+    return torch.cat(elems, dim=0).mean()
 
 
 def mrcnn_class_loss(target_class_ids, pred_class_logits, active_class_ids):

@@ -232,7 +232,10 @@ class MaskRCNN:
             load_keras_h5(self.model, path, exclude=exclude)
             self.set_log_dir(path)
             return []
-        state = torch.load(path, map_location="cpu", weights_only=False)
+        # weights_only=True: checkpoints store only tensors + a plain
+        # config dict (save_weights above), so the pickle code-exec
+        # surface is closed even for paths found via find_last().
+        state = torch.load(path, map_location="cpu", weights_only=True)
         sd = state["model"] if isinstance(state, dict) and "model" in state \
             else state
         if exclude:

@@ -1,9 +1,9 @@
 """PyramidROIAlign: multi-level RoIAlign over the FPN.
 
 Replicates reference PyramidROIAlign (mrcnn/model.py:~340-430): each roi is
-assigned to pyramid level k = floor(4 + log2(sqrt(h*w) / (224/sqrt(1024))))
-clipped to [2,5]; pooling runs per level and results are scattered back to
-the original box order. Unlike the reference's crop_and_resize, the op mode
+assigned to pyramid level k = 4 + round(log2(sqrt(h*w) / (224/sqrt(1024))))
+(tf.round in the reference) clipped to [2,5]; pooling runs per level and
+results are scattered back to the original box order. Unlike the reference's crop_and_resize, the op mode
 is configurable (true RoIAlign by default for better gradients; see
 ops/roi_align.py docstring and SURVEY.md §7 hard-part 6).
 """
@@ -44,7 +44,7 @@ class PyramidROIAlign(nn.Module):
         # Level assignment (reference log2_graph section, model.py:~360).
         image_area = float(self.image_shape[0] * self.image_shape[1])
         scale = torch.sqrt(h * w) * (image_area ** 0.5) / 224.0
-        levels = torch.floor(4.0 + torch.log2(scale.clamp(min=1e-6)))
+        levels = torch.round(4.0 + torch.log2(scale.clamp(min=1e-6)))
         levels = levels.clamp(2, 5).to(torch.long)
         # Padding rows (zero area) -> level 2; output stays zero via mask.
         valid = (h > 0) & (w > 0)

@@ -72,8 +72,9 @@ def test_gpu_vs_cpu_detection_consistency(shapes_config):
     # Random-init nets amplify bf16-vs-fp32 argmax flips, so the counts
     # are only sanity-compared; geometry must be valid either way.
     assert (n_gpu == 0) == (n_cpu == 0) or True
-    assert gpu_det[:n_gpu, :4].min() >= -1e-3
-    assert gpu_det[:n_gpu, :4].max() <= 1 + 1e-3
+    if n_gpu:  # random-init nets may legitimately yield 0 detections
+        assert gpu_det[:n_gpu, :4].min() >= -1e-3
+        assert gpu_det[:n_gpu, :4].max() <= 1 + 1e-3
     assert np.isfinite(cpu_det).all() and np.isfinite(gpu_det).all()
 
 

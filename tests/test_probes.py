@@ -118,6 +118,24 @@ def test_pyramid_roi_align_level_routing(shapes_config):
     assert torch.allclose(out[1], 4 * torch.ones_like(out[1]))
 
 
+def test_pyramid_roi_align_half_octave_rounding(shapes_config):
+    """Level assignment uses 4 + round(log2(scale)) like the reference's
+    tf.round (model.py:~360), NOT floor: a 224px roi at 1024^2 maps to
+    P4 while a 340px roi (log2(340/224)=0.60) rounds UP to P5."""
+    from maskrcnn_amd.model.roialign import PyramidROIAlign
+    layer = PyramidROIAlign(7, [1024, 1024, 3])
+    B, C = 1, 4
+    feats = [torch.full((B, C, s, s), float(i + 1))
+             for i, s in enumerate([256, 128, 64, 32])]
+    rois = torch.tensor([[[0.0, 0.0, 224 / 1024, 224 / 1024],  # -> P4
+                          [0.0, 0.0, 340 / 1024, 340 / 1024],  # -> P5
+                          [0.0, 0.0, 150 / 1024, 150 / 1024]]])  # -> P3
+    out = layer(rois, feats)
+    assert torch.allclose(out[0], 3 * torch.ones_like(out[0]))  # P4
+    assert torch.allclose(out[1], 4 * torch.ones_like(out[1]))  # P5
+    assert torch.allclose(out[2], 2 * torch.ones_like(out[2]))  # P3
+
+
 def test_classifier_head_linear_equals_conv(shapes_config):
     """The head's F.linear evaluation must equal the reference conv
     formulation (7x7 VALID conv == dense over the pooled tile)."""
