@@ -20,6 +20,13 @@ void roi_align_bwd_launch(const void*, const float*, const int*, float*, int,
                           int, int, int, int, int, long, long, long, long,
                           long, long, long, long, int, bool, bool,
                           hipStream_t);
+void roi_align_bwd_det_launch(const void*, const float*, const int*,
+                              unsigned long long*, float*, int, int, int,
+                              int, int, int, long, long, long, long, long,
+                              long, long, long, int, bool, bool, float,
+                              hipStream_t);
+void fixed_to_float_launch(const unsigned long long*, float*, long, float,
+                           hipStream_t);
 long nms_workspace_bytes(int N);
 void nms_launch(const float*, int, float, int, void*, int*, int*,
                 hipStream_t);
@@ -109,6 +116,53 @@ torch::Tensor roi_align_backward(torch::Tensor grad_out, torch::Tensor boxes,
       (int)pool_w, grad_in.stride(0), grad_in.stride(1), grad_in.stride(2),
       grad_in.stride(3), grad_out.stride(0), grad_out.stride(1),
       grad_out.stride(2), grad_out.stride(3), samp, cl, bf16, cur_stream());
+  return grad_in;
+}
+
+torch::Tensor roi_align_backward_det(torch::Tensor grad_out,
+                                     torch::Tensor boxes,
+                                     torch::Tensor box_ind, int64_t B,
+                                     int64_t C, int64_t H, int64_t W,
+                                     int64_t pool_h, int64_t pool_w,
+                                     int64_t sampling, bool aligned) {
+  // Deterministic-reduction variant: fixed-point i64 atomics (integer
+  // addition is associative -> bitwise-stable grads across runs).
+  // One host sync (amax) per call — test/debug mode, not the hot path.
+  TORCH_CHECK(grad_out.is_cuda(), "grad_out must be on GPU");
+  const bool cl = is_channels_last(grad_out);
+  if (!cl) grad_out = grad_out.contiguous();
+  auto boxes_f = boxes.to(torch::kFloat).contiguous();
+  auto ind = box_ind.to(torch::kInt).contiguous();
+  const int N = boxes.size(0);
+  auto grad_in = torch::zeros(
+      {B, C, H, W},
+      grad_out.options().dtype(torch::kFloat)
+          .memory_format(cl ? at::MemoryFormat::ChannelsLast
+                            : at::MemoryFormat::Contiguous));
+  if (N == 0) return grad_in;
+  auto acc = torch::zeros({B * C * H * W},
+                          grad_out.options().dtype(torch::kLong));
+  const float amax =
+      grad_out.abs().amax().to(torch::kFloat).item<float>();
+  if (amax == 0.f) return grad_in;
+  // Worst case every contribution (<= amax each) lands on one cell:
+  // N*PH*PW*S^2*4 <= ~2^21 for N<=2000 — keep the running sum < 2^62.
+  const float scale = exp2f(41.f - ceilf(log2f(amax)));
+  const bool bf16 = grad_out.scalar_type() == torch::kBFloat16;
+  const int samp = aligned ? std::max<int>(1, sampling) : 0;
+  roi_align_bwd_det_launch(
+      grad_out.data_ptr(), boxes_f.data_ptr<float>(), ind.data_ptr<int>(),
+      (unsigned long long*)acc.data_ptr<int64_t>(),
+      grad_in.data_ptr<float>(), N, (int)C, (int)H, (int)W, (int)pool_h,
+      (int)pool_w, grad_in.stride(0), grad_in.stride(1), grad_in.stride(2),
+      grad_in.stride(3), grad_out.stride(0), grad_out.stride(1),
+      grad_out.stride(2), grad_out.stride(3), samp, cl, bf16, scale,
+      cur_stream());
+  // acc is indexed with grad_in's strides; the flat conversion is
+  // layout-agnostic (same linear buffer).
+  fixed_to_float_launch((const unsigned long long*)acc.data_ptr<int64_t>(),
+                        grad_in.data_ptr<float>(), (long)B * C * H * W,
+                        1.f / scale, cur_stream());
   return grad_in;
 }
 
@@ -222,8 +276,11 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
       x.options().memory_format(at::MemoryFormat::ChannelsLast));
   const long M = (long)B * H * W;
   int ver = (int)version;
-  if (ver == 0)  // auto: the 128x128 glds kernels need enough M x N tiles
-    ver = (M >= 32768 && Cout >= 128) ? 2 : 1;
+  if (ver == 0)  // auto: the 128x128 glds kernels need enough M x N tiles.
+    // v3 (3-buffer counted-vmcnt + XCD swizzle) beats v2 at every
+    // large-M shape in the round-2 block1 A/B (gpurun_out/r2_conv_ab):
+    // 641 vs 603 TF @(2,256,256^2,256), 635 vs 589 @(400,256,14^2,256).
+    ver = (M >= 32768 && Cout >= 128) ? 3 : 1;
   if (ver >= 2) {
     // Persistent 64-B zero page per device (glds cannot predicate
     // out-of-range lanes; they DMA from here instead).
@@ -349,6 +406,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "RoIAlign / crop_and_resize forward (CDNA4)");
   m.def("roi_align_backward", &roi_align_backward,
         "RoIAlign backward (CDNA4, atomic scatter)");
+  m.def("roi_align_backward_det", &roi_align_backward_det,
+        "RoIAlign backward, deterministic fixed-point i64 reduction");
   m.def("nms", &nms, "bitmask greedy NMS (CDNA4)");
   m.def("nms_padded", &nms_padded,
         "sync-free NMS: (-1-padded keep [max_out], device count)",

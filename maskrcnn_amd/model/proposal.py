@@ -15,7 +15,7 @@ import torch
 import torch.nn as nn
 
 from ..ops import batched_nms_padded
-from ..utils.boxes import apply_box_deltas_t, clip_boxes_t
+from ..utils.boxes import apply_box_deltas_t
 
 
 class ProposalLayer(nn.Module):
@@ -50,8 +50,10 @@ class ProposalLayer(nn.Module):
                                  ix.unsqueeze(-1).expand(-1, -1, 4))
 
         boxes = apply_box_deltas_t(anchors_g, deltas)
-        window = torch.tensor([0., 0., 1., 1.], device=boxes.device)
-        boxes = clip_boxes_t(boxes, window)
+        # Clip to the unit window. clamp (not clip_boxes_t with a fresh
+        # torch.tensor) — creating a tensor from host data is a H2D copy,
+        # which aborts hipGraph stream capture of the serving forward.
+        boxes = boxes.clamp(0.0, 1.0)
 
         # NMS for a GROUP of images per launch (image-index offset
         # trick): per-image NMS serializes at serving batch 32, but one

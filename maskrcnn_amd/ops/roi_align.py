@@ -41,11 +41,18 @@ class _RoiAlignHIP(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_out):
+        import os
         ext = require_ext()
         boxes, box_ind = ctx.saved_tensors
         pool_h, pool_w, sampling_ratio, aligned = ctx.params
         B, C, H, W = ctx.input_shape
-        grad_in = ext.roi_align_backward(
+        # MASKRCNN_DETERMINISTIC=1: fixed-point i64 reduction — grads are
+        # bitwise-stable across runs (fp atomicAdd order is not). Costs
+        # one host sync + an i64 workspace; test/debug mode.
+        fn = (ext.roi_align_backward_det
+              if os.environ.get("MASKRCNN_DETERMINISTIC") == "1"
+              else ext.roi_align_backward)
+        grad_in = fn(
             grad_out.contiguous(), boxes, box_ind.int(), B, C, H, W,
             pool_h, pool_w, sampling_ratio, aligned)
         return grad_in.to(ctx.in_dtype), None, None, None, None, None, None

@@ -84,17 +84,8 @@ def test_gpu_kernel_matches_reference(cin, cout, h, w):
     torch.testing.assert_close(out.float(), ref, atol=0.3, rtol=0.05)
 
 
-def _xfail_v3(version):
-    """v3 (counted-vmcnt schedule) was written after round-1 GPU budget
-    ran out: exercised but not yet hardware-validated — don't let a v3
-    failure abort the -x gpu suite before its first measurement pass."""
-    return pytest.param(version, marks=pytest.mark.xfail(
-        strict=False, reason="v3 pending first GPU validation")) \
-        if version == 3 else version
-
-
 @pytest.mark.gpu
-@pytest.mark.parametrize("version", [2, _xfail_v3(3)])
+@pytest.mark.parametrize("version", [2, 3])
 @pytest.mark.parametrize("cin,cout,h,w,batch", [
     (256, 256, 64, 64, 2),    # exact-tile case
     (256, 256, 50, 30, 3),    # ragged M
@@ -119,8 +110,6 @@ def test_gpu_v2_kernel_matches_reference(cin, cout, h, w, batch, version):
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False,
-                   reason="v3 pending first GPU validation")
 def test_gpu_v3_repeated_runs_stable():
     """Race screen for the counted-vmcnt schedule: byte-identical results
     across repeated launches (cdna_hip_programming.md two-lane
@@ -206,8 +195,6 @@ def test_gpu_dgrad_via_forward_kernel():
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False,
-                   reason="wrw kernel pending first GPU validation")
 @pytest.mark.parametrize("cin,cout,h,w,batch", [
     (64, 64, 64, 48, 2),      # early resnet, ragged W segment
     (256, 256, 32, 32, 2),    # FPN shape, exact segments
@@ -236,8 +223,6 @@ def test_gpu_wrw_kernel_matches_reference(cin, cout, h, w, batch):
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False,
-                   reason="wrw kernel pending first GPU validation")
 def test_gpu_autograd_uses_hip_wrw(monkeypatch):
     """MASKRCNN_HIP_WRW=1 routes the Function's weight grad through the
     kernel and matches the default aten path."""
@@ -271,8 +256,6 @@ def test_pack_weight_fp8_roundtrip():
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False,
-                   reason="fp8 kernel pending first GPU validation")
 @pytest.mark.parametrize("cin,cout,h,w", [
     (256, 256, 32, 32),
     (64, 64, 64, 48),

@@ -97,8 +97,6 @@ def test_resume_epoch_parse(tmp_path, shapes_config):
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False,
-                   reason="fp8 kernel pending first GPU validation")
 def test_mask_head_fp8_close_to_bf16(shapes_config, monkeypatch):
     """MASKRCNN_FP8_COMPUTE routes the mask-head conv stack through the
     e4m3 kernel; detections' masks must stay close to the bf16 path."""

@@ -83,6 +83,41 @@ def test_roi_align_bwd_matches_cpu(dev, mode):
                                rtol=1e-3)
 
 
+@pytest.mark.parametrize("mode", ["crop_and_resize", "aligned"])
+def test_roi_align_bwd_deterministic(dev, mode, monkeypatch):
+    """MASKRCNN_DETERMINISTIC=1 routes backward through the fixed-point
+    i64 reduction: bitwise-identical grads across repeated runs (the fp
+    atomic path is order-dependent), and close to the CPU fp32 oracle.
+    (SURVEY §5 sanitizers row)"""
+    _require_ext()
+    from maskrcnn_amd.ops import roi_align
+    rng = np.random.RandomState(7)
+    feats = torch.randn(2, 8, 12, 12)
+    # Heavily overlapping boxes -> contended atomic targets.
+    boxes = torch.from_numpy(
+        np.clip(_rand_boxes(200, rng) * 0.3 + 0.3, 0, 1).astype(np.float32))
+    ind = torch.from_numpy(rng.randint(0, 2, 200).astype(np.int64))
+
+    f_cpu = feats.clone().requires_grad_(True)
+    out_cpu = roi_align(f_cpu, boxes, ind, (5, 5), mode=mode)
+    g = torch.randn_like(out_cpu)
+    out_cpu.backward(g)
+
+    monkeypatch.setenv("MASKRCNN_DETERMINISTIC", "1")
+    grads = []
+    for _ in range(3):
+        f_gpu = feats.to(dev).requires_grad_(True)
+        out_gpu = roi_align(f_gpu, boxes.to(dev), ind.to(dev), (5, 5),
+                            mode=mode)
+        out_gpu.backward(g.to(dev))
+        grads.append(f_gpu.grad.cpu())
+    # Bitwise-stable across runs.
+    for other in grads[1:]:
+        assert torch.equal(grads[0], other)
+    # And numerically right (fixed-point quantization ~2^-41 rel amax).
+    torch.testing.assert_close(grads[0], f_cpu.grad, atol=1e-3, rtol=1e-3)
+
+
 def test_roi_align_out_of_bounds_boxes(dev):
     """Mini-mask-relative boxes can exceed [0,1]; zeros outside."""
     _require_ext()
@@ -178,9 +213,6 @@ def test_nms_padded_gpu_matches_cpu(dev):
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False,
-                   reason="routes through detection_prep — pending "
-                          "first GPU validation")
 def test_detection_syncfree_gpu_matches_eager(dev):
     """Capture-safe DetectionLayer path on GPU (detection_prep + HIP
     nms_padded) equals the eager nonzero/gather path."""
@@ -223,8 +255,6 @@ def test_detection_syncfree_gpu_matches_eager(dev):
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False,
-                   reason="detection_prep pending first GPU validation")
 def test_detection_prep_kernel_matches_torch(dev):
     """Fused HIP prep vs the layer's torch prep block."""
     _require_ext()
