@@ -47,6 +47,8 @@ void detection_prep_launch(const void*, const void*, const void*,
                            const float*, const float*, float*, int*,
                            float*, int, int, int, float, bool,
                            hipStream_t);
+void deconv2x2_launch(const void*, const void*, const float*, const void*,
+                      void*, int, int, int, int, int, bool, hipStream_t);
 }
 
 namespace {
@@ -303,6 +305,39 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
   return out;
 }
 
+torch::Tensor deconv2x2(torch::Tensor x, torch::Tensor w_packed,
+                        torch::Tensor bias, bool relu) {
+  // x: [B,Cin,H,W] channels_last bf16; w_packed: [4, Cout, Cin] bf16
+  // tap-major (tap = di*2+dj); bias: [Cout] f32. Out: [B,Cout,2H,2W]
+  // channels_last (2x2 stride-2 transposed conv).
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "x must be [B,C,H,W] cuda");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+  TORCH_CHECK(is_channels_last(x), "x must be channels_last (NHWC)");
+  TORCH_CHECK(w_packed.dim() == 3 && w_packed.size(0) == 4,
+              "w_packed must be [4, Cout, Cin]");
+  TORCH_CHECK(w_packed.scalar_type() == torch::kBFloat16 &&
+                  w_packed.is_contiguous(),
+              "w_packed must be contiguous bf16");
+  const int B = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  const int Cout = w_packed.size(1);
+  TORCH_CHECK(w_packed.size(2) == Cin, "weight Cin mismatch");
+  TORCH_CHECK(Cin % 32 == 0, "Cin must be a multiple of 32");
+  auto bias_f = bias.to(torch::kFloat).contiguous();
+  auto out = torch::empty(
+      {B, Cout, 2 * H, 2 * W},
+      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  static std::unordered_map<int, torch::Tensor> zero_pages;
+  const int dev = x.get_device();
+  auto it = zero_pages.find(dev);
+  if (it == zero_pages.end())
+    it = zero_pages.emplace(dev, torch::zeros({32}, x.options())).first;
+  deconv2x2_launch(x.data_ptr(), w_packed.data_ptr(),
+                   bias_f.data_ptr<float>(), it->second.data_ptr(),
+                   out.data_ptr(), B, H, W, Cin, Cout, relu,
+                   cur_stream());
+  return out;
+}
+
 torch::Tensor conv3x3_fp8(torch::Tensor x, torch::Tensor w_packed,
                           torch::Tensor bias, torch::Tensor w_scale,
                           torch::Tensor a_scale, bool relu) {
@@ -431,6 +466,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused argmax/class-delta/clip detection prep (N9)",
         py::arg("rois"), py::arg("probs"), py::arg("deltas"),
         py::arg("windows"), py::arg("std_dev"), py::arg("min_conf"));
+  m.def("deconv2x2", &deconv2x2,
+        "2x2 stride-2 transposed conv + bias (+ReLU), MFMA (CDNA4)");
   m.def("conv3x3_wrw", &conv3x3_wrw,
         "3x3 conv s1p1 weight gradient (CDNA4 MFMA, split-K atomics)",
         py::arg("x"), py::arg("dy"));

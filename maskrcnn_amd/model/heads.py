@@ -12,6 +12,7 @@ deconv 2x2/2 -> 1x1 conv num_classes + sigmoid -> [B, R, C, 28, 28].
 
 from __future__ import annotations
 
+import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
@@ -152,6 +153,15 @@ class FPNMaskHead(nn.Module):
         B, N, _ = rois.shape
         x = self.roi_align(rois, feature_maps)           # [B*N,256,14,14]
         x = self._conv_stack(x)
-        x = F.relu(self.mrcnn_mask_deconv(x))
+        if (x.is_cuda and x.dtype == torch.bfloat16 and
+                x.is_contiguous(memory_format=torch.channels_last)):
+            # One MFMA GEMM + scatter (csrc/hip/deconv2x2.hip, N6);
+            # backward via aten. Fuses the ReLU.
+            from ..ops.deconv import deconv2x2_bias_relu
+            x = deconv2x2_bias_relu(x, self.mrcnn_mask_deconv.weight,
+                                    self.mrcnn_mask_deconv.bias,
+                                    relu=True)
+        else:
+            x = F.relu(self.mrcnn_mask_deconv(x))
         x = self.mrcnn_mask(x)                           # [B*N, C, 28, 28]
         return x.reshape(B, N, *x.shape[1:])
