@@ -104,6 +104,195 @@ class AddGaussianNoise(Transform):
         return image, masks
 
 
+class Sometimes(Transform):
+    """Apply `child` with probability p (imgaug.Sometimes combinator —
+    part of the reference's MASK_AUGMENTERS whitelist,
+    mrcnn/model.py:~1260)."""
+
+    def __init__(self, p, child):
+        self.p = p
+        self.child = child
+
+    def __call__(self, image, masks, rng=None):
+        rng = rng or np.random
+        if rng.rand() < self.p:
+            return self.child(image, masks, rng)
+        return image, masks
+
+
+class OneOf(Transform):
+    """Apply exactly one randomly chosen child (imgaug.OneOf)."""
+
+    def __init__(self, children):
+        self.children = list(children)
+
+    def __call__(self, image, masks, rng=None):
+        rng = rng or np.random
+        t = self.children[int(rng.randint(0, len(self.children)))]
+        return t(image, masks, rng)
+
+
+class SomeOf(Transform):
+    """Apply n randomly chosen children, in order (imgaug.SomeOf)."""
+
+    def __init__(self, n, children):
+        self.n = n
+        self.children = list(children)
+
+    def __call__(self, image, masks, rng=None):
+        rng = rng or np.random
+        k = min(self.n, len(self.children))
+        idx = sorted(rng.choice(len(self.children), size=k,
+                                replace=False).tolist())
+        for i in idx:
+            image, masks = self.children[i](image, masks, rng)
+        return image, masks
+
+
+class Affine(Transform):
+    """Rotation / scale / translation / shear — the reference's main
+    geometric augmenter (imgaug.Affine is on the mask-safe whitelist).
+    Image is interpolated bilinearly, masks nearest — both through one
+    shared inverse-affine map so they stay registered.
+
+    rotate: degrees range (lo, hi); scale: factor range; translate_frac:
+    fraction-of-dim range; shear: degrees range.
+    """
+
+    def __init__(self, rotate=(-0.0, 0.0), scale=(1.0, 1.0),
+                 translate_frac=(0.0, 0.0), shear=(0.0, 0.0), p=1.0):
+        self.rotate = rotate
+        self.scale = scale
+        self.translate_frac = translate_frac
+        self.shear = shear
+        self.p = p
+
+    def __call__(self, image, masks, rng=None):
+        rng = rng or np.random
+        if rng.rand() >= self.p:
+            return image, masks
+        from scipy import ndimage
+        H, W = image.shape[:2]
+        ang = np.deg2rad(rng.uniform(*self.rotate))
+        sc = rng.uniform(*self.scale)
+        sh = np.deg2rad(rng.uniform(*self.shear))
+        ty = rng.uniform(*self.translate_frac) * H
+        tx = rng.uniform(*self.translate_frac) * W
+        # Forward map: rotate+shear+scale about the center, then shift.
+        c, s = np.cos(ang), np.sin(ang)
+        A = np.array([[c, -np.sin(ang - sh)],
+                      [s, np.cos(ang - sh)]]) * sc
+        center = np.array([(H - 1) / 2.0, (W - 1) / 2.0])
+        # ndimage.affine_transform applies the INVERSE map to outputs.
+        Ainv = np.linalg.inv(A)
+        offset = center - Ainv @ (center + np.array([ty, tx]))
+        img = np.stack([
+            ndimage.affine_transform(image[..., ch].astype(np.float32),
+                                     Ainv, offset=offset, order=1,
+                                     mode="constant", cval=0.0)
+            for ch in range(image.shape[2])], axis=-1)
+        image = np.clip(img, 0, 255).astype(image.dtype)
+        if masks.size:
+            m = np.stack([
+                ndimage.affine_transform(
+                    masks[..., i].astype(np.float32), Ainv,
+                    offset=offset, order=0, mode="constant", cval=0.0)
+                for i in range(masks.shape[2])], axis=-1)
+            masks = m > 0.5
+        return image, masks
+
+
+class GaussianBlur(Transform):
+    """Image-only blur, sigma drawn from [lo, hi] (imgaug.GaussianBlur)."""
+
+    def __init__(self, sigma=(0.0, 2.0), p=0.5):
+        self.sigma = sigma
+        self.p = p
+
+    def __call__(self, image, masks, rng=None):
+        rng = rng or np.random
+        if rng.rand() >= self.p:
+            return image, masks
+        from scipy import ndimage
+        sig = rng.uniform(*self.sigma)
+        if sig <= 0:
+            return image, masks
+        img = ndimage.gaussian_filter(
+            image.astype(np.float32), sigma=(sig, sig, 0))
+        return np.clip(img, 0, 255).astype(image.dtype), masks
+
+
+class ContrastNormalization(Transform):
+    """Scale contrast about the per-image mean (imgaug
+    ContrastNormalization): out = mean + f * (in - mean)."""
+
+    def __init__(self, rng_range=(0.75, 1.25), p=1.0):
+        self.lo, self.hi = rng_range
+        self.p = p
+
+    def __call__(self, image, masks, rng=None):
+        rng = rng or np.random
+        if rng.rand() >= self.p:
+            return image, masks
+        f = rng.uniform(self.lo, self.hi)
+        mean = image.astype(np.float32).mean()
+        img = mean + (image.astype(np.float32) - mean) * f
+        return np.clip(img, 0, 255).astype(image.dtype), masks
+
+
+class Add(Transform):
+    """Additive brightness shift drawn from [lo, hi] (imgaug.Add)."""
+
+    def __init__(self, rng_range=(-25, 25), p=1.0):
+        self.lo, self.hi = rng_range
+        self.p = p
+
+    def __call__(self, image, masks, rng=None):
+        rng = rng or np.random
+        if rng.rand() >= self.p:
+            return image, masks
+        v = rng.uniform(self.lo, self.hi)
+        img = image.astype(np.float32) + v
+        return np.clip(img, 0, 255).astype(image.dtype), masks
+
+
+class CropAndPad(Transform):
+    """Crop (negative) or pad (positive) each side by a fraction drawn
+    from `percent`, then resize back to the original shape
+    (imgaug.CropAndPad, mask-safe whitelist member)."""
+
+    def __init__(self, percent=(-0.1, 0.1), p=1.0):
+        self.percent = percent
+        self.p = p
+
+    def __call__(self, image, masks, rng=None):
+        rng = rng or np.random
+        if rng.rand() >= self.p:
+            return image, masks
+        from ..utils.image import resize
+        H, W = image.shape[:2]
+        f = rng.uniform(*self.percent)
+        dy, dx = int(round(H * f)), int(round(W * f))
+        if dy == 0 and dx == 0:
+            return image, masks
+        if f < 0:  # crop |dy| px off each side
+            y0, x0 = -dy, -dx
+            ic = image[y0:H - y0, x0:W - x0]
+            mc = masks[y0:H - y0, x0:W - x0]
+        else:      # pad dy px on each side
+            ic = np.zeros((H + 2 * dy, W + 2 * dx) + image.shape[2:],
+                          image.dtype)
+            ic[dy:dy + H, dx:dx + W] = image
+            mc = np.zeros((H + 2 * dy, W + 2 * dx) + masks.shape[2:],
+                          masks.dtype)
+            mc[dy:dy + H, dx:dx + W] = masks
+        image = resize(ic, (H, W)).astype(image.dtype)
+        mr = resize(mc.astype(np.float32), (H, W), order=0)
+        if mr.ndim == 2:
+            mr = mr[:, :, None]
+        return image, mr > 0.5
+
+
 class Crop(Transform):
     """Random crop of `frac` of each side, resized back — geometric, so
     masks follow."""

@@ -244,3 +244,66 @@ def test_loader_augmentation_varies_across_epochs(tmp_path):
     a = tds[0]["image"].numpy()
     tds.set_epoch(3)
     assert np.array_equal(tds[0]["image"].numpy(), a)
+
+
+def test_transforms_mask_registration():
+    """Geometric transforms keep image and mask registered: a bright
+    square under its own mask stays under it after Affine/CropAndPad
+    (reference MASK_AUGMENTERS whitelist semantics, model.py:~1260)."""
+    from maskrcnn_amd.data import transforms as T
+    rng = np.random.RandomState(3)
+    H = W = 96
+    image = np.zeros((H, W, 3), np.uint8)
+    masks = np.zeros((H, W, 1), bool)
+    image[30:60, 40:70] = 200
+    masks[30:60, 40:70, 0] = True
+    aug = T.Sequential([
+        T.Affine(rotate=(-25, 25), scale=(0.8, 1.2),
+                 translate_frac=(-0.1, 0.1), shear=(-8, 8)),
+        T.CropAndPad(percent=(-0.1, 0.1)),
+    ])
+    for _ in range(5):
+        img2, m2 = aug(image, masks, rng)
+        assert img2.shape == image.shape and m2.shape == masks.shape
+        if m2[..., 0].sum() < 50:
+            continue  # transform pushed the square mostly out of frame
+        # Bright pixels inside the transformed mask; dark outside.
+        inside = img2[m2[..., 0]].mean()
+        outside = img2[~m2[..., 0]].mean()
+        assert inside > outside + 50, (inside, outside)
+
+
+def test_transforms_combinators_and_photometric():
+    from maskrcnn_amd.data import transforms as T
+    rng = np.random.RandomState(4)
+    image = (np.random.RandomState(0).rand(64, 64, 3) * 255) \
+        .astype(np.uint8)
+    masks = np.zeros((64, 64, 2), bool)
+    masks[10:30, 10:30, 0] = True
+    masks[40:60, 40:60, 1] = True
+    aug = T.Sequential([
+        T.Sometimes(0.5, T.GaussianBlur(sigma=(0.5, 1.5), p=1.0)),
+        T.OneOf([T.Multiply((0.7, 1.3)), T.Add((-30, 30)),
+                 T.ContrastNormalization((0.6, 1.4))]),
+        T.SomeOf(2, [T.Fliplr(1.0), T.Flipud(1.0), T.Rotate90(1.0)]),
+    ])
+    for _ in range(8):
+        img2, m2 = aug(image, masks, rng)
+        assert img2.shape == image.shape
+        assert m2.shape == masks.shape
+        assert img2.dtype == np.uint8 and m2.dtype == bool
+        # Photometric ops must not change mask pixel counts; geometric
+        # flips/rot90 preserve them exactly.
+        assert m2.sum() == masks.sum()
+
+
+def test_affine_identity_is_noop():
+    from maskrcnn_amd.data import transforms as T
+    rng = np.random.RandomState(5)
+    image = (np.random.RandomState(1).rand(48, 48, 3) * 255) \
+        .astype(np.uint8)
+    masks = np.zeros((48, 48, 1), bool)
+    masks[8:20, 8:20, 0] = True
+    img2, m2 = T.Affine()(image, masks, rng)
+    np.testing.assert_array_equal(m2, masks)
+    assert np.abs(img2.astype(int) - image.astype(int)).max() <= 1
