@@ -197,6 +197,12 @@ def main():
     ap.add_argument("--torch-profile", metavar="DIR", default=None,
                     help="capture a kineto (torch.profiler) chrome trace "
                          "of 3 steady steps into DIR")
+    ap.add_argument("--data", default="synthetic",
+                    choices=["synthetic", "loader"],
+                    help="synthetic = one cached on-device batch "
+                         "(BASELINE contract: pure step throughput); "
+                         "loader = real DataLoader pipeline with "
+                         "per-step HtoD inside the timed region")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -246,7 +252,40 @@ def main():
 
     use_bf16 = (config.COMPUTE_DTYPE == "bf16" and use_cuda)
 
+    loader_iter = None
+    if args.data == "loader":
+        # End-to-end variant: shapes drawn by worker processes, collated
+        # batches HtoD-copied inside the timed step.
+        from maskrcnn_amd.data.loader import make_data_loader
+        from maskrcnn_amd.data.synthetic import ShapesDataset
+        ds = ShapesDataset()
+        ds.load_shapes(64, args.image_size, args.image_size,
+                       seed=7 + rank)
+        # Pad the class table to the benched NUM_CLASSES (COCO-shaped
+        # meta: active_class_ids length must equal config.NUM_CLASSES).
+        for ci in range(4, config.NUM_CLASSES):
+            ds.add_class("shapes", ci, f"pad{ci}")
+        ds.prepare()
+        loader = make_data_loader(ds, config,
+                                  batch_size=config.BATCH_SIZE,
+                                  num_workers=2, seed=rank)
+
+        def _cycle():
+            while True:
+                for b in loader:
+                    yield b
+        loader_iter = _cycle()
+
     def one_step():
+        nonlocal batch, meta
+        if loader_iter is not None:
+            host = next(loader_iter)
+            batch = {k: v.to(device, non_blocking=True)
+                     for k, v in host.items()}
+            if config.CHANNELS_LAST and use_cuda:
+                batch["image"] = batch["image"].to(
+                    memory_format=torch.channels_last)
+            meta = parse_image_meta_t(batch["image_meta"])
         with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
                             enabled=use_bf16):
             outputs = model(batch["image"], batch["image_meta"], anchors,
@@ -335,7 +374,8 @@ def main():
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": args.dtype if use_cuda else "fp32",
-            "data": "synthetic",
+            "data": ("synthetic" if args.data == "synthetic"
+                     else "synthetic-loader-e2e"),
             "config": {
                 "model": f"{config.BACKBONE}-FPN maskrcnn",
                 "global_batch": global_batch,
