@@ -37,14 +37,14 @@ class FPN(nn.Module):
         self.use_fused = use_fused
 
     def forward(self, c2, c3, c4, c5):
-        # Fused path in eval/serving (measured 1.4-1.9x, profiles/
-        # r01_conv_ab.md). In training the end-to-end delta was within
-        # noise vs MIOpen but trended slower on one box — kept on the
-        # known-good path until the round-2 A/B (same backward either
-        # way). MASKRCNN_TRAIN_FUSED_FPN=1 enables it for that A/B.
+        # Fused path (measured 1.4-1.9x per level, profiles/
+        # r01_conv_ab.md). Default ON in training too since the round-2
+        # block2 A/B: 45.84 vs 46.86 ms/step baseline (same backward
+        # either way). MASKRCNN_TRAIN_FUSED_FPN=0 opts out.
         import os
         if self.use_fused and (not self.training or
-                               os.environ.get("MASKRCNN_TRAIN_FUSED_FPN")):
+                               os.environ.get("MASKRCNN_TRAIN_FUSED_FPN",
+                                              "1") != "0"):
             p5 = fpn_fuse(c5, self.fpn_c5p5.weight, self.fpn_c5p5.bias)
             p4 = fpn_fuse(c4, self.fpn_c4p4.weight, self.fpn_c4p4.bias, p5)
             p3 = fpn_fuse(c3, self.fpn_c3p3.weight, self.fpn_c3p3.bias, p4)

@@ -87,7 +87,7 @@ def test_vectorized_ratio_matches_loop_semantics(shapes_config,
     torch.manual_seed(0)
     vec = DetectionTargetLayer(shapes_config)
     _, cls_v, _, _ = vec(proposals, ids, boxes, masks)
-    monkeypatch.delenv("MASKRCNN_VECTORIZED_TARGETS")
+    monkeypatch.setenv("MASKRCNN_VECTORIZED_TARGETS", "0")
     torch.manual_seed(0)
     loop = DetectionTargetLayer(shapes_config)
     assert not loop.vectorized
