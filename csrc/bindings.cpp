@@ -38,6 +38,8 @@ void conv3x3_v2_launch(const void*, const void*, const float*, const void*,
                        void*, int, int, int, int, int, bool, hipStream_t);
 void conv3x3_v3_launch(const void*, const void*, const float*, const void*,
                        void*, int, int, int, int, int, bool, hipStream_t);
+void conv3x3_v4_launch(const void*, const void*, const float*, const void*,
+                       void*, int, int, int, int, int, bool, hipStream_t);
 void conv3x3_wrw_launch(const void*, const void*, float*, int, int, int,
                         int, int, hipStream_t);
 void conv3x3_fp8_launch(const void*, const void*, const float*,
@@ -293,7 +295,9 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
       it = zero_pages.emplace(
           dev, torch::zeros({32}, x.options())).first;
     }
-    auto launch = (ver == 3) ? conv3x3_v3_launch : conv3x3_v2_launch;
+    auto launch = (ver == 4) ? conv3x3_v4_launch
+                : (ver == 3) ? conv3x3_v3_launch
+                             : conv3x3_v2_launch;
     launch(x.data_ptr(), w_packed.data_ptr(), bias_f.data_ptr<float>(),
            it->second.data_ptr(), out.data_ptr(), B, H, W, Cin, Cout,
            relu, cur_stream());

@@ -85,7 +85,12 @@ def test_gpu_kernel_matches_reference(cin, cout, h, w):
 
 
 @pytest.mark.gpu
-@pytest.mark.parametrize("version", [2, 3])
+@pytest.mark.parametrize("version", [
+    2, 3,
+    pytest.param(4, marks=pytest.mark.xfail(
+        strict=False, reason="v4 (256^2-tile) pending first GPU "
+        "validation")),
+])
 @pytest.mark.parametrize("cin,cout,h,w,batch", [
     (256, 256, 64, 64, 2),    # exact-tile case
     (256, 256, 50, 30, 3),    # ragged M
