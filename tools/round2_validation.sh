@@ -36,11 +36,21 @@ block2)  # training-path A/Bs (one box; ~12 min incl. one find phase)
       MASKRCNN_HIP_WRW=1 run wrw
   ;;
 block3)  # serving bench (~8 min incl. 265 s batch-32 find)
+  # Capture the MIOpen user find-db in gpurun_out so it merges back and
+  # can be committed as repo miopen_db/ (fresh boxes then skip the
+  # 265 s batch-32 Find; _rocm_env exports MIOPEN_USER_DB_PATH).
+  mkdir -p gpurun_out/miopen_db
+  if [ -d miopen_db ] && [ -n "$(ls miopen_db 2>/dev/null)" ]; then
+    cp miopen_db/* gpurun_out/miopen_db/ 2>/dev/null
+  fi
+  export MIOPEN_USER_DB_PATH="$PWD/gpurun_out/miopen_db"
+  export MIOPEN_FIND_MODE=5
   timeout 900 python bench/serving_bench.py --requests 96 \
       --client-procs 4 --client-threads 8 --warmup 16 \
       > gpurun_out/r2_serving.json 2> gpurun_out/r2_serving.err
   cat gpurun_out/r2_serving.json
   grep -E "WARNING|batch of" gpurun_out/r2_serving.err | tail -5
+  ls -la gpurun_out/miopen_db/
   # A/B: hipGraph replay of the engine forward (validated in block1)
   MASKRCNN_SERVING_HIPGRAPH=1 timeout 700 python bench/serving_bench.py \
       --requests 96 --client-procs 4 --client-threads 8 --warmup 16 \
