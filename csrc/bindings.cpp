@@ -280,11 +280,13 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
       x.options().memory_format(at::MemoryFormat::ChannelsLast));
   const long M = (long)B * H * W;
   int ver = (int)version;
-  if (ver == 0)  // auto: the 128x128 glds kernels need enough M x N tiles.
-    // v3 (3-buffer counted-vmcnt + XCD swizzle) beats v2 at every
-    // large-M shape in the round-2 block1 A/B (gpurun_out/r2_conv_ab):
-    // 641 vs 603 TF @(2,256,256^2,256), 635 vs 589 @(400,256,14^2,256).
-    ver = (M >= 32768 && Cout >= 128) ? 3 : 1;
+  if (ver == 0)  // auto (r2_conv_ab2 A/B):
+    // v4 (256² tile, 8 waves) wins at very large M where its 1-WG/CU
+    // grid still fills the chip: 701 TF vs v3's 650 @(2,256,256²,256).
+    // v3 wins mid-M (602 vs 434 TF @(400,256,14²,256) — v4's 256-row
+    // tiles leave the grid at ~1 WG/CU there). v1 below 32768.
+    ver = (M >= 131072 && Cout >= 128) ? 4
+        : (M >= 32768 && Cout >= 128) ? 3 : 1;
   if (ver >= 2) {
     // Persistent 64-B zero page per device (glds cannot predicate
     // out-of-range lanes; they DMA from here instead).

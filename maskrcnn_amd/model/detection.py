@@ -18,7 +18,7 @@ import torch
 import torch.nn as nn
 
 from ..ops import batched_nms, batched_nms_padded
-from ..utils.boxes import apply_box_deltas_t, norm_boxes_t
+from ..utils.boxes import apply_box_deltas_t
 from ..utils.meta import parse_image_meta_t
 
 
@@ -29,6 +29,10 @@ class DetectionLayer(nn.Module):
         self.register_buffer(
             "bbox_std_dev",
             torch.tensor(config.BBOX_STD_DEV, dtype=torch.float32),
+            persistent=False)
+        self.register_buffer(
+            "_norm_shift",
+            torch.tensor([0., 0., 1., 1.], dtype=torch.float32),
             persistent=False)
 
     @torch.no_grad()
@@ -48,11 +52,13 @@ class DetectionLayer(nn.Module):
         B, N, C = probs.shape
         device = rois.device
         meta = parse_image_meta_t(image_meta)
-        image_shape = meta["image_shape"][0]
         # Windows in normalized coords (reference :~700), per image.
-        windows = norm_boxes_t(meta["window"].float(),
-                               (float(image_shape[0]),
-                                float(image_shape[1])))  # [B, 4]
+        # All-tensor math: float(shape_tensor[i]) is a host sync, and
+        # torch.tensor(...)/list-index gathers are H2D copies — any of
+        # them aborts hipGraph stream capture of the serving forward.
+        hw = meta["image_shape"][:, :2].float()            # [B, 2]
+        scale = hw.repeat(1, 2) - 1.0                      # [B,4] h-1,w-1
+        windows = (meta["window"].float() - self._norm_shift) / scale
 
         if _use_syncfree(rois):
             # Fused HIP prep (one detection_prep launch replaces the

@@ -103,7 +103,7 @@ class FPNMaskHead(nn.Module):
         from ..ops.conv_policy import use_hip_conv3x3
         if not (self._fused and x.is_cuda and not self.training and
                 use_hip_conv3x3(x.shape[0], x.shape[1], x.shape[2],
-                                x.shape[3])):
+                                x.shape[3], 256)):
             return self.convs(x)
         from ..ops.conv3x3 import conv3x3_bn_relu
         for i in range(4):
@@ -153,8 +153,10 @@ class FPNMaskHead(nn.Module):
         B, N, _ = rois.shape
         x = self.roi_align(rois, feature_maps)           # [B*N,256,14,14]
         x = self._conv_stack(x)
+        import os
         if (x.is_cuda and x.dtype == torch.bfloat16 and
-                x.is_contiguous(memory_format=torch.channels_last)):
+                x.is_contiguous(memory_format=torch.channels_last) and
+                os.environ.get("MASKRCNN_HIP_DECONV", "1") != "0"):
             # One MFMA GEMM + scatter (csrc/hip/deconv2x2.hip, N6);
             # backward via aten. Fuses the ReLU.
             from ..ops.deconv import deconv2x2_bias_relu

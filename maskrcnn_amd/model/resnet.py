@@ -86,7 +86,7 @@ class Bottleneck(nn.Module):
                 (not self.training or
                  os.environ.get("MASKRCNN_TRAIN_HIP_CONV")) and \
                 use_hip_conv3x3(out.shape[0], out.shape[1], out.shape[2],
-                                out.shape[3]):
+                                out.shape[3], self.conv2.out_channels):
             from ..ops.conv3x3 import conv3x3_bn_relu
             bn = self.bn2
             out = conv3x3_bn_relu(out, self.conv2.weight, self.conv2.bias,

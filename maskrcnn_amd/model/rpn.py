@@ -34,7 +34,7 @@ class RPN(nn.Module):
         conv = self.rpn_conv_shared
         if (x.is_cuda and not self.training and conv.stride[0] == 1 and
                 use_hip_conv3x3(x.shape[0], x.shape[1], x.shape[2],
-                                x.shape[3])):
+                                x.shape[3], conv.out_channels)):
             from ..ops.conv3x3 import conv3x3_bias_relu
             shared = conv3x3_bias_relu(x, conv.weight, conv.bias,
                                        relu=True)

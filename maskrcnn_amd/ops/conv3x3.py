@@ -114,7 +114,8 @@ class _Conv3x3Fn(torch.autograd.Function):
         if use_hip_dgrad:
             from .conv_policy import use_hip_conv3x3
             use_hip_dgrad = use_hip_conv3x3(
-                go_cl.size(0), w.size(0), go_cl.size(2), go_cl.size(3))
+                go_cl.size(0), w.size(0), go_cl.size(2), go_cl.size(3),
+                w.size(1))
         # Weight gradient: HIP split-K MFMA kernel (csrc/hip/
         # conv3x3_wrw.hip — wrw was 47% of the original MIOpen step
         # profile) when opted in; aten/MIOpen otherwise. Off by default
