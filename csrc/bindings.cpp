@@ -51,6 +51,9 @@ void detection_prep_launch(const void*, const void*, const void*,
                            hipStream_t);
 void deconv2x2_launch(const void*, const void*, const float*, const void*,
                       void*, int, int, int, int, int, bool, hipStream_t);
+void conv3x3_mx_launch(const void*, const void*, const void*, const void*,
+                       const float*, const void*, void*, int, int, int,
+                       int, int, bool, hipStream_t);
 }
 
 namespace {
@@ -344,6 +347,47 @@ torch::Tensor deconv2x2(torch::Tensor x, torch::Tensor w_packed,
   return out;
 }
 
+torch::Tensor conv3x3_mx(torch::Tensor xq, torch::Tensor xs,
+                         torch::Tensor wq, torch::Tensor ws,
+                         torch::Tensor bias, bool relu) {
+  // MX block-scaled e4m3 conv (csrc/hip/conv3x3_mx.hip).
+  // xq: [B,H,W,Cin] uint8 contiguous (NHWC bytes);
+  // xs: [B,H,W,Cin/32] uint8 e8m0; wq: [9,Cout,Cin] uint8 tap-major;
+  // ws: [9,Cout,Cin/32] uint8. Returns [B,Cout,H,W] bf16 channels_last.
+  TORCH_CHECK(xq.is_cuda() && xq.dim() == 4 && xq.is_contiguous(),
+              "xq must be contiguous [B,H,W,Cin] u8 cuda");
+  TORCH_CHECK(xq.scalar_type() == torch::kByte &&
+              xs.scalar_type() == torch::kByte &&
+              wq.scalar_type() == torch::kByte &&
+              ws.scalar_type() == torch::kByte, "u8 tensors required");
+  TORCH_CHECK(xs.is_contiguous() && wq.is_contiguous() &&
+              ws.is_contiguous(), "contiguous required");
+  const int B = xq.size(0), H = xq.size(1), W = xq.size(2),
+            Cin = xq.size(3);
+  TORCH_CHECK(Cin % 128 == 0, "Cin must be a multiple of 128 (BK)");
+  TORCH_CHECK(wq.dim() == 3 && wq.size(0) == 9 && wq.size(2) == Cin,
+              "wq must be [9, Cout, Cin]");
+  const int Cout = wq.size(1);
+  TORCH_CHECK(xs.size(3) == Cin / 32 && ws.size(2) == Cin / 32,
+              "scale tensors must have Cin/32 blocks");
+  auto bias_f = bias.to(torch::kFloat).contiguous();
+  auto out = torch::empty(
+      {B, Cout, H, W},
+      xq.options().dtype(torch::kBFloat16)
+          .memory_format(at::MemoryFormat::ChannelsLast));
+  static std::unordered_map<int, torch::Tensor> zero_pages;
+  const int dev = xq.get_device();
+  auto it = zero_pages.find(dev);
+  if (it == zero_pages.end())
+    it = zero_pages.emplace(
+        dev, torch::zeros({64}, xq.options())).first;
+  conv3x3_mx_launch(xq.data_ptr(), xs.data_ptr(), wq.data_ptr(),
+                    ws.data_ptr(), bias_f.data_ptr<float>(),
+                    it->second.data_ptr(), out.data_ptr(), B, H, W, Cin,
+                    Cout, relu, cur_stream());
+  return out;
+}
+
 torch::Tensor conv3x3_fp8(torch::Tensor x, torch::Tensor w_packed,
                           torch::Tensor bias, torch::Tensor w_scale,
                           torch::Tensor a_scale, bool relu) {
@@ -474,6 +518,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("windows"), py::arg("std_dev"), py::arg("min_conf"));
   m.def("deconv2x2", &deconv2x2,
         "2x2 stride-2 transposed conv + bias (+ReLU), MFMA (CDNA4)");
+  m.def("conv3x3_mx", &conv3x3_mx,
+        "MX block-scaled e4m3 3x3 conv (mfma_scale 16x16x128, gfx950)");
   m.def("conv3x3_wrw", &conv3x3_wrw,
         "3x3 conv s1p1 weight gradient (CDNA4 MFMA, split-K atomics)",
         py::arg("x"), py::arg("dy"));

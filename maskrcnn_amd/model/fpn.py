@@ -14,6 +14,7 @@ smoothing convs stay on MIOpen. Numerics are identical either way
 
 from __future__ import annotations
 
+import torch
 import torch.nn as nn
 import torch.nn.functional as F
 

@@ -60,6 +60,40 @@ def conv3x3_fp8(x, w_packed_u8, w_scale, bias, relu=True):
     return ext.conv3x3_fp8(xq, w_packed_u8, zb, w_scale, a_scale, relu)
 
 
+def pack_weight_mx(w: torch.Tensor):
+    """[Cout, Cin, 3, 3] -> ([9, Cout, Cin] u8 e4m3, [9, Cout, Cin/32]
+    u8 e8m0) tap-major for the MX block-scaled kernel: one power-of-two
+    scale per (tap, out-channel, 32-in-channel block)."""
+    from ..utils.fp8 import mx_quantize
+    Cout, Cin, _, _ = w.shape
+    wt = w.float().permute(2, 3, 0, 1).reshape(9, Cout, Cin)
+    q, exp = mx_quantize(wt, dim=-1)
+    return (q.view(torch.uint8).contiguous(), exp.contiguous())
+
+
+def quantize_activations_mx(x: torch.Tensor):
+    """Per-(pixel, 32-channel-block) MX quantization of a channels_last
+    activation. Returns (xq [B,H,W,C] u8 contiguous, xs [B,H,W,C/32]
+    u8 e8m0). All on GPU, no host sync."""
+    from ..utils.fp8 import mx_quantize
+    xn = x.permute(0, 2, 3, 1).float()        # NHWC values
+    q, exp = mx_quantize(xn, dim=-1)
+    return q.view(torch.uint8).contiguous(), exp.contiguous()
+
+
+def conv3x3_mx(x, wq, ws, bias, relu=True):
+    """MX block-scaled e4m3 3x3 conv at the 2x fp8 MFMA rate
+    (mfma_scale_f32_16x16x128_f8f6f4). Inference-only. x bf16/f32
+    channels_last [B,C,H,W]; (wq, ws) from pack_weight_mx. Returns
+    bf16 channels_last [B,Cout,H,W]."""
+    from ._ext import require_ext
+    ext = require_ext()
+    xq, xs = quantize_activations_mx(x)
+    zb = bias.float() if bias is not None else \
+        torch.zeros(wq.shape[1], device=x.device)
+    return ext.conv3x3_mx(xq, xs, wq, ws, zb, relu)
+
+
 def rotate_for_dgrad(w: torch.Tensor) -> torch.Tensor:
     """Weight transform making bwd-data a FORWARD conv:
     dgrad(x) = conv3x3(grad_out, w') with
