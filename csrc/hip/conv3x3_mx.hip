@@ -110,10 +110,13 @@ void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
           ? xq + ((apx_b[i] * H + sy) * (long)W + sx) * Cin + k0 +
                 d_chunk * 16
           : zeros;
+      // LDS dest must be WAVE-UNIFORM: the DMA writes lane-linear
+      // (dest + lane*16), which lands lane l at row +(l>>3),
+      // chunk l&7 — exactly the per-lane source coordinates above.
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)asrc,
           (__attribute__((address_space(3))) unsigned int*)
-              &smem[a_off(buf) + row * ROW_B + d_chunk * 16],
+              &smem[a_off(buf) + (wave * 32 + i * 8) * ROW_B],
           16, 0, 0);
       const int n = n0 + row;
       const unsigned char* bsrc = (n < Cout)
@@ -122,7 +125,7 @@ void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)bsrc,
           (__attribute__((address_space(3))) unsigned int*)
-              &smem[b_off(buf) + row * ROW_B + d_chunk * 16],
+              &smem[b_off(buf) + (wave * 32 + i * 8) * ROW_B],
           16, 0, 0);
     }
     if (wave == 0) {
@@ -138,7 +141,7 @@ void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) unsigned int*)src,
             (__attribute__((address_space(3))) unsigned int*)
-                &smem[as_off(buf) + row * 4],
+                &smem[as_off(buf) + i * 64 * 4],
             4, 0, 0);
       }
     } else if (wave == 1) {
@@ -152,7 +155,7 @@ void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) unsigned int*)src,
             (__attribute__((address_space(3))) unsigned int*)
-                &smem[bs_off(buf) + row * 4],
+                &smem[bs_off(buf) + i * 64 * 4],
             4, 0, 0);
       }
     }

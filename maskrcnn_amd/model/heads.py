@@ -156,9 +156,15 @@ class FPNMaskHead(nn.Module):
         import os
         if (x.is_cuda and x.dtype == torch.bfloat16 and
                 x.is_contiguous(memory_format=torch.channels_last) and
+                (not self.training or
+                 os.environ.get("MASKRCNN_HIP_DECONV") == "1") and
                 os.environ.get("MASKRCNN_HIP_DECONV", "1") != "0"):
             # One MFMA GEMM + scatter (csrc/hip/deconv2x2.hip, N6);
-            # backward via aten. Fuses the ReLU.
+            # fuses the ReLU. Eval/serving only by default: the r5
+            # bench A/B measured the per-step weight re-pack + aten
+            # backward mix costing ~0.7 ms/step in training (46.25 vs
+            # 45.52); MASKRCNN_HIP_DECONV=1 forces it on in training,
+            # =0 forces MIOpen everywhere.
             from ..ops.deconv import deconv2x2_bias_relu
             x = deconv2x2_bias_relu(x, self.mrcnn_mask_deconv.weight,
                                     self.mrcnn_mask_deconv.bias,

@@ -8,13 +8,17 @@ import torch
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False,
-                   reason="hipGraph serving path pending first GPU "
-                          "validation (round-2 block)")
 def test_engine_hipgraph_matches_eager(shapes_config, monkeypatch):
     """hipGraph replay of the serving forward must equal eager output
-    for the same input (engine._forward capture/replay path)."""
+    for the same input (engine._forward capture/replay path).
+
+    The eager engine is pinned to the SAME sync-free roi/detection code
+    the capture uses (capture auto-selects it), so replay must be
+    BITWISE equal to eager — any difference is a stale-buffer or race
+    bug, not tolerance noise."""
     from maskrcnn_amd.serving.engine import InferenceEngine
+    monkeypatch.setenv("MASKRCNN_SYNCFREE_ROI", "1")
+    monkeypatch.setenv("MASKRCNN_SYNCFREE_DETECTION", "1")
     torch.manual_seed(0)
     eager = InferenceEngine(shapes_config)
     monkeypatch.setenv("MASKRCNN_SERVING_HIPGRAPH", "1")
@@ -31,6 +35,31 @@ def test_engine_hipgraph_matches_eager(shapes_config, monkeypatch):
         o_e = eager.predict(images, metas)
         o_g = graphed.predict(images, metas)
         for k in o_e:
-            np.testing.assert_allclose(o_g[k], o_e[k], atol=1e-3,
-                                       rtol=1e-3)
+            np.testing.assert_array_equal(o_g[k], o_e[k], err_msg=k)
     assert len(graphed._graphs) == 1
+
+
+@pytest.mark.gpu
+def test_engine_hipgraph_close_to_gather_path(shapes_config,
+                                              monkeypatch):
+    """Replay vs the default eager gather path: same detections within
+    bf16 tolerance (the sync-free mask-combine ROI pooling rounds
+    differently from the per-level gather)."""
+    from maskrcnn_amd.serving.engine import InferenceEngine
+    torch.manual_seed(0)
+    eager = InferenceEngine(shapes_config)
+    monkeypatch.setenv("MASKRCNN_SERVING_HIPGRAPH", "1")
+    graphed = InferenceEngine(shapes_config)
+    graphed.model.load_state_dict(eager.model.state_dict())
+    H = shapes_config.IMAGE_SHAPE[0]
+    metas = np.zeros((1, shapes_config.IMAGE_META_SIZE), np.float32)
+    metas[:, 4:7] = (H, H, 3)
+    metas[:, 7:11] = (0, 0, H, H)
+    images = np.random.rand(1, H, H, 3).astype(np.float32) * 50
+    o_e = eager.predict(images, metas)
+    o_g = graphed.predict(images, metas)
+    # Box geometry of the surviving detections agrees closely; scores
+    # can reorder at bf16 ties, so compare the sorted box coordinates.
+    det_e = np.sort(o_e["mrcnn_detection"][0, :, :4], axis=0)
+    det_g = np.sort(o_g["mrcnn_detection"][0, :, :4], axis=0)
+    np.testing.assert_allclose(det_g, det_e, atol=5e-3)
