@@ -81,6 +81,18 @@ block5)  # PMC counters for the hand kernels (own run, never with traces)
   ls gpurun_out/r2_pmc
   ;;
 esac
+if [ "${1:-}" = "block7" ]; then
+  # Final default-flag sweep at 30 steps (block2's 15-step singles were
+  # within ±0.5 ms of each other; this pins the shipping combination).
+  run7() { timeout 500 python bench.py --steps 30 --warmup 5 \
+           > "gpurun_out/r7_$1.json" 2>/dev/null; cat "gpurun_out/r7_$1.json"; }
+  export MIOPEN_FIND_MODE=5
+  echo "=== all-on (shipping defaults)"; run7 all
+  echo "=== vtgt only"; MASKRCNN_TRAIN_FUSED_FPN=0 MASKRCNN_TRAIN_CL_ROI=0 run7 vtgt
+  echo "=== vtgt+fpn"; MASKRCNN_TRAIN_CL_ROI=0 run7 vtgt_fpn
+  echo "=== vtgt+clroi"; MASKRCNN_TRAIN_FUSED_FPN=0 run7 vtgt_clroi
+fi
+
 # (appended) block6: fp8-weight large-image serving (BASELINE config #5,
 # 1333 -> padded 1344; DP=8 scale-out is the driver's 8-GPU tier — this
 # measures the per-GPU replica)
