@@ -40,6 +40,8 @@ void conv3x3_v3_launch(const void*, const void*, const float*, const void*,
                        void*, int, int, int, int, int, bool, hipStream_t);
 void conv3x3_v4_launch(const void*, const void*, const float*, const void*,
                        void*, int, int, int, int, int, bool, hipStream_t);
+void conv3x3_v5_launch(const void*, const void*, const float*, const void*,
+                       void*, int, int, int, int, int, bool, hipStream_t);
 void conv3x3_wrw_launch(const void*, const void*, float*, int, int, int,
                         int, int, hipStream_t);
 void conv3x3_fp8_launch(const void*, const void*, const float*,
@@ -54,6 +56,8 @@ void deconv2x2_launch(const void*, const void*, const float*, const void*,
 void conv3x3_mx_launch(const void*, const void*, const void*, const void*,
                        const float*, const void*, void*, int, int, int,
                        int, int, bool, hipStream_t);
+void mx_probe_launch(const void*, const void*, const void*, const void*,
+                     float*, hipStream_t);
 }
 
 namespace {
@@ -300,7 +304,8 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
       it = zero_pages.emplace(
           dev, torch::zeros({32}, x.options())).first;
     }
-    auto launch = (ver == 4) ? conv3x3_v4_launch
+    auto launch = (ver == 5) ? conv3x3_v5_launch
+                : (ver == 4) ? conv3x3_v4_launch
                 : (ver == 3) ? conv3x3_v3_launch
                              : conv3x3_v2_launch;
     launch(x.data_ptr(), w_packed.data_ptr(), bias_f.data_ptr<float>(),
@@ -345,6 +350,18 @@ torch::Tensor deconv2x2(torch::Tensor x, torch::Tensor w_packed,
                    out.data_ptr(), B, H, W, Cin, Cout, relu,
                    cur_stream());
   return out;
+}
+
+torch::Tensor mx_probe(torch::Tensor a, torch::Tensor b,
+                       torch::Tensor as, torch::Tensor bs) {
+  // a,b: [16,128] u8 e4m3; as,bs: [16,4] u8 e8m0. Returns [64,4] f32
+  // (each lane's accumulators). Layout probe — test-only.
+  TORCH_CHECK(a.is_cuda() && a.is_contiguous() && b.is_contiguous() &&
+              as.is_contiguous() && bs.is_contiguous(), "contig cuda");
+  auto c = torch::zeros({64, 4}, a.options().dtype(torch::kFloat));
+  mx_probe_launch(a.data_ptr(), b.data_ptr(), as.data_ptr(),
+                  bs.data_ptr(), c.data_ptr<float>(), cur_stream());
+  return c;
 }
 
 torch::Tensor conv3x3_mx(torch::Tensor xq, torch::Tensor xs,
@@ -520,6 +537,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "2x2 stride-2 transposed conv + bias (+ReLU), MFMA (CDNA4)");
   m.def("conv3x3_mx", &conv3x3_mx,
         "MX block-scaled e4m3 3x3 conv (mfma_scale 16x16x128, gfx950)");
+  m.def("mx_probe", &mx_probe,
+        "one-instruction mfma_scale layout probe (test-only)");
   m.def("conv3x3_wrw", &conv3x3_wrw,
         "3x3 conv s1p1 weight gradient (CDNA4 MFMA, split-K atomics)",
         py::arg("x"), py::arg("dy"));

@@ -38,6 +38,10 @@ def test_engine_bf16_gpu_deterministic(shapes_config):
     metas = np.zeros((1, shapes_config.IMAGE_META_SIZE), np.float32)
     metas[:, 4:7] = (H, H, 3)
     metas[:, 7:11] = (0, 0, H, H)
+    # Warm once: cudnn.benchmark may switch MIOpen algorithms during
+    # the FIRST call's Find, which legitimately changes low bits.
+    # Steady-state predicts must then be byte-identical.
+    eng.predict(images, metas)
     out1 = eng.predict(images, metas)
     out2 = eng.predict(images, metas)
     for k in out1:
