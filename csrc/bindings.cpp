@@ -56,8 +56,8 @@ void deconv2x2_launch(const void*, const void*, const float*, const void*,
 void conv3x3_mx_launch(const void*, const void*, const void*, const void*,
                        const float*, const void*, void*, int, int, int,
                        int, int, bool, hipStream_t);
-void mx_probe_launch(const void*, const void*, const void*, const void*,
-                     float*, hipStream_t);
+void mx_probe_launch(const void*, const void*, const unsigned int*,
+                     const unsigned int*, float*, hipStream_t);
 }
 
 namespace {
@@ -288,11 +288,11 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
   const long M = (long)B * H * W;
   int ver = (int)version;
   if (ver == 0)  // auto (r2_conv_ab2 A/B):
-    // v4 (256² tile, 8 waves) wins at very large M where its 1-WG/CU
-    // grid still fills the chip: 701 TF vs v3's 650 @(2,256,256²,256).
-    // v3 wins mid-M (602 vs 434 TF @(400,256,14²,256) — v4's 256-row
-    // tiles leave the grid at ~1 WG/CU there). v1 below 32768.
-    ver = (M >= 131072 && Cout >= 128) ? 4
+    // v5 (256² tile, 8 waves, st_16x32 swizzle) wins at very large M:
+    // 725 TF vs MIOpen's ~684 @(2,256,256²,256), 761 vs 740 @Cout=512
+    // (r7_conv_ab). v3 wins mid-M (621 vs 459 @(400,256,14²,256) —
+    // the 256-row tiles leave that grid at ~1 WG/CU). v1 below 32768.
+    ver = (M >= 131072 && Cout >= 128) ? 5
         : (M >= 32768 && Cout >= 128) ? 3 : 1;
   if (ver >= 2) {
     // Persistent 64-B zero page per device (glds cannot predicate
@@ -354,13 +354,19 @@ torch::Tensor deconv2x2(torch::Tensor x, torch::Tensor w_packed,
 
 torch::Tensor mx_probe(torch::Tensor a, torch::Tensor b,
                        torch::Tensor as, torch::Tensor bs) {
-  // a,b: [16,128] u8 e4m3; as,bs: [16,4] u8 e8m0. Returns [64,4] f32
-  // (each lane's accumulators). Layout probe — test-only.
+  // a,b: [16,128] u8 e4m3; as,bs: [64] int32 — the FULL per-lane scale
+  // registers. Returns [64,4] f32 (each lane's accumulators).
+  // Layout probe — test-only.
   TORCH_CHECK(a.is_cuda() && a.is_contiguous() && b.is_contiguous() &&
               as.is_contiguous() && bs.is_contiguous(), "contig cuda");
+  TORCH_CHECK(as.numel() == 64 && bs.numel() == 64 &&
+              as.scalar_type() == torch::kInt &&
+              bs.scalar_type() == torch::kInt, "scales: [64] int32");
   auto c = torch::zeros({64, 4}, a.options().dtype(torch::kFloat));
-  mx_probe_launch(a.data_ptr(), b.data_ptr(), as.data_ptr(),
-                  bs.data_ptr(), c.data_ptr<float>(), cur_stream());
+  mx_probe_launch(a.data_ptr(), b.data_ptr(),
+                  (const unsigned int*)as.data_ptr<int>(),
+                  (const unsigned int*)bs.data_ptr<int>(),
+                  c.data_ptr<float>(), cur_stream());
   return c;
 }
 

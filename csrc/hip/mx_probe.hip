@@ -14,11 +14,14 @@ typedef __attribute__((ext_vector_type(8))) int i32x8;
 
 namespace {
 
+// a_scales / b_scales are FULL per-lane i32s ([64] u32, all 4 bytes
+// under the lane's control) so the host can map empirically which
+// (source lane, byte) the hardware reads each block's e8m0 from.
 __global__ __launch_bounds__(64)
 void mx_probe_kernel(const unsigned char* __restrict__ a_bytes,  // [16][128]
                      const unsigned char* __restrict__ b_bytes,  // [16][128]
-                     const unsigned char* __restrict__ a_scales, // [16][4]
-                     const unsigned char* __restrict__ b_scales, // [16][4]
+                     const unsigned int* __restrict__ a_scales,  // [64]
+                     const unsigned int* __restrict__ b_scales,  // [64]
                      float* __restrict__ c_out) {                // [64][4]
   const int l = threadIdx.x;
   const int row = l & 15;
@@ -32,8 +35,8 @@ void mx_probe_kernel(const unsigned char* __restrict__ a_bytes,  // [16][128]
     a.b[j] = a_bytes[row * 128 + kb * 32 + j];
     b.b[j] = b_bytes[row * 128 + kb * 32 + j];
   }
-  const int sa = a_scales[row * 4 + kb];
-  const int sb = b_scales[row * 4 + kb];
+  const int sa = (int)a_scales[l];
+  const int sb = (int)b_scales[l];
   f32x4 c = {0.f, 0.f, 0.f, 0.f};
   c = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
       a.v, b.v, c, 0, 0, 0, sa, 0, sb);
@@ -45,12 +48,12 @@ void mx_probe_kernel(const unsigned char* __restrict__ a_bytes,  // [16][128]
 
 extern "C" {
 
-void mx_probe_launch(const void* a, const void* b, const void* as,
-                     const void* bs, float* c, hipStream_t stream) {
+void mx_probe_launch(const void* a, const void* b, const unsigned int* as,
+                     const unsigned int* bs, float* c,
+                     hipStream_t stream) {
   hipLaunchKernelGGL(mx_probe_kernel, dim3(1), dim3(64), 0, stream,
                      (const unsigned char*)a, (const unsigned char*)b,
-                     (const unsigned char*)as, (const unsigned char*)bs,
-                     c);
+                     as, bs, c);
 }
 
 }  // extern "C"
