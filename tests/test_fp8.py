@@ -240,8 +240,9 @@ def test_gpu_mx_probe_scale_mapping():
         hits = []
         for lane in range(64):
             for byte in range(4):
-                sa = unit.clone()
-                sa[lane] = int(sa[lane]) + (1 << (8 * byte))  # 127->128
+                sa_np = unit.cpu().numpy().view(np.uint32).copy()
+                sa_np[lane] += np.uint32(1 << (8 * byte))  # 127->128
+                sa = torch.from_numpy(sa_np.view(np.int32)).to(unit.device)
                 tot = float(ext.mx_probe(a, b, sa, unit).sum())
                 if abs(tot - 2.0 * base) < 1e-6:
                     hits.append((lane, byte))
