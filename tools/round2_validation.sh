@@ -54,8 +54,9 @@ block3)  # serving bench (~8 min incl. 265 s batch-32 find)
   # A/B: hipGraph replay of the engine forward (validated in block1)
   MASKRCNN_SERVING_HIPGRAPH=1 timeout 700 python bench/serving_bench.py \
       --requests 96 --client-procs 4 --client-threads 8 --warmup 16 \
-      > gpurun_out/r2_serving_graph.json 2>/dev/null
+      > gpurun_out/r2_serving_graph.json 2> gpurun_out/r2_serving_graph.err
   cat gpurun_out/r2_serving_graph.json
+  tail -12 gpurun_out/r2_serving_graph.err
   ;;
 block4)  # steady-state rocprof of the tuned step (profile AFTER warmup
          # by profiling a second process on the warm box: find-db is
