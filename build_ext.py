@@ -16,10 +16,13 @@ from pathlib import Path
 
 ROOT = Path(__file__).resolve().parent
 CSRC = ROOT / "csrc"
-BUILD = ROOT / "build" / "ext"
-OUT = ROOT / "maskrcnn_amd" / "_C.so"
 
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
+# MASKRCNN_ASAN=1: host-side AddressSanitizer build (device code is not
+# instrumented — -fsanitize applies to the host pass only). Run tests
+# with LD_PRELOAD of libasan (tools/asan_check.sh). Objects/so land in
+# separate paths so a sanitized build never ships by accident.
+ASAN = os.environ.get("MASKRCNN_ASAN", "0") == "1"
 
 
 def torch_paths():
@@ -30,6 +33,8 @@ def torch_paths():
 
 
 def build(verbose: bool = True) -> Path:
+    BUILD = ROOT / "build" / ("ext_asan" if ASAN else "ext")
+    OUT = ROOT / "maskrcnn_amd" / ("_C_asan.so" if ASAN else "_C.so")
     inc, libs, cxx11 = torch_paths()
     py_inc = sysconfig.get_paths()["include"]
     BUILD.mkdir(parents=True, exist_ok=True)
@@ -45,6 +50,8 @@ def build(verbose: bool = True) -> Path:
         "-Wno-deprecated-declarations",
         "-Wno-unused-result",
     ]
+    if ASAN:
+        common += ["-fsanitize=address", "-shared-libasan", "-g"]
     incs = [f"-I{p}" for p in inc] + [f"-I{py_inc}", f"-I{CSRC}"]
 
     sources = sorted(CSRC.glob("hip/*.hip")) + \
@@ -75,6 +82,8 @@ def build(verbose: bool = True) -> Path:
         ["-ltorch", "-ltorch_cpu", "-ltorch_python", "-lc10",
          "-lc10_hip", "-ltorch_hip",
          f"-Wl,-rpath,{libs[0]}"]
+    if ASAN:
+        link += ["-fsanitize=address", "-shared-libasan"]
     if verbose:
         print(" ".join(link[:6]), "...")
     subprocess.run(link, check=True)
