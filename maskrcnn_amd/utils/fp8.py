@@ -125,3 +125,12 @@ def mx_dequantize(q: torch.Tensor, exp: torch.Tensor, dim: int = -1,
     e = exp.movedim(dim, -1).float().unsqueeze(-1) - 127.0
     out = qb * torch.pow(2.0, e)
     return out.reshape(shape).movedim(-1, dim).to(dtype)
+
+
+def mx_scale_src(row: int, kblock: int):
+    """(row, 32-element k-block) -> (source lane, byte) of the
+    mfma_scale_f32_16x16x128 A-scale operand, as measured by the
+    binary-encoded probe (tests/test_fp8.py
+    test_gpu_mx_probe_scale_mapping). Placeholder: the naive
+    data-lane-symmetric rule; superseded by the probe table."""
+    return (16 * kblock + row, 0)
