@@ -51,7 +51,13 @@ def main():
         name="mask_rcnn", signature_name="serving_default"))
     req.inputs["input_image"] = numpy_to_tensor(molded)
     req.inputs["input_image_meta"] = numpy_to_tensor(meta)
-    req.output_filter = ["mrcnn_detection", "mrcnn_mask"]
+    # mrcnn_mask_top: detected-class masks only ([D,28,28] — exactly
+    # what unmold consumes), 81x fewer response bytes than the full
+    # per-class tensor. MASKRCNN_CLIENT_FULL_MASK=1 benchmarks the
+    # stock TF-Serving-shaped response instead.
+    full = os.environ.get("MASKRCNN_CLIENT_FULL_MASK") == "1"
+    req.output_filter = ["mrcnn_detection",
+                         "mrcnn_mask" if full else "mrcnn_mask_top"]
     req_bytes_cached = req.encode()  # serialize once per worker
 
     lats = []

@@ -295,10 +295,32 @@ class ModelServer:
                 request.inputs["input_anchors"]).astype(np.float32)
         # The filter is pushed into the engine: unrequested outputs skip
         # the DtoH transfer entirely.
+        #
+        # Extension (wire-compatible): output_filter entry
+        # "mrcnn_mask_top" returns the DETECTED-CLASS mask only,
+        # [B, D, 28, 28] — what unmold_detections actually consumes —
+        # instead of the full [B, D, 28, 28, C] tensor (81x fewer bytes
+        # at 81 COCO classes; the full mrcnn_mask stays available for
+        # stock TF-Serving clients).
+        want = list(request.output_filter) or None
+        mask_top = bool(want) and "mrcnn_mask_top" in want
+        if mask_top:
+            want = [w for w in want if w != "mrcnn_mask_top"]
+            for need in ("mrcnn_mask", "mrcnn_detection"):
+                if need not in want:
+                    want.append(need)
         outputs, version = state.predict(
-            images, metas, anchors,
-            outputs=list(request.output_filter) or None,
+            images, metas, anchors, outputs=want,
             version=want_version)
+        if mask_top:
+            det = outputs["mrcnn_detection"]          # [B, D, 6]
+            full = outputs["mrcnn_mask"]              # [B, D, mh, mw, C]
+            cls = det[:, :, 4].astype(np.int64)       # [B, D]
+            B, D = cls.shape
+            bb = np.arange(B)[:, None]
+            dd = np.arange(D)[None, :]
+            outputs["mrcnn_mask_top"] = \
+                full[bb, dd, :, :, cls]               # [B, D, mh, mw]
         if request.output_filter:
             outputs = {k: v for k, v in outputs.items()
                        if k in request.output_filter}

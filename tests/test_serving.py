@@ -816,3 +816,37 @@ def test_serving_bench_cpu_smoke():
     assert rec["metric"] == "serving_images_per_sec"
     assert rec["value"] > 0
     assert rec["steps"] == 8
+
+
+def test_grpc_mask_top_extension(served_model):
+    """output_filter entry "mrcnn_mask_top" returns the detected-class
+    mask [B,D,mh,mw] (81x fewer bytes than mrcnn_mask) and matches a
+    gather of the full tensor; stock mrcnn_mask stays available."""
+    server, cfg, _ = served_model
+    import grpc
+    from maskrcnn_amd.serving.client import PREDICT_METHOD
+    channel = grpc.insecure_channel(f"localhost:{server.grpc_port}")
+    stub = channel.unary_unary(
+        PREDICT_METHOD, request_serializer=lambda r: r.encode(),
+        response_deserializer=PredictResponse.decode)
+    H = cfg.IMAGE_SHAPE[0]
+    rng = np.random.RandomState(3)
+    req = PredictRequest(model_spec=ModelSpec(name="mask_rcnn"))
+    req.inputs["input_image"] = numpy_to_tensor(
+        rng.rand(1, H, H, 3).astype(np.float32) * 60)
+    meta = np.zeros((1, cfg.IMAGE_META_SIZE), np.float32)
+    meta[:, 4:7] = (H, H, 3)
+    meta[:, 7:11] = (0, 0, H, H)
+    req.inputs["input_image_meta"] = numpy_to_tensor(meta)
+    req.output_filter = ["mrcnn_detection", "mrcnn_mask",
+                         "mrcnn_mask_top"]
+    resp = stub(req, timeout=120)
+    det = tensor_to_numpy(resp.outputs["mrcnn_detection"])
+    full = tensor_to_numpy(resp.outputs["mrcnn_mask"])
+    top = tensor_to_numpy(resp.outputs["mrcnn_mask_top"])
+    assert top.shape == full.shape[:4]
+    cls = det[:, :, 4].astype(np.int64)
+    B, D = cls.shape
+    ref = full[np.arange(B)[:, None], np.arange(D)[None, :], :, :, cls]
+    np.testing.assert_array_equal(top, ref)
+    assert top.nbytes * (full.shape[-1]) == full.nbytes
