@@ -35,7 +35,13 @@ def main():
                                                 numpy_to_tensor)
 
     H = args.image_size
-    molded = np.random.randn(1, H, H, 3).astype(np.float32) * 40
+    # uint8 wire extension (default): the raw image ships at 1/4 the
+    # bytes and the server molds on-GPU. MASKRCNN_CLIENT_FP32=1 sends
+    # the stock molded-fp32 TF-Serving payload instead.
+    if os.environ.get("MASKRCNN_CLIENT_FP32") == "1":
+        molded = np.random.randn(1, H, H, 3).astype(np.float32) * 40
+    else:
+        molded = (np.random.rand(1, H, H, 3) * 255).astype(np.uint8)
     meta = np.zeros((1, 1 + 3 + 3 + 4 + 1 + args.num_classes), np.float32)
     meta[:, 4:7] = (H, H, 3)
     meta[:, 7:11] = (0, 0, H, H)

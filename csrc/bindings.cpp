@@ -391,8 +391,10 @@ torch::Tensor conv3x3_mx(torch::Tensor xq, torch::Tensor xs,
   TORCH_CHECK(wq.dim() == 3 && wq.size(0) == 9 && wq.size(2) == Cin,
               "wq must be [9, Cout, Cin]");
   const int Cout = wq.size(1);
-  TORCH_CHECK(xs.size(3) == Cin / 32 && ws.size(2) == Cin / 32,
-              "scale tensors must have Cin/32 blocks");
+  TORCH_CHECK(xs.size(3) == Cin / 64 && ws.size(2) == Cin / 64,
+              "scale tensors must have Cin/64 blocks (the gfx950 "
+              "mfma_scale 16x16x128 applies one e8m0 per 64 k at "
+              "OPSEL=0 — probe-measured)");
   auto bias_f = bias.to(torch::kFloat).contiguous();
   auto out = torch::empty(
       {B, Cout, H, W},

@@ -4,22 +4,23 @@
 // MFMA (16x16x32) runs at the BF16 rate on gfx950, so conv3x3_fp8.hip
 // can never beat bf16 on compute; the MX instruction
 // v_mfma_scale_f32_16x16x128_f8f6f4 runs K=128 per issue (~5 PF dense,
-// 4647 TF measured on MX GEMM) with one e8m0 scale per 32-element
-// K-block per row.
+// 4647 TF measured on MX GEMM) with block-scaled e8m0 exponents.
 //
-// Layout (CK ck/utility/amd_xdlops.hpp intrin_mfma_scale_f32_16x16x128
-// f8f6f4<16,16> and xdlops_gemm.hpp mfma_type: num_input_blks=4,
-// k_per_blk=32): operand lane l holds row l%16, K-block l/16 (32
-// consecutive e4m3 bytes); scale operand is an i32 whose OPSEL-selected
-// byte (we use 0) is that block's e8m0 exponent; C fragment is the
-// usual 16x16 layout (c[r]: row (l/16)*4+r, col l%16).
+// Layout (pinned EMPIRICALLY by tests/test_fp8.py probes on MI355X,
+// gpurun_out/r10_probe.log): operand lane l holds row l%16, K-block
+// l/16 (32 consecutive e4m3 bytes); C fragment is the usual 16x16
+// layout (c[r]: row (l/16)*4+r, col l%16). The SCALE operand at
+// OPSEL=(0,0) is read from byte 0 of lane `row` for k in [0,64) and
+// byte 0 of lane `16+row` for k in [64,128) — ONE e8m0 per 64
+// consecutive k (not per 32); quantization uses block=64 to match
+// (utils/fp8.py mx_quantize(block=64), mx_scale_src).
 //
 // Structure: conv3x3_v2's 128x128-tile double-buffered glds skeleton
 // with BK=128 fp8 bytes per K-step; A/B data staged via width-16 glds,
-// per-row scale words via width-4 glds.
+// per-row scale bytes via width-2 glds.
 //
-// Inputs: xq [B,H,W,Cin] e4m3 bytes (channels_last), xs [B,H,W,Cin/32]
-// e8m0 bytes; wq [9,Cout,Cin] e4m3 tap-major, ws [9,Cout,Cin/32] e8m0.
+// Inputs: xq [B,H,W,Cin] e4m3 bytes (channels_last), xs [B,H,W,Cin/64]
+// e8m0 bytes; wq [9,Cout,Cin] e4m3 tap-major, ws [9,Cout,Cin/64] e8m0.
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
@@ -33,7 +34,7 @@ constexpr int BN = 128;
 constexpr int BK = 128;                  // fp8 bytes per k-step
 constexpr int ROW_B = BK;                // 128 B per LDS row
 constexpr int TILE_B = BM * ROW_B;       // 16 KiB per tile
-constexpr int SCL_B = BM * 4;            // 512 B of scales per tile
+constexpr int SCL_B = BM * 2;            // 2 block-64 scales per row
 
 __global__ __launch_bounds__(256)
 void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
@@ -53,7 +54,7 @@ void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
   const int wave = tid >> 6;
   const int wave_m = wave >> 1;
   const int wave_n = wave & 1;
-  const int KS = Cin / 32;               // scale bytes per pixel
+  const int KS = Cin / 64;               // block-64 scale bytes/pixel
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // [buf][A 16K | B 16K | As 512 | Bs 512]
@@ -79,7 +80,7 @@ void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
     apx_y[i] = (int)(rest % H);
     apx_b[i] = rest / H;
   }
-  // Scale staging (width-4 glds): wave 0 stages A scales, wave 1 B
+  // Scale staging (width-2 glds): wave 0 stages A scales, wave 1 B
   // scales; 2 instrs of 64 rows each. Lane l -> row l.
   int spx_x[2], spx_y[2];
   long spx_b[2];
@@ -136,13 +137,13 @@ void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
         const bool ok =
             sm_ok[i] && sy >= 0 && sy < H && sx >= 0 && sx < W;
         const unsigned char* src = ok
-            ? xs + ((spx_b[i] * H + sy) * (long)W + sx) * KS + k0 / 32
+            ? xs + ((spx_b[i] * H + sy) * (long)W + sx) * KS + k0 / 64
             : zeros;
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) unsigned int*)src,
             (__attribute__((address_space(3))) unsigned int*)
-                &smem[as_off(buf) + i * 64 * 4],
-            4, 0, 0);
+                &smem[as_off(buf) + i * 64 * 2],
+            2, 0, 0);
       }
     } else if (wave == 1) {
 #pragma unroll
@@ -150,13 +151,13 @@ void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
         const int row = i * 64 + lane;
         const int n = n0 + row;
         const unsigned char* src = (n < Cout)
-            ? ws + ((long)tap * Cout + n) * KS + k0 / 32
+            ? ws + ((long)tap * Cout + n) * KS + k0 / 64
             : zeros;
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) unsigned int*)src,
             (__attribute__((address_space(3))) unsigned int*)
-                &smem[bs_off(buf) + i * 64 * 4],
-            4, 0, 0);
+                &smem[bs_off(buf) + i * 64 * 2],
+            2, 0, 0);
       }
     }
   };
@@ -172,6 +173,13 @@ void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
     const int cur = t & 1;
     if (t + 1 < n_steps) stage(t + 1, cur ^ 1);
 
+    // Scale registers: the instruction reads byte 0 of lanes 0..31
+    // only — lane r carries row r's k[0,64) scale, lane 16+r row r's
+    // k[64,128) scale (probe-measured; lanes 32..63 ignored). So each
+    // lane loads the scale for the (row, half) it CARRIES, not its own
+    // fragment's.
+    const int s_carry_row = lane & 15;
+    const int s_carry_half = (lane >> 4) & 1;
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const int arow = wave_m * 64 + mi * 16 + frag_row;
@@ -180,8 +188,9 @@ void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
       i32x4 ahi = *(const i32x4*)(ap + 16);
       i32x8 a = {alo[0], alo[1], alo[2], alo[3],
                  ahi[0], ahi[1], ahi[2], ahi[3]};
-      const int sa =
-          (unsigned char)smem[as_off(cur) + arow * 4 + frag_kb];
+      const int sa_row = wave_m * 64 + mi * 16 + s_carry_row;
+      const int sa = (unsigned char)
+          smem[as_off(cur) + sa_row * 2 + s_carry_half];
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         const int bcol = wave_n * 64 + ni * 16 + frag_row;
@@ -190,8 +199,9 @@ void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
         i32x4 bhi = *(const i32x4*)(bp + 16);
         i32x8 b = {blo[0], blo[1], blo[2], blo[3],
                    bhi[0], bhi[1], bhi[2], bhi[3]};
-        const int sb =
-            (unsigned char)smem[bs_off(cur) + bcol * 4 + frag_kb];
+        const int sb_col = wave_n * 64 + ni * 16 + s_carry_row;
+        const int sb = (unsigned char)
+            smem[bs_off(cur) + sb_col * 2 + s_carry_half];
         acc[mi][ni] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
             a, b, acc[mi][ni], 0 /*cbsz: e4m3*/, 0 /*blgp: e4m3*/,
             0 /*OpselA*/, sa, 0 /*OpselB*/, sb);

@@ -61,23 +61,25 @@ def conv3x3_fp8(x, w_packed_u8, w_scale, bias, relu=True):
 
 
 def pack_weight_mx(w: torch.Tensor):
-    """[Cout, Cin, 3, 3] -> ([9, Cout, Cin] u8 e4m3, [9, Cout, Cin/32]
+    """[Cout, Cin, 3, 3] -> ([9, Cout, Cin] u8 e4m3, [9, Cout, Cin/64]
     u8 e8m0) tap-major for the MX block-scaled kernel: one power-of-two
-    scale per (tap, out-channel, 32-in-channel block)."""
+    scale per (tap, out-channel, 64-in-channel block) — the gfx950
+    mfma_scale 16x16x128 scale granularity at OPSEL=0
+    (probe-measured, utils/fp8.py mx_scale_src)."""
     from ..utils.fp8 import mx_quantize
     Cout, Cin, _, _ = w.shape
     wt = w.float().permute(2, 3, 0, 1).reshape(9, Cout, Cin)
-    q, exp = mx_quantize(wt, dim=-1)
+    q, exp = mx_quantize(wt, dim=-1, block=64)
     return (q.view(torch.uint8).contiguous(), exp.contiguous())
 
 
 def quantize_activations_mx(x: torch.Tensor):
-    """Per-(pixel, 32-channel-block) MX quantization of a channels_last
-    activation. Returns (xq [B,H,W,C] u8 contiguous, xs [B,H,W,C/32]
+    """Per-(pixel, 64-channel-block) MX quantization of a channels_last
+    activation. Returns (xq [B,H,W,C] u8 contiguous, xs [B,H,W,C/64]
     u8 e8m0). All on GPU, no host sync."""
     from ..utils.fp8 import mx_quantize
     xn = x.permute(0, 2, 3, 1).float()        # NHWC values
-    q, exp = mx_quantize(xn, dim=-1)
+    q, exp = mx_quantize(xn, dim=-1, block=64)
     return q.view(torch.uint8).contiguous(), exp.contiguous()
 
 

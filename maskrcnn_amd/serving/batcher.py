@@ -113,8 +113,11 @@ class DynamicBatcher:
                     item = self.queue.get(timeout=remaining)
                 except queue.Empty:
                     break
-                if item[0].shape[1:] != first[0].shape[1:]:
-                    # Different molded shape: flush current, start new.
+                if (item[0].shape[1:] != first[0].shape[1:] or
+                        item[0].dtype != first[0].dtype):
+                    # Different molded shape (or fp32-vs-uint8 wire
+                    # format): flush current, start new — a mixed
+                    # concat would silently upcast unmolded uint8.
                     self._run(batch)
                     batch = [item]
                     n = item[0].shape[0]

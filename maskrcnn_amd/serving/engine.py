@@ -158,7 +158,22 @@ class InferenceEngine:
             # NHWC numpy IS the channels_last layout of an NCHW tensor:
             # upload as-is (pinned, async), then view-permute — no host
             # transpose and no extra device copy when CHANNELS_LAST.
-            up = self._upload(input_image.astype(np.float32, copy=False))
+            #
+            # uint8 extension: an un-molded uint8 image uploads at 1/4
+            # the bytes (wire AND PCIe); mold (float - MEAN_PIXEL) runs
+            # on the GPU. fp32 input stays the stock molded contract.
+            if input_image.dtype == np.uint8:
+                up = self._upload(input_image)
+                mean = getattr(self, "_mean_pixel_t", None)
+                if mean is None or mean.device != self.device:
+                    mean = torch.tensor(
+                        self.config.MEAN_PIXEL,
+                        dtype=torch.float32, device=self.device)
+                    self._mean_pixel_t = mean
+                up = up.float() - mean.view(1, 1, 1, 3)
+            else:
+                up = self._upload(
+                    input_image.astype(np.float32, copy=False))
             images = up.permute(0, 3, 1, 2)
             if self.config.CHANNELS_LAST and self.device.type == "cuda":
                 images = images.contiguous(

@@ -283,8 +283,11 @@ class ModelServer:
                 f"model {name!r} version {want_version} not loaded "
                 f"(serving {sorted(state.versions) or state.version})")
         try:
-            images = tensor_to_numpy(request.inputs["input_image"]) \
-                .astype(np.float32)
+            images = tensor_to_numpy(request.inputs["input_image"])
+            if images.dtype != np.uint8:
+                # fp32 = the stock molded contract; uint8 = wire/PCIe
+                # extension, molded on the GPU (engine.predict).
+                images = images.astype(np.float32)
             metas = tensor_to_numpy(request.inputs["input_image_meta"]) \
                 .astype(np.float32)
         except KeyError as e:

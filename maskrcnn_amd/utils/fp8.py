@@ -93,19 +93,21 @@ MX_BLOCK = 32
 _E4M3_EMAX = 8  # floor(log2(448))
 
 
-def mx_quantize(x: torch.Tensor, dim: int = -1):
-    """Block-32 MX quantization along `dim`.
+def mx_quantize(x: torch.Tensor, dim: int = -1, block: int = MX_BLOCK):
+    """Block-scaled MX quantization along `dim` (default block 32; the
+    gfx950 mfma_scale 16x16x128 instruction applies ONE e8m0 per 64
+    consecutive k at OPSEL=0 — measured by the r10 scale-mapping probe —
+    so the conv kernel quantizes with block=64).
 
-    Returns (q e4m3fn same shape, exp uint8 with dim shrunk by 32):
+    Returns (q e4m3fn same shape, exp uint8 with dim shrunk by block):
     x ~= q.float() * 2^(exp - 127) broadcast over each block.
-    `dim`'s size must be a multiple of 32.
     """
     dim = dim % x.dim()
     K = x.shape[dim]
-    assert K % MX_BLOCK == 0, f"dim size {K} not a multiple of 32"
+    assert K % block == 0, f"dim size {K} not a multiple of {block}"
     xf = x.float().movedim(dim, -1)
     shape = xf.shape
-    xb = xf.reshape(*shape[:-1], K // MX_BLOCK, MX_BLOCK)
+    xb = xf.reshape(*shape[:-1], K // block, block)
     amax = xb.abs().amax(dim=-1, keepdim=True).clamp(min=2.0 ** -126)
     e = torch.floor(torch.log2(amax)) - _E4M3_EMAX
     e = e.clamp(-127, 127)
@@ -117,11 +119,12 @@ def mx_quantize(x: torch.Tensor, dim: int = -1):
 
 
 def mx_dequantize(q: torch.Tensor, exp: torch.Tensor, dim: int = -1,
-                  dtype=torch.float32) -> torch.Tensor:
+                  dtype=torch.float32,
+                  block: int = MX_BLOCK) -> torch.Tensor:
     dim = dim % q.dim()
     qf = q.float().movedim(dim, -1)
     shape = qf.shape
-    qb = qf.reshape(*shape[:-1], shape[-1] // MX_BLOCK, MX_BLOCK)
+    qb = qf.reshape(*shape[:-1], shape[-1] // block, block)
     e = exp.movedim(dim, -1).float().unsqueeze(-1) - 127.0
     out = qb * torch.pow(2.0, e)
     return out.reshape(shape).movedim(-1, dim).to(dtype)
@@ -129,8 +132,8 @@ def mx_dequantize(q: torch.Tensor, exp: torch.Tensor, dim: int = -1,
 
 def mx_scale_src(row: int, kblock: int):
     """(row, 32-element k-block) -> (source lane, byte) of the
-    mfma_scale_f32_16x16x128 A-scale operand, as measured by the
-    binary-encoded probe (tests/test_fp8.py
-    test_gpu_mx_probe_scale_mapping). Placeholder: the naive
-    data-lane-symmetric rule; superseded by the probe table."""
-    return (16 * kblock + row, 0)
+    mfma_scale_f32_16x16x128 A-scale operand at OPSEL=(0,0), as
+    MEASURED by the binary-encoded probe on MI355X (gpurun_out/
+    r10_probe.log): blocks {0,1} read byte 0 of lane `row`, blocks
+    {2,3} byte 0 of lane `16+row` — one e8m0 per 64 consecutive k."""
+    return (16 * (kblock // 2) + row, 0)

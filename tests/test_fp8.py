@@ -125,11 +125,11 @@ def test_pack_weight_mx_roundtrip():
     from maskrcnn_amd.ops.conv3x3 import pack_weight_mx
     from maskrcnn_amd.utils.fp8 import FP8_DTYPE, mx_dequantize
     torch.manual_seed(0)
-    w = torch.randn(8, 32, 3, 3) * 0.3
+    w = torch.randn(8, 64, 3, 3) * 0.3
     q, exp = pack_weight_mx(w)
-    assert q.shape == (9, 8, 32) and exp.shape == (9, 8, 1)
-    deq = mx_dequantize(q.view(FP8_DTYPE), exp, dim=-1)
-    ref = w.float().permute(2, 3, 0, 1).reshape(9, 8, 32)
+    assert q.shape == (9, 8, 64) and exp.shape == (9, 8, 1)
+    deq = mx_dequantize(q.view(FP8_DTYPE), exp, dim=-1, block=64)
+    ref = w.float().permute(2, 3, 0, 1).reshape(9, 8, 64)
     err = (deq - ref).abs().max() / ref.abs().max()
     assert float(err) < 0.07, float(err)
 
@@ -168,9 +168,9 @@ def test_gpu_mx_kernel_matches_mx_reference(cin, cout, h, w, batch):
 
     # Reference: conv with the SAME quantized operands in fp32.
     xq, xs = quantize_activations_mx(x)
-    x_deq = mx_dequantize(xq.view(FP8_DTYPE), xs, dim=-1) \
+    x_deq = mx_dequantize(xq.view(FP8_DTYPE), xs, dim=-1, block=64) \
         .permute(0, 3, 1, 2)                       # NCHW fp32
-    w_deq = mx_dequantize(wq.view(FP8_DTYPE), ws, dim=-1) \
+    w_deq = mx_dequantize(wq.view(FP8_DTYPE), ws, dim=-1, block=64) \
         .reshape(3, 3, cout, cin).permute(2, 3, 0, 1)
     ref = F.relu(F.conv2d(x_deq, w_deq, b, padding=1))
     torch.testing.assert_close(out.float(), ref, atol=0.1, rtol=0.05)

@@ -850,3 +850,35 @@ def test_grpc_mask_top_extension(served_model):
     ref = full[np.arange(B)[:, None], np.arange(D)[None, :], :, :, cls]
     np.testing.assert_array_equal(top, ref)
     assert top.nbytes * (full.shape[-1]) == full.nbytes
+
+
+def test_grpc_uint8_image_extension(served_model):
+    """uint8 input_image (wire extension, 1/4 the bytes) is molded
+    server-side and must match the stock client-molded fp32 request."""
+    server, cfg, _ = served_model
+    import grpc
+    from maskrcnn_amd.serving.client import PREDICT_METHOD
+    channel = grpc.insecure_channel(f"localhost:{server.grpc_port}")
+    stub = channel.unary_unary(
+        PREDICT_METHOD, request_serializer=lambda r: r.encode(),
+        response_deserializer=PredictResponse.decode)
+    H = cfg.IMAGE_SHAPE[0]
+    rng = np.random.RandomState(9)
+    raw = (rng.rand(1, H, H, 3) * 255).astype(np.uint8)
+    meta = np.zeros((1, cfg.IMAGE_META_SIZE), np.float32)
+    meta[:, 4:7] = (H, H, 3)
+    meta[:, 7:11] = (0, 0, H, H)
+
+    def call(img):
+        req = PredictRequest(model_spec=ModelSpec(name="mask_rcnn"))
+        req.inputs["input_image"] = numpy_to_tensor(img)
+        req.inputs["input_image_meta"] = numpy_to_tensor(meta)
+        req.output_filter = ["mrcnn_detection"]
+        return tensor_to_numpy(
+            stub(req, timeout=120).outputs["mrcnn_detection"])
+
+    molded = raw.astype(np.float32) - np.array(cfg.MEAN_PIXEL,
+                                               np.float32)
+    det_fp32 = call(molded)
+    det_u8 = call(raw)
+    np.testing.assert_allclose(det_u8, det_fp32, atol=1e-5)
