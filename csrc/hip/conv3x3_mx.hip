@@ -129,35 +129,28 @@ void conv3x3_mx_kernel(const unsigned char* __restrict__ xq,
               &smem[b_off(buf) + (wave * 32 + i * 8) * ROW_B],
           16, 0, 0);
     }
-    if (wave == 0) {
-#pragma unroll
-      for (int i = 0; i < 2; ++i) {
-        const int row = i * 64 + lane;
+    // Scales: plain global loads + ds_write (tiny traffic; the
+    // per-step __syncthreads drains lgkm+vmcnt, so no glds discipline
+    // needed). Threads 0..127: A rows; 128..255: B rows.
+    {
+      const int r = tid & 127;
+      if (tid < 128) {
+        const int i = r >> 6;
         const int sy = spx_y[i] + dy, sx = spx_x[i] + dx;
         const bool ok =
             sm_ok[i] && sy >= 0 && sy < H && sx >= 0 && sx < W;
         const unsigned char* src = ok
             ? xs + ((spx_b[i] * H + sy) * (long)W + sx) * KS + k0 / 64
             : zeros;
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)src,
-            (__attribute__((address_space(3))) unsigned int*)
-                &smem[as_off(buf) + i * 64 * 2],
-            2, 0, 0);
-      }
-    } else if (wave == 1) {
-#pragma unroll
-      for (int i = 0; i < 2; ++i) {
-        const int row = i * 64 + lane;
-        const int n = n0 + row;
+        *(unsigned short*)&smem[as_off(buf) + r * 2] =
+            *(const unsigned short*)src;
+      } else {
+        const int n = n0 + r;
         const unsigned char* src = (n < Cout)
             ? ws + ((long)tap * Cout + n) * KS + k0 / 64
             : zeros;
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)src,
-            (__attribute__((address_space(3))) unsigned int*)
-                &smem[bs_off(buf) + i * 64 * 2],
-            2, 0, 0);
+        *(unsigned short*)&smem[bs_off(buf) + r * 2] =
+            *(const unsigned short*)src;
       }
     }
   };
