@@ -135,8 +135,6 @@ def test_pack_weight_mx_roundtrip():
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False,
-                   reason="MX kernel pending first GPU validation")
 @pytest.mark.parametrize("cin,cout,h,w,batch", [
     (256, 256, 14, 14, 8),     # mask-head shape
     (256, 256, 32, 32, 2),

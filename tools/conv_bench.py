@@ -152,7 +152,44 @@ def bench_fp8():
               f"bf16v2={bf:7.3f}ms {bf / f8:5.2f}x")
 
 
+def bench_mx():
+    """MX block-scaled e4m3 conv (K=128 mfma_scale) vs the best bf16
+    hand kernel for the shape. Reports both end-to-end (incl. the
+    on-the-fly activation quant pass) and kernel-only time."""
+    from maskrcnn_amd.ops._ext import require_ext
+    from maskrcnn_amd.ops.conv3x3 import (conv3x3_mx, pack_weight,
+                                          pack_weight_mx,
+                                          quantize_activations_mx)
+    ext = require_ext()
+    dev = torch.device("cuda:0")
+    print("\nconv3x3 MX e4m3 (K=128 block-scaled) vs bf16 best:")
+    for (B, Cin, H, W, Cout) in [(2, 256, 64, 64, 256),
+                                 (2, 256, 256, 256, 256),
+                                 (400, 256, 14, 14, 256),
+                                 (16, 256, 84, 84, 256)]:
+        x = torch.randn(B, Cin, H, W, device=dev).bfloat16() \
+            .to(memory_format=torch.channels_last)
+        w = (torch.randn(Cout, Cin, 3, 3, device=dev) * 0.05)
+        b = torch.randn(Cout, device=dev)
+        wq, ws = pack_weight_mx(w)
+        wp = pack_weight(w)
+        bf = b.float()
+        mx = timeit(lambda: conv3x3_mx(x, wq, ws, b, relu=True))
+        xq, xs = quantize_activations_mx(x)
+        mxk = timeit(lambda: ext.conv3x3_mx(xq, xs, wq, ws, bf, True))
+        M = B * H * W
+        ver = 5 if M >= 131072 else (3 if M >= 32768 else 1)
+        bf16 = timeit(lambda: ext.conv3x3_fwd(x, wp, bf, True, ver))
+        flops = 2.0 * B * H * W * Cout * Cin * 9
+        print(f"{str((B, Cin, H, W, Cout)):>28} "
+              f"mx+q={mx:7.3f}ms({flops / (mx * 1e-3) / 1e12:5.0f}TF) "
+              f"mx-k={mxk:7.3f}ms({flops / (mxk * 1e-3) / 1e12:5.0f}TF) "
+              f"bf16v{ver}={bf16:7.3f}ms ratio={bf16 / mx:5.2f}x "
+              f"(kernel-only {bf16 / mxk:5.2f}x)")
+
+
 if __name__ == "__main__":
     main()
     bench_wrw()
     bench_fp8()
+    bench_mx()
