@@ -58,6 +58,7 @@ void conv3x3_mx_launch(const void*, const void*, const void*, const void*,
                        int, int, bool, hipStream_t);
 void mx_probe_launch(const void*, const void*, const unsigned int*,
                      const unsigned int*, float*, hipStream_t);
+void mx_quant_launch(const void*, void*, void*, long, int, hipStream_t);
 }
 
 namespace {
@@ -352,6 +353,26 @@ torch::Tensor deconv2x2(torch::Tensor x, torch::Tensor w_packed,
   return out;
 }
 
+std::tuple<torch::Tensor, torch::Tensor> mx_quant(torch::Tensor x) {
+  // x: [B,C,H,W] bf16 channels_last (C % 64 == 0). Returns
+  // (xq [B,H,W,C] u8 e4m3, xs [B,H,W,C/64] u8 e8m0) — the conv3x3_mx
+  // input format — in one fused pass.
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+              x.scalar_type() == torch::kBFloat16, "bf16 [B,C,H,W]");
+  TORCH_CHECK(is_channels_last(x), "channels_last required");
+  const int B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  TORCH_CHECK(C % 64 == 0, "C must be a multiple of 64");
+  const long M = (long)B * H * W;
+  const int KS = C / 64;
+  auto q = torch::empty({B, H, W, C}, x.options().dtype(torch::kByte));
+  auto s = torch::empty({B, H, W, KS},
+                        x.options().dtype(torch::kByte));
+  if (M)
+    mx_quant_launch(x.data_ptr(), q.data_ptr(), s.data_ptr(), M * KS,
+                    KS, cur_stream());
+  return {q, s};
+}
+
 torch::Tensor mx_probe(torch::Tensor a, torch::Tensor b,
                        torch::Tensor as, torch::Tensor bs) {
   // a,b: [16,128] u8 e4m3; as,bs: [64] int32 — the FULL per-lane scale
@@ -545,6 +566,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "2x2 stride-2 transposed conv + bias (+ReLU), MFMA (CDNA4)");
   m.def("conv3x3_mx", &conv3x3_mx,
         "MX block-scaled e4m3 3x3 conv (mfma_scale 16x16x128, gfx950)");
+  m.def("mx_quant", &mx_quant,
+        "fused MX block-64 activation quantization (bf16 NHWC -> e4m3)");
   m.def("mx_probe", &mx_probe,
         "one-instruction mfma_scale layout probe (test-only)");
   m.def("conv3x3_wrw", &conv3x3_wrw,

@@ -76,7 +76,14 @@ def pack_weight_mx(w: torch.Tensor):
 def quantize_activations_mx(x: torch.Tensor):
     """Per-(pixel, 64-channel-block) MX quantization of a channels_last
     activation. Returns (xq [B,H,W,C] u8 contiguous, xs [B,H,W,C/64]
-    u8 e8m0). All on GPU, no host sync."""
+    u8 e8m0). All on GPU, no host sync. One fused HIP pass
+    (csrc/hip/mx_quant.hip) when eligible — the eager-torch version is
+    ~6 kernels and cost 2-3x the MX conv itself (r13_conv_ab)."""
+    ext = load_ext()
+    if (ext is not None and x.is_cuda and x.dtype == torch.bfloat16 and
+            x.is_contiguous(memory_format=torch.channels_last) and
+            x.size(1) % 64 == 0):
+        return ext.mx_quant(x)
     from ..utils.fp8 import mx_quantize
     xn = x.permute(0, 2, 3, 1).float()        # NHWC values
     q, exp = mx_quantize(xn, dim=-1, block=64)

@@ -272,3 +272,31 @@ def test_gpu_mx_probe_scale_mapping():
     expected = {(m, blk): mx_scale_src(m, blk)
                 for m in range(16) for blk in range(4)}
     assert amap == expected, amap
+
+
+@pytest.mark.gpu
+def test_gpu_mx_quant_kernel_matches_torch():
+    """Fused HIP MX quantization == the torch mx_quantize reference
+    (same block-64 exponents; values equal after dequantization —
+    v_cvt_pk_fp8_f32 and torch's fp8 cast both round-to-nearest-even)."""
+    import torch
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from maskrcnn_amd.ops._ext import require_ext
+    from maskrcnn_amd.utils.fp8 import FP8_DTYPE, mx_dequantize, \
+        mx_quantize
+    ext = require_ext()
+    torch.manual_seed(0)
+    x = (torch.randn(3, 128, 17, 23, device="cuda") *
+         torch.logspace(-3, 3, 128, device="cuda").view(1, -1, 1, 1)) \
+        .bfloat16().to(memory_format=torch.channels_last)
+    q, s = ext.mx_quant(x)
+    assert q.shape == (3, 17, 23, 128) and s.shape == (3, 17, 23, 2)
+    qt, st = mx_quantize(x.permute(0, 2, 3, 1).float(), dim=-1,
+                         block=64)
+    np_eq = (s == st.cuda()).float().mean().item()
+    assert np_eq > 0.999, np_eq  # identical exponents
+    deq_k = mx_dequantize(q.view(FP8_DTYPE), s, dim=-1, block=64)
+    deq_t = mx_dequantize(qt.view(FP8_DTYPE) if qt.dtype != FP8_DTYPE
+                          else qt, st, dim=-1, block=64).cuda()
+    torch.testing.assert_close(deq_k, deq_t, atol=1e-3, rtol=1e-2)
