@@ -115,10 +115,12 @@ class FPNMaskHead(nn.Module):
         return x
 
     def _conv_stack_fp8(self, x):
-        """e4m3-compute stack: BN-folded weights quantized once and
-        cached (weights are static in serving; the cache is dropped by
-        train() via the _fp8_cache attribute check in forward)."""
-        from ..ops.conv3x3 import conv3x3_fp8, pack_weight_fp8
+        """fp8-compute stack via the MX block-scaled kernel (K=128
+        mfma_scale — the 2x-rate path; the non-MX e4m3 kernel runs at
+        the bf16 MFMA rate and measured 0.5x, r2_conv_ab). BN-folded
+        weights are MX-quantized once and cached (static in serving;
+        the cache is dropped by train() via the _fp8_cache check)."""
+        from ..ops.conv3x3 import conv3x3_mx, pack_weight_mx
         if getattr(self, "_fp8_cache", None) is None:
             import torch as _t
             cache = []
@@ -129,13 +131,12 @@ class FPNMaskHead(nn.Module):
                     s = bn.weight * (bn.running_var + bn.eps).rsqrt()
                     w_eff = conv.weight * s.view(-1, 1, 1, 1)
                     b_eff = bn.bias + (conv.bias - bn.running_mean) * s
-                    qp, scale = pack_weight_fp8(w_eff)
-                    cache.append((qp.to(x.device),
-                                  scale.to(x.device),
+                    wq, ws = pack_weight_mx(w_eff)
+                    cache.append((wq.to(x.device), ws.to(x.device),
                                   b_eff.float().to(x.device)))
             self._fp8_cache = cache
-        for qp, scale, b_eff in self._fp8_cache:
-            x = conv3x3_fp8(x, qp, scale, b_eff, relu=True)
+        for wq, ws, b_eff in self._fp8_cache:
+            x = conv3x3_mx(x, wq, ws, b_eff, relu=True)
         return x
 
     def train(self, mode: bool = True):

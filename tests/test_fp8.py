@@ -175,7 +175,6 @@ def test_gpu_mx_kernel_matches_mx_reference(cin, cout, h, w, batch):
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False, reason="layout probe — informational")
 def test_gpu_mx_probe_layout():
     """Empirically pin the mfma_scale_f32_16x16x128 operand layout:
     one-hot A at (m0,k0) and B at (n0,k0) under the assumed
@@ -211,7 +210,6 @@ def test_gpu_mx_probe_layout():
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False, reason="layout probe — informational")
 def test_gpu_mx_probe_scale_mapping():
     """Recover the FULL A-scale source map of mfma_scale_f32_16x16x128:
     for each (row, k-block), which (lane, byte) of the scale operand
