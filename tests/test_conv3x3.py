@@ -85,12 +85,7 @@ def test_gpu_kernel_matches_reference(cin, cout, h, w):
 
 
 @pytest.mark.gpu
-@pytest.mark.parametrize("version", [
-    2, 3, 4,
-    pytest.param(5, marks=pytest.mark.xfail(
-        strict=False, reason="v5 (st_16x32 swizzle) pending first GPU "
-        "validation")),
-])
+@pytest.mark.parametrize("version", [2, 3, 4, 5])
 @pytest.mark.parametrize("cin,cout,h,w,batch", [
     (256, 256, 64, 64, 2),    # exact-tile case
     (256, 256, 50, 30, 3),    # ragged M
