@@ -64,7 +64,8 @@ class _FpnFuseFn(torch.autograd.Function):
             go.contiguous(
                 memory_format=torch.channels_last if go.is_cuda
                 else torch.contiguous_format),
-            x, w4, None, [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+            x.to(go.dtype), w4, None, [1, 1], [0, 0], [1, 1], False,
+            [0, 0], 1,
             [ctx.needs_input_grad[0], ctx.needs_input_grad[1], False])
         grad_x = gx.to(x.dtype) if ctx.needs_input_grad[0] else None
         grad_w = gw.view_as(w).to(w.dtype) \
