@@ -47,6 +47,12 @@ def main():
     ap.add_argument("--backbone", default="resnet101")
     ap.add_argument("--weight-dtype", default="fp32",
                     choices=["fp32", "fp8"])
+    ap.add_argument("--server-procs", type=int, default=1,
+                    help="N independent ModelServer replicas on ONE "
+                         "GPU (separate processes, separate ports, "
+                         "clients round-robin): bypasses the ~28 "
+                         "unary-msgs/s gRPC-python per-process ingress "
+                         "ceiling; the engines share cuda:0")
     args = ap.parse_args()
 
     import logging
@@ -75,22 +81,58 @@ def main():
                        weight_dtype=args.weight_dtype)
     del model
 
+    servers = []
+    targets = []
     server = ModelServer("mask_rcnn", base, grpc_port=0, rest_port=None,
                          max_batch_size=args.max_batch,
                          batch_timeout_micros=args.batch_timeout_micros,
                          config=cfg)
     server.start(warmup=True)
-    target = f"127.0.0.1:{server.grpc_port}"
+    servers.append(server)
+    targets.append(f"127.0.0.1:{server.grpc_port}")
+    # Extra same-GPU replicas as SUBPROCESSES (each with its own gRPC
+    # C-core + GIL). The committed miopen_db find-db keeps their
+    # warmups cheap.
+    extra = []
+    if args.server_procs > 1:
+        for i in range(args.server_procs - 1):
+            port = 18600 + i
+            pr = subprocess.Popen(
+                [sys.executable, "-m", "maskrcnn_amd.serving.server",
+                 "--model_name", "mask_rcnn",
+                 "--model_base_path", base,
+                 "--port", str(port), "--rest_api_port", "0",
+                 "--enable_batching",
+                 "--max_batch_size", str(args.max_batch),
+                 "--batch_timeout_micros",
+                 str(args.batch_timeout_micros)],
+                stderr=subprocess.DEVNULL)
+            extra.append(pr)
+            targets.append(f"127.0.0.1:{port}")
+        # Wait for the extra replicas to come up.
+        from maskrcnn_amd.serving.client import ModelServiceClient
+        for t in targets[1:]:
+            deadline = time.time() + 600
+            while time.time() < deadline:
+                try:
+                    st = ModelServiceClient(t, timeout=5.0) \
+                        .get_model_status("mask_rcnn")
+                    if st and st[0]["state"] == "AVAILABLE":
+                        break
+                except Exception:
+                    pass
+                time.sleep(2.0)
 
     def run_clients(total_requests):
         per = max(1, total_requests // args.client_procs)
         procs = [subprocess.Popen(
-            [sys.executable, WORKER, "--target", target,
+            [sys.executable, WORKER,
+             "--target", targets[i % len(targets)],
              "--requests", str(per),
              "--threads", str(args.client_threads),
              "--image-size", str(args.image_size)],
             stdout=subprocess.PIPE, text=True)
-            for _ in range(args.client_procs)]
+            for i in range(args.client_procs)]
         lats = []
         elapsed = 0.0
         n_err = 0
@@ -137,6 +179,7 @@ def main():
         "wall_s": round(wall, 2),
         "config": {
             "model": f"{args.backbone}-FPN maskrcnn serving",
+            "server_procs": args.server_procs,
             "global_batch": args.max_batch,
             "image_size": args.image_size,
             "clients": f"{args.client_procs}p x {args.client_threads}t",

@@ -667,7 +667,8 @@ def main():  # pragma: no cover
         ap.error("--model_base_path or --model_config_file required")
     server = ModelServer(
         args.model_name, args.model_base_path, grpc_port=args.port,
-        rest_port=args.rest_api_port, metrics_port=args.metrics_port,
+        rest_port=args.rest_api_port or None,  # 0 = disable REST
+        metrics_port=args.metrics_port,
         devices=devices, model_configs=model_configs,
         max_batch_size=max_batch, batch_timeout_micros=timeout_us)
     server.start()
