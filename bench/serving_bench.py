@@ -117,7 +117,8 @@ def main():
                 try:
                     st = ModelServiceClient(t, timeout=5.0) \
                         .get_model_status("mask_rcnn")
-                    if st and st[0]["state"] == "AVAILABLE":
+                    # 30 = ModelVersionStatus.State.AVAILABLE
+                    if st and st[0]["state"] in (30, "AVAILABLE"):
                         break
                 except Exception:
                     pass

@@ -882,3 +882,88 @@ def test_grpc_uint8_image_extension(served_model):
     det_fp32 = call(molded)
     det_u8 = call(raw)
     np.testing.assert_allclose(det_u8, det_fp32, atol=1e-5)
+
+
+def test_cli_replicas_same_export(tmp_path):
+    """Two independent server CLI processes on one export (the
+    bench --server-procs scenario): both come AVAILABLE with REST
+    disabled (--rest_api_port 0) and both answer gRPC Predict."""
+    import signal
+    import socket
+    import subprocess
+    import sys
+
+    import grpc
+
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.client import (ModelServiceClient,
+                                             PREDICT_METHOD)
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+
+    class TinyConfig(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 100
+        POST_NMS_ROIS_INFERENCE = 10
+
+    cfg = TinyConfig()
+    m = MaskRCNN("inference", cfg, model_dir=str(tmp_path))
+    base = str(tmp_path / "models" / "rep_model")
+    export_saved_model(m, cfg, base, version=1)
+
+    def free_port():
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        p = s.getsockname()[1]
+        s.close()
+        return p
+
+    ports = [free_port(), free_port()]
+    procs = [subprocess.Popen(
+        [sys.executable, "-m", "maskrcnn_amd.serving.server",
+         "--model_name", "rep_model", "--model_base_path", base,
+         "--port", str(p), "--rest_api_port", "0"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+        for p in ports]
+    try:
+        H = cfg.IMAGE_SHAPE[0]
+        req = PredictRequest(model_spec=ModelSpec(name="rep_model"))
+        req.inputs["input_image"] = numpy_to_tensor(
+            np.zeros((1, H, H, 3), np.float32))
+        meta = np.zeros((1, cfg.IMAGE_META_SIZE), np.float32)
+        meta[:, 4:7] = (H, H, 3)
+        meta[:, 7:11] = (0, 0, H, H)
+        req.inputs["input_image_meta"] = numpy_to_tensor(meta)
+        req.output_filter = ["mrcnn_detection"]
+        for port, proc in zip(ports, procs):
+            st = None
+            for _ in range(300):
+                if proc.poll() is not None:
+                    raise AssertionError(proc.stdout.read()[-2000:])
+                try:
+                    st = ModelServiceClient(
+                        f"127.0.0.1:{port}", timeout=2.0) \
+                        .get_model_status("rep_model")
+                    # 30 = ModelVersionStatus.State.AVAILABLE
+                    if st and st[0]["state"] in (30, "AVAILABLE"):
+                        break
+                except Exception:
+                    pass
+                time.sleep(0.2)
+            assert st and st[0]["state"] in (30, "AVAILABLE")
+            channel = grpc.insecure_channel(f"127.0.0.1:{port}")
+            stub = channel.unary_unary(
+                PREDICT_METHOD, request_serializer=lambda r: r.encode(),
+                response_deserializer=PredictResponse.decode)
+            resp = stub(req, timeout=120)
+            assert tensor_to_numpy(
+                resp.outputs["mrcnn_detection"]).shape[0] == 1
+    finally:
+        for proc in procs:
+            proc.send_signal(signal.SIGTERM)
+        for proc in procs:
+            try:
+                proc.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                proc.kill()
+                proc.wait(timeout=10)
