@@ -118,8 +118,14 @@ class DynamicBatcher:
                     # Different molded shape (or fp32-vs-uint8 wire
                     # format): flush current, start new — a mixed
                     # concat would silently upcast unmolded uint8.
+                    # `first` MUST follow the new batch: comparing later
+                    # items against the stale head let a uint8 request
+                    # join an fp32 batch (upcast to raw 0..255 floats —
+                    # garbage detections; caught by
+                    # test_grpc_mixed_uint8_fp32_clients).
                     self._run(batch)
                     batch = [item]
+                    first = item
                     n = item[0].shape[0]
                     t0 = time.monotonic()
                     continue

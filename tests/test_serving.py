@@ -967,3 +967,56 @@ def test_cli_replicas_same_export(tmp_path):
             except subprocess.TimeoutExpired:
                 proc.kill()
                 proc.wait(timeout=10)
+
+
+def test_grpc_mixed_uint8_fp32_clients(served_model):
+    """A uint8-extension client and a stock fp32 client hitting the
+    SAME server concurrently get identical results (the batcher must
+    not concat mixed-dtype batches — it splits on dtype change)."""
+    import threading
+
+    import grpc
+    from maskrcnn_amd.serving.client import PREDICT_METHOD
+    server, cfg, _ = served_model
+    channel = grpc.insecure_channel(f"localhost:{server.grpc_port}")
+    stub = channel.unary_unary(
+        PREDICT_METHOD, request_serializer=lambda r: r.encode(),
+        response_deserializer=PredictResponse.decode)
+    H = cfg.IMAGE_SHAPE[0]
+    rng = np.random.RandomState(11)
+    raw = (rng.rand(1, H, H, 3) * 255).astype(np.uint8)
+    molded = raw.astype(np.float32) - np.array(cfg.MEAN_PIXEL,
+                                               np.float32)
+    meta = np.zeros((1, cfg.IMAGE_META_SIZE), np.float32)
+    meta[:, 4:7] = (H, H, 3)
+    meta[:, 7:11] = (0, 0, H, H)
+
+    def make_req(img):
+        r = PredictRequest(model_spec=ModelSpec(name="mask_rcnn"))
+        r.inputs["input_image"] = numpy_to_tensor(img)
+        r.inputs["input_image_meta"] = numpy_to_tensor(meta)
+        r.output_filter = ["mrcnn_detection"]
+        return r
+
+    results = {}
+
+    def call(tag, img):
+        results[tag] = tensor_to_numpy(
+            stub(make_req(img), timeout=120).outputs["mrcnn_detection"])
+
+    # Interleave several of each so both dtypes share batcher windows.
+    threads = []
+    for i in range(3):
+        threads.append(threading.Thread(
+            target=call, args=(f"u8_{i}", raw)))
+        threads.append(threading.Thread(
+            target=call, args=(f"f32_{i}", molded)))
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    for i in range(3):
+        np.testing.assert_allclose(results[f"u8_{i}"],
+                                   results["f32_0"], atol=1e-5)
+        np.testing.assert_allclose(results[f"f32_{i}"],
+                                   results["f32_0"], atol=1e-5)
