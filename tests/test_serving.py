@@ -1020,3 +1020,44 @@ def test_grpc_mixed_uint8_fp32_clients(served_model):
                                    results["f32_0"], atol=1e-5)
         np.testing.assert_allclose(results[f"f32_{i}"],
                                    results["f32_0"], atol=1e-5)
+
+
+def test_batcher_mixed_shapes_concurrent(shapes_config):
+    """Three resolutions interleaved concurrently: every request gets
+    results for ITS OWN shape (regression for the stale-batch-head bug:
+    items after a shape flush were compared against the old head, so a
+    later odd-shape request could be concatenated into the wrong batch
+    — np.concatenate would raise, failing the whole window)."""
+    import threading
+
+    import torch
+    from maskrcnn_amd.serving.batcher import DynamicBatcher
+    from maskrcnn_amd.serving.engine import InferenceEngine
+    torch.manual_seed(0)
+    eng = InferenceEngine(shapes_config)
+    b = DynamicBatcher(eng, max_batch_size=8, batch_timeout_micros=4000)
+    H = shapes_config.IMAGE_SHAPE[0]
+    sizes = [H, H + 64, H + 128]
+    rng = np.random.RandomState(5)
+    results = {}
+
+    def call(tag, s):
+        img = (rng.rand(1, s, s, 3) * 50).astype(np.float32)
+        meta = np.zeros((1, shapes_config.IMAGE_META_SIZE), np.float32)
+        meta[:, 4:7] = (s, s, 3)
+        meta[:, 7:11] = (0, 0, s, s)
+        results[tag] = b.submit(
+            img, meta, None,
+            outputs=["mrcnn_detection", "mrcnn_mask"]).result(180)
+
+    threads = [threading.Thread(target=call, args=(f"{s}_{i}", s))
+               for i in range(2) for s in sizes]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    b.close()
+    assert len(results) == 6
+    for tag, out in results.items():
+        assert out["mrcnn_detection"].shape[0] == 1, tag
+        assert np.isfinite(out["mrcnn_detection"]).all(), tag
