@@ -531,6 +531,7 @@ std::vector<torch::Tensor> detection_prep(
 }  // namespace
 
 void register_rle(pybind11::module_& m);
+void register_unmold(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("roi_align_forward", &roi_align_forward,
@@ -549,6 +550,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("w"), py::arg("bias"),
         py::arg("up") = py::none());
   register_rle(m);
+  register_unmold(m);
   m.def("conv3x3_fwd", &conv3x3_fwd,
         "3x3 conv s1p1 + bias + optional ReLU (CDNA4 MFMA implicit GEMM)",
         py::arg("x"), py::arg("w_packed"), py::arg("bias"),

@@ -379,6 +379,17 @@ class MaskRCNN:
         # (strided copy of N full images); this is ~60 ms.
         from ..utils.image import resize as _resize
         Ho, Wo = original_image_shape[:2]
+        # Native path: one fused C++ pass (bilinear resize + threshold
+        # + paste, at::parallel_for over detections) — ~10x the
+        # threaded python loop below at 100 dets.
+        from ..ops._ext import load_ext as _load_ext
+        _ext = _load_ext()
+        if N > 0 and _ext is not None and hasattr(_ext, "unmold_masks"):
+            fm = _ext.unmold_masks(
+                torch.from_numpy(np.ascontiguousarray(masks)),
+                torch.from_numpy(np.ascontiguousarray(boxes)).long(),
+                int(Ho), int(Wo), 0.5).numpy()
+            return boxes, class_ids, scores, np.transpose(fm, (1, 2, 0))
         # [N, H, W] internally (contiguous pastes and contiguous
         # masks[:, :, i] slices for callers); returned transposed as
         # the reference's [H, W, N] convention — a view, no copy.

@@ -5,6 +5,7 @@ tested against in tests/test_ops_gpu.py.
 """
 
 import numpy as np
+import pytest
 import torch
 
 from maskrcnn_amd.ops import batched_nms, crop_and_resize, nms, roi_align
@@ -128,3 +129,56 @@ def test_nms_padded_matches_nms():
     assert n == ref.shape[0]
     np.testing.assert_array_equal(keep[:n].numpy(), ref.numpy())
     assert (keep[n:] == -1).all()
+
+
+def test_native_unmold_matches_python_path():
+    """csrc/unmold.cpp vs the threaded python resize+paste loop.
+
+    The native op fuses bilinear resize (F.interpolate align_corners
+    =False semantics) + 0.5 threshold + paste; it must agree with the
+    utils.image.resize-based fallback pixel for pixel.
+    """
+    import maskrcnn_amd.ops._ext as E
+    from maskrcnn_amd.model.mask_rcnn import MaskRCNN
+    ext = E.load_ext()
+    if ext is None or not hasattr(ext, "unmold_masks"):
+        pytest.skip("native extension not built")
+    rng = np.random.default_rng(7)
+    N = 23
+    det = np.zeros((N + 2, 6), dtype=np.float32)
+    det[:N, 0] = rng.uniform(0, 0.8, N)
+    det[:N, 1] = rng.uniform(0, 0.8, N)
+    det[:N, 2] = det[:N, 0] + rng.uniform(0.01, 0.2, N)
+    det[:N, 3] = det[:N, 1] + rng.uniform(0.01, 0.2, N)
+    det[:N, 4] = rng.integers(1, 5, N)
+    det[:N, 5] = rng.uniform(0.5, 1, N)  # rows N..N+1 zero-pad
+    mask = rng.random((N + 2, 5, 28, 28)).astype(np.float32)
+    args = (det, mask, (512, 640, 3), (512, 512, 3), (0, 0, 512, 512))
+    got = MaskRCNN.unmold_detections(*args)
+    orig = E.load_ext
+    E.load_ext = lambda *a, **k: None
+    try:
+        ref = MaskRCNN.unmold_detections(*args)
+    finally:
+        E.load_ext = orig
+    np.testing.assert_array_equal(got[0], ref[0])
+    np.testing.assert_array_equal(got[1], ref[1])
+    np.testing.assert_array_equal(got[2], ref[2])
+    mism = (got[3] != ref[3]).sum()
+    assert mism <= got[3].size * 1e-6, f"{mism} mask pixels differ"
+
+
+def test_native_unmold_empty_and_clamped():
+    from maskrcnn_amd.ops._ext import load_ext
+    ext = load_ext()
+    if ext is None or not hasattr(ext, "unmold_masks"):
+        pytest.skip("native extension not built")
+    out = ext.unmold_masks(torch.zeros(0, 28, 28),
+                           torch.zeros(0, 4, dtype=torch.long), 64, 64, 0.5)
+    assert out.shape == (0, 64, 64)
+    # Box partially outside the canvas: clamped, no crash, outside empty.
+    m = torch.ones(1, 28, 28)
+    out = ext.unmold_masks(m, torch.tensor([[-8, 50, 40, 90]]), 64, 64, 0.5)
+    assert out.shape == (1, 64, 64)
+    assert bool(out[0, :40, 50:].all())
+    assert int(out[0].sum()) == 40 * 14
