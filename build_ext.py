@@ -91,5 +91,31 @@ def build(verbose: bool = True) -> Path:
     return OUT
 
 
+def build_ingress(verbose: bool = True) -> Path:
+    """Build maskrcnn_amd/_ingress.so — the native HTTP/2 gRPC ingress.
+
+    Host-only C++ (no HIP, no torch): pybind11 + the system
+    libnghttp2.so.14 (HPACK). Linked with -l:NAME.so.14 because the
+    image ships the runtime library but not the dev symlink/headers.
+    """
+    import pybind11
+    OUT = ROOT / "maskrcnn_amd" / "_ingress.so"
+    src = CSRC / "ingress.cpp"
+    if OUT.exists() and OUT.stat().st_mtime > src.stat().st_mtime:
+        return OUT
+    py_inc = sysconfig.get_paths()["include"]
+    cmd = ["g++", "-O2", "-std=c++17", "-shared", "-fPIC",
+           str(src), "-o", str(OUT),
+           f"-I{pybind11.get_include()}", f"-I{py_inc}",
+           "-l:libnghttp2.so.14", "-lpthread",
+           "-fvisibility=hidden"]
+    if verbose:
+        print(" ".join(cmd[:6]), "...")
+    subprocess.run(cmd, check=True)
+    print(f"built {OUT}")
+    return OUT
+
+
 if __name__ == "__main__":
     build()
+    build_ingress()

@@ -197,7 +197,8 @@ class ModelServer:
                  devices=None, max_batch_size: int = 32,
                  batch_timeout_micros: int = 2000,
                  poll_interval_s: float = 5.0, config=None,
-                 model_configs=None, version_policy: str = "latest"):
+                 model_configs=None, version_policy: str = "latest",
+                 grpc_ingress: str = "python"):
         """Single model (model_name + model_base_path, TF-Serving's
         --model_name/--model_base_path mode) or several (model_configs =
         parse_model_config(models.config), the --model_config_file
@@ -224,6 +225,10 @@ class ModelServer:
         self.rest_port = rest_port
         self.metrics_port = metrics_port
         self.poll_interval_s = poll_interval_s
+        if grpc_ingress not in ("python", "cpp"):
+            raise ValueError(f"grpc_ingress must be python|cpp, "
+                             f"got {grpc_ingress!r}")
+        self.grpc_ingress = grpc_ingress
         self._stop = threading.Event()
         self._grpc_server = None
         self._rest_thread = None
@@ -340,6 +345,39 @@ class ModelServer:
                             for d in range(torch.cuda.device_count())))
         return resp
 
+    # ------------------------------------------- context-free handlers
+    # Shared by the grpc-python and native-C++ ingresses. Raise
+    # KeyError for NOT_FOUND; anything else maps to INTERNAL.
+    def _handle_metadata(self, request: GetModelMetadataRequest):
+        state = self._get_state(request.model_spec.name)
+        eng = state.engine
+        sigs = {"serving_default": build_signature(eng.config)} \
+            if eng is not None else {}
+        return GetModelMetadataResponse(
+            model_spec=ModelSpec(name=state.name, version=state.version),
+            signature_defs=sigs)
+
+    def _handle_status(self, request):
+        from .wireproto import (GetModelStatusResponse, ModelVersionStatus,
+                                STATE_AVAILABLE)
+        state = self._get_state(request.model_spec.name)
+        statuses = [ModelVersionStatus(version=v, state=STATE_AVAILABLE)
+                    for v in sorted(state.versions)] or \
+            [ModelVersionStatus(version=0, state=0)]
+        return GetModelStatusResponse(model_version_status=statuses)
+
+    def _handle_reload(self, request):
+        from .wireproto import ReloadConfigResponse, StatusProto
+        try:
+            self.reload_config(
+                [{"name": c.name, "base_path": c.base_path}
+                 for c in request.model_configs])
+            return ReloadConfigResponse(status=StatusProto(0, ""))
+        except Exception as e:  # noqa: BLE001
+            log.exception("reload config failed")
+            return ReloadConfigResponse(
+                status=StatusProto(13, str(e)))  # 13 = INTERNAL
+
     # -------------------------------------------------------------- grpc
     def _build_grpc(self):
         import grpc
@@ -347,58 +385,27 @@ class ModelServer:
 
         outer = self
 
-        def predict(request: PredictRequest, context):
-            try:
-                return outer._handle_predict(request)
-            except KeyError as e:
-                if _PROM:
-                    ERRS.labels("grpc").inc()
-                context.abort(grpc.StatusCode.NOT_FOUND, str(e))
-            except Exception as e:  # noqa: BLE001
-                if _PROM:
-                    ERRS.labels("grpc").inc()
-                log.exception("Predict failed")
-                context.abort(grpc.StatusCode.INTERNAL, str(e))
+        def _guard(fn):
+            def call(request, context):
+                try:
+                    return fn(request)
+                except KeyError as e:
+                    if _PROM:
+                        ERRS.labels("grpc").inc()
+                    context.abort(grpc.StatusCode.NOT_FOUND, str(e))
+                except Exception as e:  # noqa: BLE001
+                    if _PROM:
+                        ERRS.labels("grpc").inc()
+                    log.exception("%s failed", fn.__name__)
+                    context.abort(grpc.StatusCode.INTERNAL, str(e))
+            return call
 
-        def get_model_metadata(request: GetModelMetadataRequest, context):
-            try:
-                state = outer._get_state(request.model_spec.name)
-            except KeyError as e:
-                context.abort(grpc.StatusCode.NOT_FOUND, str(e))
-            eng = state.engine
-            sigs = {"serving_default": build_signature(eng.config)} \
-                if eng is not None else {}
-            return GetModelMetadataResponse(
-                model_spec=ModelSpec(name=state.name,
-                                     version=state.version),
-                signature_defs=sigs)
+        predict = _guard(outer._handle_predict)
+        get_model_metadata = _guard(outer._handle_metadata)
+        get_model_status = _guard(outer._handle_status)
+        handle_reload_config = _guard(outer._handle_reload)
 
-        from .wireproto import (GetModelStatusRequest,
-                                GetModelStatusResponse, ModelVersionStatus,
-                                ReloadConfigRequest, ReloadConfigResponse,
-                                StatusProto, STATE_AVAILABLE)
-
-        def get_model_status(request: GetModelStatusRequest, context):
-            try:
-                state = outer._get_state(request.model_spec.name)
-            except KeyError as e:
-                context.abort(grpc.StatusCode.NOT_FOUND, str(e))
-            statuses = [ModelVersionStatus(version=v,
-                                           state=STATE_AVAILABLE)
-                        for v in sorted(state.versions)] or \
-                [ModelVersionStatus(version=0, state=0)]
-            return GetModelStatusResponse(model_version_status=statuses)
-
-        def handle_reload_config(request: ReloadConfigRequest, context):
-            try:
-                outer.reload_config(
-                    [{"name": c.name, "base_path": c.base_path}
-                     for c in request.model_configs])
-                return ReloadConfigResponse(status=StatusProto(0, ""))
-            except Exception as e:  # noqa: BLE001
-                log.exception("reload config failed")
-                return ReloadConfigResponse(
-                    status=StatusProto(13, str(e)))  # 13 = INTERNAL
+        from .wireproto import (GetModelStatusRequest, ReloadConfigRequest)
 
         handler = grpc.method_handlers_generic_handler(
             "tensorflow.serving.PredictionService",
@@ -429,6 +436,50 @@ class ModelServer:
         bound = server.add_insecure_port(f"[::]:{self.grpc_port}")
         self.grpc_port = bound
         return server
+
+    # ------------------------------------------------------ cpp ingress
+    def _build_cpp_ingress(self):
+        """Native HTTP/2 gRPC front-end (csrc/ingress.cpp).
+
+        The TF-Serving analog of the C++ grpc daemon:
+        sockets, HTTP/2 framing, HPACK and flow control run in C++
+        threads with no GIL; Python is entered once per complete
+        request for wireproto decode -> handler -> encode.
+        """
+        from .. import _ingress
+        from .wireproto import (GetModelStatusRequest, ReloadConfigRequest)
+
+        svc = "/tensorflow.serving."
+        methods = {
+            svc + "PredictionService/Predict":
+                (PredictRequest.decode, self._handle_predict),
+            svc + "PredictionService/GetModelMetadata":
+                (GetModelMetadataRequest.decode, self._handle_metadata),
+            svc + "ModelService/GetModelStatus":
+                (GetModelStatusRequest.decode, self._handle_status),
+            svc + "ModelService/HandleReloadConfigRequest":
+                (ReloadConfigRequest.decode, self._handle_reload),
+        }
+
+        def dispatch(path, payload):
+            ent = methods.get(path)
+            if ent is None:
+                return (12, f"unknown method {path!r}", b"")
+            decode, fn = ent
+            try:
+                return (0, "", fn(decode(payload)).encode())
+            except KeyError as e:
+                if _PROM:
+                    ERRS.labels("grpc").inc()
+                return (5, str(e), b"")  # NOT_FOUND
+            except Exception as e:  # noqa: BLE001
+                if _PROM:
+                    ERRS.labels("grpc").inc()
+                log.exception("%s failed (cpp ingress)", path)
+                return (13, str(e), b"")  # INTERNAL
+
+        srv = _ingress.IngressServer(self.grpc_port, dispatch, 16)
+        return srv
 
     # -------------------------------------------------------------- rest
     def _build_rest_app(self):
@@ -584,8 +635,13 @@ class ModelServer:
                 state.warmup()
         if self.metrics_port and _PROM:
             start_http_server(self.metrics_port)
-        self._grpc_server = self._build_grpc()
-        self._grpc_server.start()
+        if self.grpc_ingress == "cpp":
+            self._grpc_server = self._build_cpp_ingress()
+            self._grpc_server.start()
+            self.grpc_port = self._grpc_server.port
+        else:
+            self._grpc_server = self._build_grpc()
+            self._grpc_server.start()
 
         if self.rest_port is not None:
             import uvicorn
@@ -617,7 +673,10 @@ class ModelServer:
     def stop(self, grace: float = 1.0):
         self._stop.set()
         if self._grpc_server is not None:
-            self._grpc_server.stop(grace).wait()
+            if self.grpc_ingress == "cpp":
+                self._grpc_server.stop()
+            else:
+                self._grpc_server.stop(grace).wait()
         if self._uvicorn_server is not None:
             self._uvicorn_server.should_exit = True
         for state in self.models.values():
@@ -626,7 +685,10 @@ class ModelServer:
                     b.close()
 
     def wait(self):
-        self._grpc_server.wait_for_termination()
+        if self.grpc_ingress == "cpp":
+            self._stop.wait()
+        else:
+            self._grpc_server.wait_for_termination()
 
 
 def main():  # pragma: no cover
@@ -647,6 +709,10 @@ def main():  # pragma: no cover
                     help="one engine per GPU: cuda:0..cuda:N-1")
     ap.add_argument("--max_batch_size", type=int, default=32)
     ap.add_argument("--batch_timeout_micros", type=int, default=2000)
+    ap.add_argument("--grpc_ingress", choices=("python", "cpp"),
+                    default="python",
+                    help="cpp = native HTTP/2 front-end "
+                         "(csrc/ingress.cpp, no GIL on the wire path)")
     args = ap.parse_args()
     logging.basicConfig(level=logging.INFO)
     devices = None
@@ -670,7 +736,8 @@ def main():  # pragma: no cover
         rest_port=args.rest_api_port or None,  # 0 = disable REST
         metrics_port=args.metrics_port,
         devices=devices, model_configs=model_configs,
-        max_batch_size=max_batch, batch_timeout_micros=timeout_us)
+        max_batch_size=max_batch, batch_timeout_micros=timeout_us,
+        grpc_ingress=args.grpc_ingress)
     server.start()
     server.wait()
 

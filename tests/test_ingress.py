@@ -1,0 +1,201 @@
+"""Native C++ gRPC ingress (csrc/ingress.cpp) against real grpc-python
+clients: HTTP/2 framing, HPACK (the client side Huffman-encodes), flow
+control with multi-MB messages both ways, grpc-status trailers, and the
+full ModelServer Predict path with grpc_ingress="cpp".
+
+Reference parity target: TF-Serving's C++ grpc ModelServer front-end
+(SURVEY.md N11 / §2.2 S2).
+"""
+
+import threading
+
+import numpy as np
+import pytest
+
+from maskrcnn_amd.serving.wireproto import (ModelSpec, PredictRequest,
+                                            PredictResponse,
+                                            numpy_to_tensor)
+
+pytest.importorskip("grpc")
+ingress = pytest.importorskip("maskrcnn_amd._ingress")
+
+
+def _raw_stub(port, method="/tensorflow.serving.PredictionService/Predict"):
+    import grpc
+    ch = grpc.insecure_channel(
+        f"127.0.0.1:{port}",
+        options=[("grpc.max_receive_message_length", 256 << 20),
+                 ("grpc.max_send_message_length", 256 << 20)])
+    return ch, ch.unary_unary(method, request_serializer=lambda b: b,
+                              response_deserializer=lambda b: b)
+
+
+def test_ingress_echo_small_and_large():
+    def handler(path, payload):
+        return (0, "", bytes(payload))
+
+    srv = ingress.IngressServer(0, handler, 4)
+    srv.start()
+    try:
+        ch, stub = _raw_stub(srv.port)
+        assert stub(b"hello", timeout=10) == b"hello"
+        # 8 MB each way: exceeds the 64 KB default stream window many
+        # times over -> exercises WINDOW_UPDATE handling both ways.
+        big = bytes(range(256)) * (8 * 1024 * 1024 // 256)
+        assert stub(big, timeout=30) == big
+        ch.close()
+    finally:
+        srv.stop()
+
+
+def test_ingress_error_status_and_unknown_method():
+    import grpc
+
+    def handler(path, payload):
+        if path.endswith("/Fail"):
+            return (5, "model not here", b"")
+        return (0, "", b"ok")
+
+    srv = ingress.IngressServer(0, handler, 2)
+    srv.start()
+    try:
+        ch, stub = _raw_stub(srv.port, "/tensorflow.serving.Svc/Fail")
+        with pytest.raises(grpc.RpcError) as e:
+            stub(b"x", timeout=10)
+        assert e.value.code() == grpc.StatusCode.NOT_FOUND
+        assert "model not here" in e.value.details()
+        ch.close()
+    finally:
+        srv.stop()
+
+
+def test_ingress_concurrent_streams():
+    """Several threads on SEPARATE channels plus several on one shared
+    channel (h2 stream multiplexing on a single connection)."""
+    def handler(path, payload):
+        return (0, "", bytes(payload[::-1]))
+
+    srv = ingress.IngressServer(0, handler, 8)
+    srv.start()
+    errs = []
+    try:
+        shared_ch, shared_stub = _raw_stub(srv.port)
+
+        def worker(i, stub):
+            try:
+                msg = bytes([i]) * (64 * 1024 + i)
+                for _ in range(5):
+                    assert stub(msg, timeout=30) == msg[::-1]
+            except Exception as e:  # noqa: BLE001
+                errs.append(e)
+
+        threads = [threading.Thread(target=worker, args=(i, shared_stub))
+                   for i in range(4)]
+        own = []
+        for i in range(4, 8):
+            ch, stub = _raw_stub(srv.port)
+            own.append(ch)
+            threads.append(threading.Thread(target=worker,
+                                            args=(i, stub)))
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        assert not errs, errs
+        shared_ch.close()
+        for ch in own:
+            ch.close()
+    finally:
+        srv.stop()
+
+
+# ---------------------------------------------------------------- server
+@pytest.fixture(scope="module")
+def cpp_served_model(tmp_path_factory):
+    from maskrcnn_amd.model.mask_rcnn import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+    from maskrcnn_amd.serving.server import ModelServer
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+
+    class SmallConfig(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 500
+        POST_NMS_ROIS_INFERENCE = 50
+        DETECTION_MIN_CONFIDENCE = 0.0
+
+    cfg = SmallConfig()
+    tmp = tmp_path_factory.mktemp("serving_cpp")
+    m = MaskRCNN("inference", cfg, model_dir=str(tmp))
+    base = str(tmp / "models" / "mask_rcnn")
+    export_saved_model(m, cfg, base, version=1)
+    server = ModelServer("mask_rcnn", base, grpc_port=0, rest_port=None,
+                         max_batch_size=4, batch_timeout_micros=1000,
+                         poll_interval_s=0.5, config=cfg,
+                         grpc_ingress="cpp")
+    server.start()
+    yield server, cfg, m
+    server.stop()
+
+
+def test_cpp_ingress_predict_matches_local(cpp_served_model):
+    server, cfg, m = cpp_served_model
+    from maskrcnn_amd.serving.client import GrpcServingClient
+    client = GrpcServingClient(f"localhost:{server.grpc_port}", cfg)
+    img = (np.random.RandomState(3).rand(96, 120, 3) * 255).astype(np.uint8)
+    r = client.detect([img])[0]
+    assert r["masks"].shape[:2] == img.shape[:2]
+    local = m.detect([img])[0]
+    np.testing.assert_allclose(r["rois"], local["rois"])
+    np.testing.assert_array_equal(r["class_ids"], local["class_ids"])
+
+
+def test_cpp_ingress_not_found_and_filter(cpp_served_model):
+    server, cfg, _ = cpp_served_model
+    import grpc
+    from maskrcnn_amd.serving.client import PREDICT_METHOD
+    channel = grpc.insecure_channel(f"localhost:{server.grpc_port}")
+    stub = channel.unary_unary(
+        PREDICT_METHOD, request_serializer=lambda r: r.encode(),
+        response_deserializer=PredictResponse.decode)
+    H = cfg.IMAGE_SHAPE[0]
+    req = PredictRequest(model_spec=ModelSpec(name="mask_rcnn"))
+    req.inputs["input_image"] = numpy_to_tensor(
+        np.zeros((1, H, H, 3), np.float32))
+    req.inputs["input_image_meta"] = numpy_to_tensor(
+        np.zeros((1, cfg.IMAGE_META_SIZE), np.float32))
+    req.output_filter = ["mrcnn_detection"]
+    resp = stub(req, timeout=60)
+    assert list(resp.outputs) == ["mrcnn_detection"]
+    assert resp.model_spec.version == 1
+
+    bad = PredictRequest(model_spec=ModelSpec(name="nope"))
+    bad.inputs["input_image"] = req.inputs["input_image"]
+    bad.inputs["input_image_meta"] = req.inputs["input_image_meta"]
+    with pytest.raises(grpc.RpcError) as e:
+        stub(bad, timeout=30)
+    assert e.value.code() == grpc.StatusCode.NOT_FOUND
+    channel.close()
+
+
+def test_cpp_ingress_metadata_and_status(cpp_served_model):
+    server, cfg, _ = cpp_served_model
+    from maskrcnn_amd.serving.client import ModelServiceClient
+    from maskrcnn_amd.serving.wireproto import STATE_AVAILABLE
+    c = ModelServiceClient(f"localhost:{server.grpc_port}")
+    states = c.get_model_status("mask_rcnn")
+    assert any(s["version"] == 1 and s["state"] == STATE_AVAILABLE
+               for s in states)
+
+    import grpc
+    from maskrcnn_amd.serving.wireproto import (GetModelMetadataRequest,
+                                                GetModelMetadataResponse)
+    ch = grpc.insecure_channel(f"localhost:{server.grpc_port}")
+    stub = ch.unary_unary(
+        "/tensorflow.serving.PredictionService/GetModelMetadata",
+        request_serializer=lambda r: r.encode(),
+        response_deserializer=GetModelMetadataResponse.decode)
+    resp = stub(GetModelMetadataRequest(
+        model_spec=ModelSpec(name="mask_rcnn"),
+        metadata_field=["signature_def"]), timeout=30)
+    assert "serving_default" in resp.signature_defs
+    ch.close()
