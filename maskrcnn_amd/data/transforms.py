@@ -318,3 +318,48 @@ class Crop(Transform):
         if mr.ndim == 2:
             mr = mr[:, :, None]
         return image, mr > 0.5
+
+
+class PiecewiseAffine(Transform):
+    """Grid-displacement warp (imgaug.PiecewiseAffine equivalent; the
+    last member of the reference's mask-safe whitelist,
+    mrcnn/model.py:~1260). A coarse nb_rows x nb_cols grid of control
+    points is jittered by N(0, scale * size) and bilinearly upsampled
+    to a dense displacement field; image sampled order-1, masks
+    order-0, through the SAME field so they stay registered."""
+
+    def __init__(self, scale=(0.01, 0.05), nb_rows=4, nb_cols=4, p=1.0):
+        self.scale = scale if isinstance(scale, tuple) else (scale, scale)
+        self.nb_rows = nb_rows
+        self.nb_cols = nb_cols
+        self.p = p
+
+    def __call__(self, image, masks, rng=None):
+        rng = rng or np.random
+        if rng.rand() >= self.p:
+            return image, masks
+        from scipy import ndimage
+        H, W = image.shape[:2]
+        s = rng.uniform(*self.scale)
+        if s <= 0:
+            return image, masks
+        dy = rng.randn(self.nb_rows, self.nb_cols) * s * H
+        dx = rng.randn(self.nb_rows, self.nb_cols) * s * W
+        zoom = (H / self.nb_rows, W / self.nb_cols)
+        dy = ndimage.zoom(dy, zoom, order=1)[:H, :W]
+        dx = ndimage.zoom(dx, zoom, order=1)[:H, :W]
+        yy, xx = np.mgrid[0:H, 0:W].astype(np.float32)
+        coords = np.stack([yy + dy, xx + dx])
+        img = np.stack([
+            ndimage.map_coordinates(image[..., ch].astype(np.float32),
+                                    coords, order=1, mode="reflect")
+            for ch in range(image.shape[2])], axis=-1)
+        image = np.clip(img, 0, 255).astype(image.dtype)
+        if masks.size:
+            m = np.stack([
+                ndimage.map_coordinates(masks[..., i].astype(np.float32),
+                                        coords, order=0, mode="constant",
+                                        cval=0.0)
+                for i in range(masks.shape[2])], axis=-1)
+            masks = m > 0.5
+        return image, masks

@@ -307,3 +307,31 @@ def test_affine_identity_is_noop():
     img2, m2 = T.Affine()(image, masks, rng)
     np.testing.assert_array_equal(m2, masks)
     assert np.abs(img2.astype(int) - image.astype(int)).max() <= 1
+
+
+def test_piecewise_affine_registration():
+    """PiecewiseAffine warps image and masks through the same
+    displacement field (last reference whitelist member)."""
+    from maskrcnn_amd.data import transforms as T
+    rng = np.random.RandomState(9)
+    H = W = 96
+    image = np.zeros((H, W, 3), np.uint8)
+    masks = np.zeros((H, W, 1), bool)
+    image[30:60, 40:70] = 200
+    masks[30:60, 40:70, 0] = True
+    aug = T.PiecewiseAffine(scale=(0.02, 0.04))
+    moved = False
+    for _ in range(5):
+        img2, m2 = aug(image, masks, rng)
+        assert img2.shape == image.shape and m2.shape == masks.shape
+        inside = img2[m2[..., 0]].mean()
+        outside = img2[~m2[..., 0]].mean()
+        assert inside > outside + 50, (inside, outside)
+        if (m2 != masks).any():
+            moved = True
+    assert moved, "warp never displaced any mask pixel"
+    # scale 0 or p=0 -> noop
+    img3, m3 = T.PiecewiseAffine(scale=(0.0, 0.0))(image, masks, rng)
+    np.testing.assert_array_equal(img3, image)
+    img4, m4 = T.PiecewiseAffine(p=0.0)(image, masks, rng)
+    np.testing.assert_array_equal(img4, image)
