@@ -199,3 +199,75 @@ def test_cpp_ingress_metadata_and_status(cpp_served_model):
         metadata_field=["signature_def"]), timeout=30)
     assert "serving_default" in resp.signature_defs
     ch.close()
+
+
+# ------------------------------------------------------------ robustness
+def _h2_frame(ftype, flags, sid, payload=b""):
+    import struct
+    n = len(payload)
+    return struct.pack(">BHBBI", n >> 16, n & 0xffff, ftype, flags,
+                       sid) + payload
+
+
+def test_ingress_survives_garbage_and_truncation():
+    """Malformed input must only kill its own connection: bad preface,
+    random bytes, truncated frames, HPACK garbage — then a VALID
+    request on a fresh channel still succeeds."""
+    import socket
+
+    def handler(path, payload):
+        return (0, "", b"alive")
+
+    srv = ingress.IngressServer(0, handler, 2)
+    srv.start()
+    try:
+        attacks = [
+            b"GET / HTTP/1.1\r\n\r\n",                    # wrong preface
+            b"PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n" + b"\xff" * 64,
+            b"PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n" +
+            _h2_frame(0x4, 0, 0) +                         # settings
+            _h2_frame(0x1, 0x4, 1, b"\xff\xfe\xfd"),       # HPACK junk
+            b"PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n" +
+            _h2_frame(0x1, 0x4, 1, b"\x88")[:11],          # truncated
+            b"PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n" +
+            b"\xff\xff\xff\x00\x00\x00\x00\x00\x01",       # 16MB+ len
+        ]
+        for a in attacks:
+            s = socket.create_connection(("127.0.0.1", srv.port),
+                                         timeout=5)
+            s.sendall(a)
+            s.settimeout(2)
+            try:
+                while s.recv(4096):
+                    pass
+            except socket.timeout:
+                pass
+            s.close()
+        # server still alive and correct
+        ch, stub = _raw_stub(srv.port)
+        assert stub(b"ping", timeout=10) == b"alive"
+        ch.close()
+    finally:
+        srv.stop()
+
+
+def test_ingress_handler_exception_maps_to_internal():
+    import grpc
+
+    def handler(path, payload):
+        raise RuntimeError("kaboom")
+
+    srv = ingress.IngressServer(0, handler, 2)
+    srv.start()
+    try:
+        ch, stub = _raw_stub(srv.port)
+        with pytest.raises(grpc.RpcError) as e:
+            stub(b"x", timeout=10)
+        assert e.value.code() == grpc.StatusCode.INTERNAL
+        assert "kaboom" in e.value.details()
+        # still serving afterwards
+        with pytest.raises(grpc.RpcError):
+            stub(b"y", timeout=10)
+        ch.close()
+    finally:
+        srv.stop()
