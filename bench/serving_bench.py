@@ -47,6 +47,11 @@ def main():
     ap.add_argument("--backbone", default="resnet101")
     ap.add_argument("--weight-dtype", default="fp32",
                     choices=["fp32", "fp8"])
+    ap.add_argument("--grpc-ingress", default="python",
+                    choices=["python", "cpp"],
+                    help="cpp = native HTTP/2 front-end "
+                         "(csrc/ingress.cpp); applies to the in-process "
+                         "server and the --server-procs replicas")
     ap.add_argument("--server-procs", type=int, default=1,
                     help="N independent ModelServer replicas on ONE "
                          "GPU (separate processes, separate ports, "
@@ -86,7 +91,7 @@ def main():
     server = ModelServer("mask_rcnn", base, grpc_port=0, rest_port=None,
                          max_batch_size=args.max_batch,
                          batch_timeout_micros=args.batch_timeout_micros,
-                         config=cfg)
+                         config=cfg, grpc_ingress=args.grpc_ingress)
     server.start(warmup=True)
     servers.append(server)
     targets.append(f"127.0.0.1:{server.grpc_port}")
@@ -105,7 +110,8 @@ def main():
                  "--enable_batching",
                  "--max_batch_size", str(args.max_batch),
                  "--batch_timeout_micros",
-                 str(args.batch_timeout_micros)],
+                 str(args.batch_timeout_micros),
+                 "--grpc_ingress", args.grpc_ingress],
                 stderr=subprocess.DEVNULL)
             extra.append(pr)
             targets.append(f"127.0.0.1:{port}")
