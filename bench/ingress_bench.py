@@ -139,12 +139,32 @@ def bench_python(workers, n, size):
         server.stop(0)
 
 
+def gil_load_thread(stop):
+    """Simulates the model-runner/batcher thread of a busy server: pure
+    Python work holding the GIL in short bursts (collation, unmold,
+    proto bookkeeping)."""
+    x = 0
+    while not stop.is_set():
+        for i in range(10000):
+            x += i * i
+        time.sleep(0.0002)  # brief release, like a torch op boundary
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--workers", type=int, default=8)
     ap.add_argument("--n", type=int, default=30)
     ap.add_argument("--size", type=int, default=1024)
+    ap.add_argument("--gil-load", action="store_true",
+                    help="run a GIL-hungry background thread in the "
+                         "server process (the busy-model scenario)")
     args = ap.parse_args()
+    stop = None
+    if args.gil_load:
+        import threading
+        stop = threading.Event()
+        threading.Thread(target=gil_load_thread, args=(stop,),
+                         daemon=True).start()
     req_mb = args.size * args.size * 3 / 1e6
     out = {"request_mb": round(req_mb, 2), "workers": args.workers,
            "n_per_worker": args.n}
@@ -157,6 +177,9 @@ def main():
               f"(p50 {p50:.1f} ms)")
     out["speedup"] = round(out["cpp"]["msgs_per_s"] /
                            out["python"]["msgs_per_s"], 2)
+    out["gil_load"] = bool(args.gil_load)
+    if stop is not None:
+        stop.set()
     print(json.dumps(out))
 
 
