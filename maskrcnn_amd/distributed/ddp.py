@@ -100,13 +100,39 @@ class DistributedTrainer:
                 p.register_post_accumulate_grad_hook(self._on_grad))
         self._reset_pending()
 
+        self._sync = True  # no_sync() flips this during grad accumulation
+
         # Broadcast initial weights + buffers from rank 0 (reference
         # ParallelModel shares one weight copy; we replicate explicitly).
+        # COALESCED: R101-FPN has ~450 state tensors; one broadcast per
+        # tensor is ~450 collective launches at init. Pack same-dtype
+        # tensors into <=64 MB flat chunks -> ~10 launches.
         with torch.no_grad():
+            chunk, chunk_bytes, dtype = [], 0, None
+            LIMIT = 64 * 1024 * 1024
+
+            def flush():
+                if not chunk:
+                    return
+                flat = torch.cat([t.reshape(-1) for t in chunk])
+                dist.broadcast(flat, src=0, group=self.group)
+                off = 0
+                for t in chunk:
+                    n = t.numel()
+                    t.copy_(flat[off:off + n].reshape(t.shape))
+                    off += n
+                chunk.clear()
+
             for t in model.state_dict().values():
-                if t.is_floating_point() or t.dtype in (
-                        torch.int32, torch.int64):
-                    dist.broadcast(t, src=0, group=self.group)
+                if not (t.is_floating_point() or t.dtype in (
+                        torch.int32, torch.int64)):
+                    continue
+                if t.dtype != dtype or chunk_bytes >= LIMIT:
+                    flush()
+                    dtype, chunk_bytes = t.dtype, 0
+                chunk.append(t)
+                chunk_bytes += t.numel() * t.element_size()
+            flush()
 
     def _reset_pending(self):
         for b in self.buckets:
@@ -114,6 +140,8 @@ class DistributedTrainer:
 
     @torch.no_grad()
     def _on_grad(self, p):
+        if not self._sync:
+            return  # no_sync(): grads accumulate locally in p.grad
         b, i = self._param_slot[p]
         off, n = b.offsets[i], b.numels[i]
         b.buffer[off:off + n].copy_(p.grad.reshape(-1).to(self.comm_dtype))
@@ -121,6 +149,25 @@ class DistributedTrainer:
         if b.pending == 0:
             b.work = dist.all_reduce(b.buffer, op=dist.ReduceOp.SUM,
                                      group=self.group, async_op=True)
+
+    def no_sync(self):
+        """Context manager: skip gradient synchronization (gradient
+        accumulation). Grads accumulate in param.grad across backwards;
+        the LAST backward runs outside no_sync() and its hooks copy the
+        fully-accumulated grads into the buckets, so exactly one
+        all-reduce happens per optimizer step (torch DDP no_sync
+        semantics; the reference's ParallelModel had no equivalent)."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def ctx():
+            prev = self._sync
+            self._sync = False
+            try:
+                yield
+            finally:
+                self._sync = prev
+        return ctx()
 
     @torch.no_grad()
     def finalize_backward(self):

@@ -131,10 +131,59 @@ def run_staged(rank, world):
     return result
 
 
+def run_accum(rank, world):
+    """no_sync() gradient accumulation: 2 local microbatches per rank,
+    one all-reduce. Verifies against an in-process replay of EVERY
+    rank's microbatches (seeds are deterministic) through a copy of the
+    broadcast model."""
+    import copy
+
+    import torch
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(100 + rank)
+    model = torch.nn.Sequential(
+        torch.nn.Linear(16, 32), torch.nn.ReLU(),
+        torch.nn.Linear(32, 4))
+    from maskrcnn_amd.distributed import DistributedTrainer
+    ddp = DistributedTrainer(model, bucket_mb=0.01)
+    ref = copy.deepcopy(model)  # post-broadcast weights
+
+    def batch(r, step):
+        g = torch.Generator().manual_seed(1000 + 10 * r + step)
+        return (torch.randn(4, 16, generator=g),
+                torch.randn(4, 4, generator=g))
+
+    # Microbatch 0 under no_sync (accumulate), microbatch 1 synced.
+    with ddp.no_sync():
+        x, y = batch(rank, 0)
+        torch.nn.functional.mse_loss(model(x), y).backward()
+    x, y = batch(rank, 1)
+    torch.nn.functional.mse_loss(model(x), y).backward()
+    ddp.finalize_backward()
+
+    # Expected: mean over ranks of (sum of that rank's microbatch grads).
+    for r in range(world):
+        for s in range(2):
+            x, y = batch(r, s)
+            torch.nn.functional.mse_loss(ref(x), y).backward()
+    ok = True
+    for p_m, p_r in zip(model.parameters(), ref.parameters()):
+        if not torch.allclose(p_m.grad, p_r.grad / world, atol=1e-5):
+            ok = False
+    result = {"rank": rank, "match": ok,
+              "grad_sum": float(sum(p.grad.abs().sum()
+                                    for p in model.parameters()))}
+    dist.destroy_process_group()
+    return result
+
+
 if __name__ == "__main__":
     mode = sys.argv[1]
     rank = int(sys.argv[2])
     world = int(sys.argv[3])
     fn = {"linear": run_linear, "maskrcnn": run_maskrcnn,
-          "staged": run_staged}[mode]
+          "staged": run_staged,
+          "accum": run_accum}[mode]
     print(json.dumps(fn(rank, world)), flush=True)

@@ -190,3 +190,14 @@ def test_bench_script_single_process(tmp_path):
                       if ln.startswith("{")][0])
     assert rec["n_gpus"] == 1 and rec["value"] > 0
     assert rec["data"] == "synthetic"
+
+
+@pytest.mark.timeout(300)
+def test_no_sync_gradient_accumulation():
+    """no_sync() accumulates locally; the synced backward reduces the
+    full accumulated grad exactly once (torch-DDP semantics)."""
+    res = _run_world("accum")
+    for r in res.values():
+        assert r["match"], res
+    assert res[0]["grad_sum"] == pytest.approx(res[1]["grad_sum"],
+                                               rel=1e-6)
