@@ -29,6 +29,7 @@
 #include <unistd.h>
 
 #include <atomic>
+#include <chrono>
 #include <condition_variable>
 #include <cstring>
 #include <deque>
@@ -230,12 +231,28 @@ class IngressServer {
       workers_.emplace_back([this] { worker_loop(); });
   }
 
-  void stop() {
+  void stop(double drain_s = 1.0) {
     if (!running_.exchange(false)) return;
-    ::shutdown(listen_fd_, SHUT_RDWR);
+    ::shutdown(listen_fd_, SHUT_RDWR);  // no new connections
     {
       std::lock_guard<std::mutex> lk(qmu_);
       qcv_.notify_all();
+    }
+    {  // Drain: give queued + in-flight requests up to drain_s to
+       // finish and their responses to flush BEFORE killing the
+       // connections (TF-Serving's graceful-shutdown behavior).
+      py::gil_scoped_release rel;  // workers need the GIL to finish
+      auto deadline = std::chrono::steady_clock::now() +
+          std::chrono::duration<double>(drain_s);
+      while (std::chrono::steady_clock::now() < deadline) {
+        bool idle;
+        {
+          std::lock_guard<std::mutex> lk(qmu_);
+          idle = queue_.empty() && in_flight_.load() == 0;
+        }
+        if (idle) break;
+        std::this_thread::sleep_for(std::chrono::milliseconds(10));
+      }
     }
     {  // kick every live connection so reader threads return
       std::lock_guard<std::mutex> lk(cmu_);
@@ -532,8 +549,10 @@ class IngressServer {
         if (!running_ && queue_.empty()) return;
         req = std::move(queue_.front());
         queue_.pop_front();
+        ++in_flight_;
       }
       serve_one(req);
+      --in_flight_;
     }
   }
 
@@ -680,6 +699,7 @@ class IngressServer {
   std::mutex qmu_;
   std::condition_variable qcv_;
   std::deque<Request> queue_;
+  std::atomic<int> in_flight_{0};
 };
 
 }  // namespace
@@ -692,6 +712,6 @@ PYBIND11_MODULE(_ingress, m) {
            py::arg("handler"), py::arg("workers") = 16)
       .def("start", &IngressServer::start,
            py::call_guard<py::gil_scoped_release>())
-      .def("stop", &IngressServer::stop)
+      .def("stop", &IngressServer::stop, py::arg("drain_s") = 1.0)
       .def_property_readonly("port", &IngressServer::port);
 }

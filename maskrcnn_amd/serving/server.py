@@ -674,7 +674,7 @@ class ModelServer:
         self._stop.set()
         if self._grpc_server is not None:
             if self.grpc_ingress == "cpp":
-                self._grpc_server.stop()
+                self._grpc_server.stop(drain_s=grace)
             else:
                 self._grpc_server.stop(grace).wait()
         if self._uvicorn_server is not None:

@@ -271,3 +271,32 @@ def test_ingress_handler_exception_maps_to_internal():
         ch.close()
     finally:
         srv.stop()
+
+
+def test_ingress_stop_drains_in_flight():
+    """stop() lets queued/in-flight requests finish and flush their
+    responses before tearing connections down (graceful shutdown)."""
+    import time
+
+    def handler(path, payload):
+        time.sleep(0.3)
+        return (0, "", b"drained")
+
+    srv = ingress.IngressServer(0, handler, 2)
+    srv.start()
+    ch, stub = _raw_stub(srv.port)
+    result = {}
+
+    def call():
+        try:
+            result["resp"] = stub(b"x", timeout=10)
+        except Exception as e:  # noqa: BLE001
+            result["err"] = e
+
+    t = threading.Thread(target=call)
+    t.start()
+    time.sleep(0.1)  # request is in flight inside the handler
+    srv.stop(drain_s=5.0)
+    t.join(timeout=10)
+    ch.close()
+    assert result.get("resp") == b"drained", result
