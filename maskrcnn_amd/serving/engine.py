@@ -19,6 +19,18 @@ from .._rocm_env import setup_rocm_perf
 from ..config import Config
 from ..utils.anchors import AnchorCache
 
+try:  # per-stage serve-loop timers (SURVEY §5 tracing row)
+    from prometheus_client import Histogram
+    _STAGE = Histogram(
+        "predict_stage_seconds", "engine predict stage wall time",
+        ["stage"],  # upload=HtoD+mold, forward=launch side,
+                    # dtoh=drain+transfer (forward is async: kernel
+                    # time lands in dtoh unless hipGraph replay syncs)
+        buckets=(.001, .0025, .005, .01, .025, .05, .1, .25, .5, 1, 2.5))
+    _PROM = True
+except ImportError:  # pragma: no cover
+    _PROM = False
+
 
 class InferenceEngine:
     def __init__(self, config: Config, state_dict=None, device=None,
@@ -153,8 +165,11 @@ class InferenceEngine:
         """
         import torch
 
+        import time as _time
+
         B, H, W, _ = input_image.shape
         with self._lock, torch.no_grad():
+            _t0 = _time.monotonic()
             # NHWC numpy IS the channels_last layout of an NCHW tensor:
             # upload as-is (pinned, async), then view-permute — no host
             # transpose and no extra device copy when CHANNELS_LAST.
@@ -190,7 +205,9 @@ class InferenceEngine:
             else:
                 anchors = self.anchor_cache.get((H, W, 3),
                                                 device=self.device)
+            _t1 = _time.monotonic()
             out = self._forward(images, metas, anchors)
+            _t2 = _time.monotonic()
             # Trim the padded batch rows GPU-side BEFORE DtoH (a padded
             # batch-32 output set is ~1 GB) and transfer only the
             # requested outputs (a full set adds ~12 MB/image of rpn/roi
@@ -205,6 +222,19 @@ class InferenceEngine:
                     # Torch [B,D,C,mh,mw] -> TF layout [B,D,mh,mw,C].
                     t = t.permute(0, 1, 3, 4, 2).contiguous()
                 result[name] = t.cpu().numpy()
+            _t3 = _time.monotonic()
+            # upload = HtoD + GPU mold; forward = launch side (async);
+            # dtoh = execution drain + transfer. Kept as attributes for
+            # tests/tools and exported to Prometheus when available.
+            self.last_stage_ms = {
+                "upload": (_t1 - _t0) * 1e3,
+                "forward": (_t2 - _t1) * 1e3,
+                "dtoh": (_t3 - _t2) * 1e3,
+            }
+            if _PROM:
+                _STAGE.labels("upload").observe(_t1 - _t0)
+                _STAGE.labels("forward").observe(_t2 - _t1)
+                _STAGE.labels("dtoh").observe(_t3 - _t2)
             return result
 
     @classmethod

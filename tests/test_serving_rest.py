@@ -202,3 +202,29 @@ def test_rest_prometheus_metrics_path(rest_server):
     assert r.status_code == 200
     assert b"predict_requests_total" in r.content or \
         b"batch_queue_wait" in r.content
+
+
+def test_engine_stage_timers(shapes_config):
+    """engine.predict records per-stage wall times (SURVEY §5 tracing:
+    per-stage CPU timers in the serve loop) and exports them to the
+    prometheus registry when available."""
+    import numpy as np
+    from maskrcnn_amd.serving.engine import InferenceEngine
+
+    cfg = shapes_config
+    eng = InferenceEngine(cfg, device="cpu", use_bf16=False)
+    H = cfg.IMAGE_SHAPE[0]
+    img = np.zeros((1, H, H, 3), np.float32)
+    meta = np.zeros((1, cfg.IMAGE_META_SIZE), np.float32)
+    eng.predict(img, meta, outputs=["mrcnn_detection"])
+    st = eng.last_stage_ms
+    assert set(st) == {"upload", "forward", "dtoh"}
+    assert all(v >= 0 for v in st.values())
+    assert st["forward"] > 0
+    try:
+        from prometheus_client import REGISTRY
+        sample = REGISTRY.get_sample_value(
+            "predict_stage_seconds_count", {"stage": "forward"})
+        assert sample and sample >= 1
+    except ImportError:
+        pass
