@@ -362,7 +362,7 @@ class IngressServer {
           conn->write_frame(F_SETTINGS, FL_END_STREAM, 0, nullptr, 0);
           break;
         case F_PING:
-          if (!(flags & FL_END_STREAM))
+          if (!(flags & FL_END_STREAM) && len == 8)
             conn->write_frame(F_PING, FL_END_STREAM, 0, p, len);
           break;
         case F_WINDOW_UPDATE: {
@@ -394,7 +394,30 @@ class IngressServer {
             if (off + 5 > end) return;
             off += 5;
           }
-          auto st = std::make_shared<Stream>();
+          std::shared_ptr<Stream> st;
+          {
+            std::lock_guard<std::mutex> lk(conn->mu);
+            auto it = conn->streams.find(sid);
+            if (it != conn->streams.end() && it->second->headers_done)
+              st = it->second;  // trailing HEADERS on an open stream
+          }
+          if (st) {
+            // Request trailers: MUST still run through the HPACK
+            // inflater (the dynamic table is connection-wide state;
+            // skipping a header block would desync every later
+            // request), then half-close without clobbering the body.
+            // Trailers carry no :path, so finish_headers leaves
+            // st->path intact and dispatches on end_stream as usual.
+            st->header_block.assign(p + off, p + end);
+            st->end_stream |= bool(flags & FL_END_STREAM);
+            if (flags & FL_END_HEADERS) {
+              if (!finish_headers(conn, sid, st)) return;
+            } else {
+              continuation_sid = sid;
+            }
+            break;
+          }
+          st = std::make_shared<Stream>();
           st->send_window = conn->peer_initial_window;
           st->end_stream = flags & FL_END_STREAM;
           st->header_block.assign(p + off, p + end);

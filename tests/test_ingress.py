@@ -356,3 +356,61 @@ def test_ingress_random_frame_fuzz():
         ch.close()
     finally:
         srv.stop()
+
+
+def test_ingress_request_trailers_half_close():
+    """A client that half-closes via trailing HEADERS (instead of
+    END_STREAM on DATA) still gets served, and the trailer block keeps
+    the connection's HPACK table in sync."""
+    import socket
+    import struct
+
+    def handler(path, payload):
+        return (0, "", b"trailer-ok:" + bytes(payload))
+
+    srv = ingress.IngressServer(0, handler, 2)
+    srv.start()
+    try:
+        s = socket.create_connection(("127.0.0.1", srv.port), timeout=5)
+        s.sendall(b"PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n" +
+                  struct.pack(">BHBBI", 0, 0, 0x4, 0, 0))  # SETTINGS
+
+        def lit(name, value):
+            return (b"\x00" + bytes([len(name)]) + name +
+                    bytes([len(value)]) + value)
+
+        hdrs = (b"\x83"  # :method POST (static idx 3)
+                + lit(b":path", b"/svc/M") + lit(b":scheme", b"http")
+                + lit(b"content-type", b"application/grpc"))
+        s.sendall(struct.pack(">BHBBI", 0, len(hdrs), 0x1, 0x4, 1)
+                  + hdrs)                       # HEADERS, END_HEADERS
+        body = b"\x00" + struct.pack(">I", 3) + b"abc"
+        s.sendall(struct.pack(">BHBBI", 0, len(body), 0x0, 0x0, 1)
+                  + body)                       # DATA, no END_STREAM
+        trl = lit(b"grpc-client-note", b"bye")
+        s.sendall(struct.pack(">BHBBI", 0, len(trl), 0x1, 0x5, 1)
+                  + trl)                        # trailers: EH+ES
+        # Read frames until the response DATA arrives.
+        s.settimeout(5)
+        buf = b""
+        deadline = 50
+        payload = None
+        while deadline and payload is None:
+            deadline -= 1
+            chunk = s.recv(65536)
+            if not chunk:
+                break
+            buf += chunk
+            while len(buf) >= 9:
+                n = (buf[0] << 16) | (buf[1] << 8) | buf[2]
+                if len(buf) < 9 + n:
+                    break
+                ftype = buf[3]
+                frame = buf[9:9 + n]
+                buf = buf[9 + n:]
+                if ftype == 0x0:  # DATA
+                    payload = frame[5:]
+        assert payload == b"trailer-ok:abc", payload
+        s.close()
+    finally:
+        srv.stop()
