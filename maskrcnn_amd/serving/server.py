@@ -310,25 +310,17 @@ class ModelServer:
         # instead of the full [B, D, 28, 28, C] tensor (81x fewer bytes
         # at 81 COCO classes; the full mrcnn_mask stays available for
         # stock TF-Serving clients).
+        # mrcnn_mask_top is resolved INSIDE the engine (GPU-side class
+        # gather before DtoH — 81x fewer transfer bytes); the engine
+        # needs mrcnn_detection listed so clients that filtered it out
+        # still work, and it ignores unknown names.
         want = list(request.output_filter) or None
-        mask_top = bool(want) and "mrcnn_mask_top" in want
-        if mask_top:
-            want = [w for w in want if w != "mrcnn_mask_top"]
-            for need in ("mrcnn_mask", "mrcnn_detection"):
-                if need not in want:
-                    want.append(need)
+        if want and "mrcnn_mask_top" in want and \
+                "mrcnn_detection" not in want:
+            want = want + ["mrcnn_detection"]
         outputs, version = state.predict(
             images, metas, anchors, outputs=want,
             version=want_version)
-        if mask_top:
-            det = outputs["mrcnn_detection"]          # [B, D, 6]
-            full = outputs["mrcnn_mask"]              # [B, D, mh, mw, C]
-            cls = det[:, :, 4].astype(np.int64)       # [B, D]
-            B, D = cls.shape
-            bb = np.arange(B)[:, None]
-            dd = np.arange(D)[None, :]
-            outputs["mrcnn_mask_top"] = \
-                full[bb, dd, :, :, cls]               # [B, D, mh, mw]
         if request.output_filter:
             outputs = {k: v for k, v in outputs.items()
                        if k in request.output_filter}
