@@ -137,21 +137,26 @@ class DynamicBatcher:
         if not batch:
             return
         try:
-            images = np.concatenate([b[0] for b in batch], axis=0)
+            # Images are passed as the REQUEST LIST: the engine
+            # collates them straight into its pinned staging buffer
+            # (one memcpy pass; a host-side np.concatenate of 32 x
+            # 1024² requests was an extra ~100 MB / ~20 ms per batch
+            # under the GIL).
+            image_parts = [b[0] for b in batch]
             metas = np.concatenate([b[1] for b in batch], axis=0)
             anchors = batch[0][2]
             # Union of the batch's output filters (None = everything).
             outputs = None
             if all(b[3] is not None for b in batch):
                 outputs = sorted({n for b in batch for n in b[3]})
-            real_n = images.shape[0]
+            real_n = sum(p.shape[0] for p in image_parts)
+            padded_n = real_n
             if self.pad_to_max and real_n < self.max_batch_size:
                 pad = self.max_batch_size - real_n
-                images = np.concatenate(
-                    [images, np.zeros((pad,) + images.shape[1:],
-                                      images.dtype)], axis=0)
+                padded_n = self.max_batch_size
                 # Padded metas replicate row 0 (real) so shape fields
-                # parsed from the meta stay valid.
+                # parsed from the meta stay valid. Image pad rows are
+                # zero-filled by the engine's collate.
                 metas = np.concatenate(
                     [metas, np.repeat(metas[:1], pad, axis=0)], axis=0)
             t0 = time.monotonic()
@@ -159,16 +164,17 @@ class DynamicBatcher:
                 for *_r, fut in batch:
                     QUEUE_WAIT.observe(
                         t0 - getattr(fut, "_enqueued_at", t0))
-            out = self.engine.predict(images, metas, anchors,
-                                      trim_to=real_n, outputs=outputs)
+            out = self.engine.predict(
+                image_parts, metas, anchors, trim_to=real_n,
+                outputs=outputs, pad_to=padded_n)
             dt = time.monotonic() - t0
             if _PROM:
                 INFER_TIME.observe(dt)
             log.info("batch of %d (padded %d) in %.3fs", real_n,
-                     images.shape[0], dt)
+                     padded_n, dt)
             self.stats["batches"] += 1
             self.stats["requests"] += len(batch)
-            self.stats["batch_size_sum"] += images.shape[0]
+            self.stats["batch_size_sum"] += padded_n
             off = 0
             for img, _, _, _, fut in batch:
                 k = img.shape[0]

@@ -86,6 +86,43 @@ class InferenceEngine:
         buf.copy_(t)
         return buf.to(self.device, non_blocking=True)
 
+    def _upload_batch(self, parts, total: int):
+        """Collate a list of [bi, H, W, C] arrays (same dtype/shape
+        tail) into ONE pinned [total, H, W, C] buffer — each request's
+        bytes are copied exactly once, and rows past the real batch
+        (pad-to-max) are zero-filled."""
+        import torch
+        if len(parts) == 1 and parts[0].shape[0] == total:
+            return self._upload(parts[0])
+        tail = parts[0].shape[1:]
+        np_dtype = parts[0].dtype
+        if self.device.type != "cuda":
+            outn = np.zeros((total,) + tail, np_dtype)
+            off = 0
+            for p in parts:
+                outn[off:off + p.shape[0]] = p
+                off += p.shape[0]
+            return torch.from_numpy(outn)
+        ref = torch.from_numpy(
+            np.empty((0,) + tail, np_dtype))  # dtype carrier
+        key = ((total,) + tail, ref.dtype)
+        buf = self._pinned.get(key)
+        if buf is None:
+            if len(self._pinned) >= 8:
+                self._pinned.pop(next(iter(self._pinned)))
+            buf = torch.empty((total,) + tail, dtype=ref.dtype,
+                              pin_memory=True)
+            self._pinned[key] = buf
+        off = 0
+        for p in parts:
+            n = p.shape[0]
+            buf[off:off + n].copy_(
+                torch.from_numpy(np.ascontiguousarray(p)))
+            off += n
+        if off < total:
+            buf[off:].zero_()  # pinned buffers are reused: clear pad
+        return buf.to(self.device, non_blocking=True)
+
     #: the 7 signature outputs (S1 order)
     OUTPUT_NAMES = ("mrcnn_detection", "mrcnn_class", "mrcnn_bbox",
                     "mrcnn_mask", "rois", "rpn_class", "rpn_bbox")
@@ -151,15 +188,21 @@ class InferenceEngine:
         self._graphs[(tuple(images.shape), tuple(anchors.shape))] = st
         return st
 
-    def predict(self, input_image: np.ndarray,
+    def predict(self, input_image,
                 input_image_meta: np.ndarray,
                 input_anchors: Optional[np.ndarray] = None,
                 trim_to: Optional[int] = None,
-                outputs: Optional[list] = None
+                outputs: Optional[list] = None,
+                pad_to: Optional[int] = None
                 ) -> Dict[str, np.ndarray]:
         """Run the Predict signature.
 
-        input_image: [B, H, W, 3] molded float32 (NHWC, TF convention).
+        input_image: [B, H, W, 3] molded float32 (NHWC, TF convention)
+        — or a LIST of such arrays (the batcher's request list): parts
+        are collated directly into the pinned staging buffer, skipping
+        the host-side np.concatenate (~100 MB / ~20 ms per uint8
+        batch-32 under the GIL). pad_to zero-fills trailing rows up to
+        the target batch (pad-to-max batching).
         Returns the 7 named outputs as numpy (TF output layout:
         mrcnn_mask as [B, D, 28, 28, C]).
         """
@@ -167,7 +210,12 @@ class InferenceEngine:
 
         import time as _time
 
-        B, H, W, _ = input_image.shape
+        parts = (list(input_image)
+                 if isinstance(input_image, (list, tuple))
+                 else [input_image])
+        _, H, W, _c = parts[0].shape
+        B = pad_to if pad_to is not None \
+            else sum(p.shape[0] for p in parts)
         with self._lock, torch.no_grad():
             _t0 = _time.monotonic()
             # NHWC numpy IS the channels_last layout of an NCHW tensor:
@@ -177,8 +225,8 @@ class InferenceEngine:
             # uint8 extension: an un-molded uint8 image uploads at 1/4
             # the bytes (wire AND PCIe); mold (float - MEAN_PIXEL) runs
             # on the GPU. fp32 input stays the stock molded contract.
-            if input_image.dtype == np.uint8:
-                up = self._upload(input_image)
+            if parts[0].dtype == np.uint8:
+                up = self._upload_batch(parts, B)
                 mean = getattr(self, "_mean_pixel_t", None)
                 if mean is None or mean.device != self.device:
                     mean = torch.tensor(
@@ -187,8 +235,9 @@ class InferenceEngine:
                     self._mean_pixel_t = mean
                 up = up.float() - mean.view(1, 1, 1, 3)
             else:
-                up = self._upload(
-                    input_image.astype(np.float32, copy=False))
+                up = self._upload_batch(
+                    [p.astype(np.float32, copy=False) for p in parts],
+                    B)
             images = up.permute(0, 3, 1, 2)
             if self.config.CHANNELS_LAST and self.device.type == "cuda":
                 images = images.contiguous(

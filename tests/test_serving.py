@@ -183,7 +183,16 @@ class _FakeEngine:
         self.calls = []
 
     def predict(self, images, metas, anchors=None, trim_to=None,
-                outputs=None):
+                outputs=None, pad_to=None):
+        # Batcher contract: images is the request LIST (engine-side
+        # collate); pad_to is the padded batch target.
+        if isinstance(images, (list, tuple)):
+            images = np.concatenate(images, axis=0)
+        if pad_to is not None and images.shape[0] < pad_to:
+            pad = pad_to - images.shape[0]
+            images = np.concatenate(
+                [images, np.zeros((pad,) + images.shape[1:],
+                                  images.dtype)], axis=0)
         self.calls.append(images.shape[0])
         time.sleep(0.01)
         out = images.sum(axis=(1, 2, 3))
