@@ -54,8 +54,11 @@ def build(verbose: bool = True) -> Path:
         common += ["-fsanitize=address", "-shared-libasan", "-g"]
     incs = [f"-I{p}" for p in inc] + [f"-I{py_inc}", f"-I{CSRC}"]
 
+    # ingress.cpp is a SEPARATE host-only product (_ingress.so via
+    # build_ingress — needs -lnghttp2); sweeping it into _C.so leaves
+    # an undefined nghttp2 symbol that breaks the core import.
     sources = sorted(CSRC.glob("hip/*.hip")) + \
-        sorted(CSRC.glob("*.cpp"))
+        sorted(p for p in CSRC.glob("*.cpp") if p.name != "ingress.cpp")
     objs = []
 
     def compile_one(src: Path) -> Path:
