@@ -1070,3 +1070,66 @@ def test_batcher_mixed_shapes_concurrent(shapes_config):
     for tag, out in results.items():
         assert out["mrcnn_detection"].shape[0] == 1, tag
         assert np.isfinite(out["mrcnn_detection"]).all(), tag
+
+
+def test_server_cli_cpp_ingress(tmp_path):
+    """python -m maskrcnn_amd.serving.server --grpc_ingress cpp: boots,
+    answers a real gRPC Predict through the native daemon, dies clean."""
+    import signal
+    import socket
+    import subprocess
+    import sys
+
+    pytest.importorskip("maskrcnn_amd._ingress")
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+
+    class TinyConfig(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 100
+        POST_NMS_ROIS_INFERENCE = 10
+        DETECTION_MIN_CONFIDENCE = 0.0
+
+    cfg = TinyConfig()
+    m = MaskRCNN("inference", cfg, model_dir=str(tmp_path))
+    base = str(tmp_path / "models" / "cli_cpp")
+    export_saved_model(m, cfg, base, version=1)
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    gport = s.getsockname()[1]
+    s.close()
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "maskrcnn_amd.serving.server",
+         "--model_name", "cli_cpp", "--model_base_path", base,
+         "--port", str(gport), "--rest_api_port", "0",
+         "--enable_batching", "--max_batch_size", "2",
+         "--batch_timeout_micros", "1000",
+         "--grpc_ingress", "cpp"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+    try:
+        from maskrcnn_amd.serving.client import GrpcServingClient
+        img = (np.random.RandomState(5).rand(96, 96, 3) * 255) \
+            .astype(np.uint8)
+        r = None
+        deadline = time.time() + 120
+        while time.time() < deadline:
+            if proc.poll() is not None:
+                raise AssertionError(proc.stdout.read()[-2000:])
+            try:
+                client = GrpcServingClient(f"127.0.0.1:{gport}", cfg,
+                                           model_name="cli_cpp",
+                                           timeout=10.0)
+                r = client.detect([img])[0]
+                break
+            except Exception:
+                time.sleep(0.5)
+        assert r is not None, "no Predict answer through cpp ingress"
+        assert r["masks"].shape[:2] == img.shape[:2]
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=20)
+        except subprocess.TimeoutExpired:
+            proc.kill()
