@@ -35,6 +35,11 @@ except ImportError:  # pragma: no cover
     _PROM = False
 
 
+class BatcherQueueFull(RuntimeError):
+    """max_enqueued_batches exceeded — TF-Serving sheds load here
+    (RESOURCE_EXHAUSTED) rather than buffering unboundedly."""
+
+
 class DynamicBatcher:
     def __init__(self, engine, max_batch_size: int = 32,
                  batch_timeout_micros: int = 2000,
@@ -67,13 +72,21 @@ class DynamicBatcher:
         """Enqueue one request (batch dim must be 1..max_batch_size).
         `outputs` restricts which signature outputs are computed back to
         the host (None = all 7). Returns a Future resolving to the
-        outputs dict (numpy)."""
+        outputs dict (numpy). Raises BatcherQueueFull when
+        max_enqueued_batches worth of requests are already waiting
+        (TF-Serving load-shedding semantics — the RPC maps it to
+        RESOURCE_EXHAUSTED instead of buffering unboundedly)."""
         if self._stop:
             raise RuntimeError("batcher closed")
         fut: Future = Future()
         fut._enqueued_at = time.monotonic()
-        self.queue.put((input_image, input_image_meta, input_anchors,
-                        outputs, fut))
+        try:
+            self.queue.put_nowait((input_image, input_image_meta,
+                                   input_anchors, outputs, fut))
+        except queue.Full:
+            raise BatcherQueueFull(
+                f"batching queue full ({self.queue.maxsize} requests "
+                f"waiting); retry later")
         return fut
 
     def close(self):

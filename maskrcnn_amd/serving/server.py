@@ -166,7 +166,10 @@ class ModelState:
             try:
                 fut = batcher.submit(images, metas, anchors,
                                      outputs=outputs)
-            except RuntimeError:
+            except RuntimeError as e:
+                from .batcher import BatcherQueueFull
+                if isinstance(e, BatcherQueueFull):
+                    raise  # load shedding, not a hot-swap race
                 continue
             try:
                 return fut.result(timeout=300.0), version
@@ -377,6 +380,8 @@ class ModelServer:
 
         outer = self
 
+        from .batcher import BatcherQueueFull
+
         def _guard(fn):
             def call(request, context):
                 try:
@@ -385,6 +390,11 @@ class ModelServer:
                     if _PROM:
                         ERRS.labels("grpc").inc()
                     context.abort(grpc.StatusCode.NOT_FOUND, str(e))
+                except BatcherQueueFull as e:
+                    if _PROM:
+                        ERRS.labels("grpc").inc()
+                    context.abort(grpc.StatusCode.RESOURCE_EXHAUSTED,
+                                  str(e))
                 except Exception as e:  # noqa: BLE001
                     if _PROM:
                         ERRS.labels("grpc").inc()
@@ -453,6 +463,8 @@ class ModelServer:
                 (ReloadConfigRequest.decode, self._handle_reload),
         }
 
+        from .batcher import BatcherQueueFull
+
         def dispatch(path, payload):
             ent = methods.get(path)
             if ent is None:
@@ -464,6 +476,10 @@ class ModelServer:
                 if _PROM:
                     ERRS.labels("grpc").inc()
                 return (5, str(e), b"")  # NOT_FOUND
+            except BatcherQueueFull as e:
+                if _PROM:
+                    ERRS.labels("grpc").inc()
+                return (8, str(e), b"")  # RESOURCE_EXHAUSTED
             except Exception as e:  # noqa: BLE001
                 if _PROM:
                     ERRS.labels("grpc").inc()
@@ -609,6 +625,10 @@ class ModelServer:
             except Exception as e:  # noqa: BLE001
                 if _PROM:
                     ERRS.labels("rest").inc()
+                from .batcher import BatcherQueueFull
+                if isinstance(e, BatcherQueueFull):
+                    return JSONResponse({"error": str(e)},
+                                        status_code=429)
                 log.exception("REST predict failed")
                 return JSONResponse({"error": str(e)}, status_code=500)
 

@@ -1133,3 +1133,31 @@ def test_server_cli_cpp_ingress(tmp_path):
             proc.wait(timeout=20)
         except subprocess.TimeoutExpired:
             proc.kill()
+
+
+def test_batcher_load_shedding():
+    """max_enqueued_batches full -> BatcherQueueFull (TF-Serving
+    RESOURCE_EXHAUSTED semantics), not unbounded buffering."""
+    from maskrcnn_amd.serving.batcher import (BatcherQueueFull,
+                                              DynamicBatcher)
+
+    class Slow:
+        def predict(self, images, metas, anchors=None, trim_to=None,
+                    outputs=None, pad_to=None):
+            time.sleep(0.5)
+            n = trim_to if trim_to is not None else 1
+            return {"out": np.zeros(n)}
+
+    b = DynamicBatcher(Slow(), max_batch_size=1,
+                       batch_timeout_micros=100,
+                       max_enqueued_batches=2)
+    img = np.zeros((1, 8, 8, 3), np.float32)
+    meta = np.zeros((1, 5), np.float32)
+    futs = [b.submit(img, meta)]  # taken by the worker quickly
+    with pytest.raises(BatcherQueueFull):
+        for _ in range(8):  # maxsize=2 -> must trip well before 8
+            futs.append(b.submit(img, meta))
+            time.sleep(0.01)
+    for f in futs:
+        f.result(timeout=30)
+    b.close()
