@@ -113,6 +113,16 @@ def parse_model_config(path: str) -> List[Dict]:
         mvp = c.get("model_version_policy")
         if isinstance(mvp, dict) and "all" in mvp:
             entry["version_policy"] = "all"
+        # version_labels { key: 'stable' value: 2 } (repeated) — the
+        # TF-Serving canary/rollback mechanism: clients address
+        # ModelSpec.version_label instead of a number.
+        vl = c.get("version_labels")
+        if vl:
+            if isinstance(vl, dict):
+                vl = [vl]
+            entry["version_labels"] = {
+                str(e["key"]): int(e["value"]) for e in vl
+                if "key" in e and "value" in e}
         out.append(entry)
     return out
 

@@ -58,12 +58,17 @@ class ModelState:
     def __init__(self, name: str, base_path: str, device=None,
                  devices=None, max_batch_size: int = 32,
                  batch_timeout_micros: int = 2000, config=None,
-                 version_policy: str = "latest"):
+                 version_policy: str = "latest", version_labels=None):
         """version_policy: 'latest' (TF-Serving default — new version
         replaces old) or 'all' (every discovered version stays
-        servable; requests pin via model_spec.version)."""
+        servable; requests pin via model_spec.version).
+        version_labels: {'stable': 1, 'canary': 2} — clients address
+        ModelSpec.version_label (the TF-Serving canary mechanism;
+        implies the labeled versions must be loaded, so usually
+        combined with version_policy='all')."""
         assert version_policy in ("latest", "all"), version_policy
         self.name = name
+        self.version_labels = dict(version_labels or {})
         self.base_path = base_path
         self.devices = list(devices) if devices else [device]
         self.max_batch_size = max_batch_size
@@ -221,8 +226,8 @@ class ModelServer:
             mc["name"]: ModelState(
                 mc["name"], mc["base_path"],
                 **{**self._model_kwargs,
-                   **({"version_policy": mc["version_policy"]}
-                      if "version_policy" in mc else {})})
+                   **{k: mc[k] for k in ("version_policy",
+                                         "version_labels") if k in mc}})
             for mc in model_configs}
         self.grpc_port = grpc_port
         self.rest_port = rest_port
@@ -286,6 +291,14 @@ class ModelServer:
         if sig != "serving_default":
             raise KeyError(f"unknown signature {sig!r}")
         want_version = request.model_spec.version
+        label = getattr(request.model_spec, "version_label", "")
+        if want_version is None and label:
+            # TF-Serving version_labels routing ('stable'/'canary').
+            if label not in state.version_labels:
+                raise KeyError(
+                    f"model {name!r} has no version label {label!r} "
+                    f"(labels: {sorted(state.version_labels)})")
+            want_version = state.version_labels[label]
         if want_version is not None and want_version not in state.versions:
             raise KeyError(
                 f"model {name!r} version {want_version} not loaded "
@@ -568,6 +581,7 @@ class ModelServer:
                                     status_code=404)
             name = model_and_verb[:-len(":predict")]
             want_version = None
+            want_label = None
             if "/versions/" in name:   # /v1/models/{m}/versions/{v}:predict
                 name, _, v = name.partition("/versions/")
                 try:
@@ -575,10 +589,18 @@ class ModelServer:
                 except ValueError:
                     return JSONResponse(
                         {"error": f"bad version {v!r}"}, status_code=400)
+            elif "/labels/" in name:   # /v1/models/{m}/labels/{l}:predict
+                name, _, want_label = name.partition("/labels/")
             state = outer.models.get(name)
             if state is None:
                 return JSONResponse(
                     {"error": f"model {name} not found"}, status_code=404)
+            if want_label is not None:
+                if want_label not in state.version_labels:
+                    return JSONResponse(
+                        {"error": f"model {name} has no version label "
+                                  f"{want_label!r}"}, status_code=404)
+                want_version = state.version_labels[want_label]
             if want_version is not None and \
                     want_version not in state.versions:
                 return JSONResponse(

@@ -1161,3 +1161,76 @@ def test_batcher_load_shedding():
     for f in futs:
         f.result(timeout=30)
     b.close()
+
+
+def test_version_labels_routing(tmp_path):
+    """TF-Serving version_labels: ModelSpec.version_label routes grpc
+    Predicts, /v1/models/{m}/labels/{l}:predict routes REST, and the
+    models.config `version_labels {key value}` entries parse."""
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+    from maskrcnn_amd.serving.server import ModelServer
+
+    class TinyConfig(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 100
+        POST_NMS_ROIS_INFERENCE = 10
+        DETECTION_MIN_CONFIDENCE = 0.0
+
+    cfg = TinyConfig()
+    m = MaskRCNN("inference", cfg, model_dir=str(tmp_path))
+    base = str(tmp_path / "models" / "vl")
+    export_saved_model(m, cfg, base, version=1)
+    export_saved_model(m, cfg, base, version=2)
+    server = ModelServer(
+        model_configs=[{"name": "vl", "base_path": base,
+                        "version_policy": "all",
+                        "version_labels": {"stable": 1, "canary": 2}}],
+        grpc_port=0, rest_port=None, max_batch_size=1,
+        batch_timeout_micros=100, config=cfg)
+    server.start()
+    try:
+        import grpc
+        from maskrcnn_amd.serving.client import PREDICT_METHOD
+        ch = grpc.insecure_channel(f"localhost:{server.grpc_port}")
+        stub = ch.unary_unary(
+            PREDICT_METHOD, request_serializer=lambda r: r.encode(),
+            response_deserializer=PredictResponse.decode)
+        H = cfg.IMAGE_SHAPE[0]
+        meta = np.zeros((1, cfg.IMAGE_META_SIZE), np.float32)
+        meta[:, 4:7] = (H, H, 3)
+        meta[:, 7:11] = (0, 0, H, H)
+
+        def req(label):
+            r = PredictRequest(model_spec=ModelSpec(
+                name="vl", version_label=label))
+            r.inputs["input_image"] = numpy_to_tensor(
+                np.zeros((1, H, H, 3), np.float32))
+            r.inputs["input_image_meta"] = numpy_to_tensor(meta)
+            r.output_filter = ["mrcnn_detection"]
+            return r
+
+        assert stub(req("stable"), timeout=60).model_spec.version == 1
+        assert stub(req("canary"), timeout=60).model_spec.version == 2
+        with pytest.raises(grpc.RpcError) as e:
+            stub(req("nope"), timeout=30)
+        assert e.value.code() == grpc.StatusCode.NOT_FOUND
+        ch.close()
+    finally:
+        server.stop()
+
+
+def test_model_config_version_labels_parse(tmp_path):
+    from maskrcnn_amd.serving.config_file import parse_model_config
+    mc = tmp_path / "models.config"
+    mc.write_text(
+        "model_config_list { config {\n"
+        "  name: 'm' base_path: '/x'\n"
+        "  model_version_policy { all {} }\n"
+        "  version_labels { key: 'stable' value: 1 }\n"
+        "  version_labels { key: 'canary' value: 2 }\n"
+        "} }\n")
+    cfgs = parse_model_config(str(mc))
+    assert cfgs[0]["version_policy"] == "all"
+    assert cfgs[0]["version_labels"] == {"stable": 1, "canary": 2}

@@ -228,3 +228,52 @@ def test_engine_stage_timers(shapes_config):
         assert sample and sample >= 1
     except ImportError:
         pass
+
+
+def test_rest_version_labels_route(tmp_path):
+    """/v1/models/{m}/labels/{l}:predict (TF-Serving REST label
+    addressing) — exercised in-process via the FastAPI TestClient."""
+    from fastapi.testclient import TestClient
+
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+    from maskrcnn_amd.serving.server import ModelServer
+
+    class TinyConfig(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 100
+        POST_NMS_ROIS_INFERENCE = 10
+        DETECTION_MIN_CONFIDENCE = 0.0
+
+    cfg = TinyConfig()
+    m = MaskRCNN("inference", cfg, model_dir=str(tmp_path))
+    base = str(tmp_path / "models" / "vlr")
+    export_saved_model(m, cfg, base, version=1)
+    export_saved_model(m, cfg, base, version=2)
+    server = ModelServer(
+        model_configs=[{"name": "vlr", "base_path": base,
+                        "version_policy": "all",
+                        "version_labels": {"stable": 1}}],
+        grpc_port=0, rest_port=None, max_batch_size=1,
+        batch_timeout_micros=100, config=cfg)
+    for st in server.models.values():
+        st.maybe_reload()
+    app = server._build_rest_app()
+    client = TestClient(app)
+    H = cfg.IMAGE_SHAPE[0]
+    meta = np.zeros((1, cfg.IMAGE_META_SIZE), np.float32)
+    meta[:, 4:7] = (H, H, 3)
+    meta[:, 7:11] = (0, 0, H, H)
+    body = {"inputs": {
+        "input_image": np.zeros((1, H, H, 3), np.float32).tolist(),
+        "input_image_meta": meta.tolist()}}
+    r = client.post("/v1/models/vlr/labels/stable:predict", json=body)
+    assert r.status_code == 200, r.text
+    assert "mrcnn_detection" in r.json()["outputs"]
+    r = client.post("/v1/models/vlr/labels/nope:predict", json=body)
+    assert r.status_code == 404
+    for st in server.models.values():
+        for slot in st.versions.values():
+            for b in slot["batchers"]:
+                b.close()
