@@ -63,12 +63,16 @@ def build_signature(config) -> SignatureDef:
 
 def export_saved_model(model, config, base_dir: str, version: int = 1,
                        extra_meta: dict = None,
-                       weight_dtype: str = "fp32") -> str:
+                       weight_dtype: str = "fp32",
+                       warmup_batches=None) -> str:
     """Export a (torch) Mask R-CNN model as a SavedModel-layout directory.
 
     model: MaskRCNNModel (or the MaskRCNN facade; .model is used).
     weight_dtype: "fp32" or "fp8" (e4m3fn per-out-channel quantized
     conv/linear weights — BASELINE config #5).
+    warmup_batches: list of batch sizes (e.g. [32]) — writes TF-Serving
+    `assets.extra/tf_serving_warmup_requests` (TFRecord of
+    PredictionLogs) replayed by the server at load time.
     Returns the version directory path.
     """
     import torch
@@ -106,6 +110,26 @@ def export_saved_model(model, config, base_dir: str, version: int = 1,
         index["meta"] = extra_meta
     with open(os.path.join(var_dir, "variables.index"), "w") as f:
         json.dump(index, f, indent=1, default=str)
+
+    if warmup_batches:
+        import numpy as np
+        from .warmup import write_warmup_requests
+        from .wireproto import (ModelSpec, PredictRequest,
+                                numpy_to_tensor)
+        H, W, _ = config.IMAGE_SHAPE
+        reqs = []
+        for b in warmup_batches:
+            r = PredictRequest(model_spec=ModelSpec(
+                name="", signature_name="serving_default"))
+            # uint8 zeros: 1/4 the record bytes; engine molds on-GPU.
+            r.inputs["input_image"] = numpy_to_tensor(
+                np.zeros((b, H, W, 3), np.uint8))
+            meta = np.zeros((b, config.IMAGE_META_SIZE), np.float32)
+            meta[:, 4:7] = (H, W, 3)
+            meta[:, 7:11] = (0, 0, H, W)
+            r.inputs["input_image_meta"] = numpy_to_tensor(meta)
+            reqs.append(r)
+        write_warmup_requests(export_dir, reqs)
     return export_dir
 
 

@@ -112,6 +112,32 @@ class ModelState:
             eng, max_batch_size=self.max_batch_size,
             batch_timeout_micros=self.batch_timeout_micros)
             for eng in engines]
+        # TF-Serving model_warmup: replay assets.extra/
+        # tf_serving_warmup_requests (TFRecord of PredictionLogs)
+        # through every replica BEFORE the version goes AVAILABLE, so
+        # the first live request pays no MIOpen-Find/first-launch cost.
+        try:
+            from .warmup import read_warmup_requests
+            from .wireproto import tensor_to_numpy
+            reqs = read_warmup_requests(export_dir)
+            for i, wreq in enumerate(reqs):
+                images = tensor_to_numpy(wreq.inputs["input_image"])
+                if images.dtype != np.uint8:
+                    images = images.astype(np.float32)
+                metas = tensor_to_numpy(
+                    wreq.inputs["input_image_meta"]).astype(np.float32)
+                anchors = None
+                if "input_anchors" in wreq.inputs:
+                    anchors = tensor_to_numpy(
+                        wreq.inputs["input_anchors"]).astype(np.float32)
+                for eng in engines:
+                    eng.predict(images, metas, anchors)
+            if reqs:
+                log.info("replayed %d warmup request(s) for %s v%d",
+                         len(reqs), self.name, version)
+        except Exception:  # noqa: BLE001 — warmup must never block load
+            log.exception("model warmup replay failed (%s v%d)",
+                          self.name, version)
         return {"engines": engines, "batchers": batchers}
 
     def maybe_reload(self) -> bool:

@@ -1234,3 +1234,84 @@ def test_model_config_version_labels_parse(tmp_path):
     cfgs = parse_model_config(str(mc))
     assert cfgs[0]["version_policy"] == "all"
     assert cfgs[0]["version_labels"] == {"stable": 1, "canary": 2}
+
+
+def test_warmup_tfrecord_roundtrip(tmp_path):
+    """TFRecord framing (masked crc32c) + PredictionLog wire encode/
+    decode roundtrip, plus TF-mask crc self-check."""
+    from maskrcnn_amd.serving.warmup import (decode_prediction_log,
+                                             encode_prediction_log,
+                                             read_warmup_requests,
+                                             tfrecord_read,
+                                             tfrecord_write,
+                                             write_warmup_requests)
+    recs = [b"abc", b"", b"\x00" * 1000]
+    p = str(tmp_path / "r.tfrecord")
+    tfrecord_write(p, recs)
+    assert tfrecord_read(p) == recs
+    # corrupt one data byte -> crc error
+    blob = bytearray(open(p, "rb").read())
+    blob[12] ^= 0xFF
+    open(p, "wb").write(bytes(blob))
+    with pytest.raises(ValueError):
+        tfrecord_read(p)
+
+    req = PredictRequest(model_spec=ModelSpec(name="m"))
+    req.inputs["input_image"] = numpy_to_tensor(
+        np.zeros((2, 4, 4, 3), np.uint8))
+    req.inputs["input_image_meta"] = numpy_to_tensor(
+        np.ones((2, 5), np.float32))
+    back = decode_prediction_log(encode_prediction_log(req))
+    np.testing.assert_array_equal(
+        tensor_to_numpy(back.inputs["input_image_meta"]),
+        np.ones((2, 5), np.float32))
+
+    vd = str(tmp_path / "1")
+    write_warmup_requests(vd, [req])
+    assert os.path.isfile(
+        os.path.join(vd, "assets.extra", "tf_serving_warmup_requests"))
+    got = read_warmup_requests(vd)
+    assert len(got) == 1
+    assert tensor_to_numpy(got[0].inputs["input_image"]).shape == \
+        (2, 4, 4, 3)
+
+
+def test_model_warmup_replayed_at_load(tmp_path):
+    """export(warmup_batches=[2]) -> ModelState load replays the warmup
+    request through the engine before the version is AVAILABLE."""
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+    from maskrcnn_amd.serving.server import ModelState
+
+    class TinyConfig(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 100
+        POST_NMS_ROIS_INFERENCE = 10
+
+    cfg = TinyConfig()
+    m = MaskRCNN("inference", cfg, model_dir=str(tmp_path))
+    base = str(tmp_path / "models" / "wu")
+    export_saved_model(m, cfg, base, version=1, warmup_batches=[2])
+    assert os.path.isfile(os.path.join(
+        base, "1", "assets.extra", "tf_serving_warmup_requests"))
+
+    calls = []
+    import maskrcnn_amd.serving.engine as E
+    orig = E.InferenceEngine.predict
+
+    def spy(self, *a, **k):
+        calls.append(1)
+        return orig(self, *a, **k)
+
+    E.InferenceEngine.predict = spy
+    try:
+        st = ModelState("wu", base, max_batch_size=2,
+                        batch_timeout_micros=100, config=cfg)
+        st.maybe_reload()
+        assert len(calls) >= 1, "warmup request was not replayed"
+    finally:
+        E.InferenceEngine.predict = orig
+        for slot in st.versions.values():
+            for b in slot["batchers"]:
+                b.close()
