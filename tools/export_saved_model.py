@@ -25,6 +25,13 @@ def main():
     ap.add_argument("--num-classes", type=int, default=81)
     ap.add_argument("--backbone", default="resnet101")
     ap.add_argument("--image-size", type=int, default=1024)
+    ap.add_argument("--weight-dtype", default="fp32",
+                    choices=["fp32", "fp8"])
+    ap.add_argument("--warmup-batches", type=int, nargs="*",
+                    default=None,
+                    help="write assets.extra/tf_serving_warmup_requests"
+                         " for these batch sizes (e.g. --warmup-batches"
+                         " 32); the server replays them at load")
     args = ap.parse_args()
 
     from maskrcnn_amd.config import Config
@@ -44,7 +51,9 @@ def main():
     model.load_weights(args.checkpoint, by_name=True)
     out = export_saved_model(model, cfg, args.export_dir,
                              version=args.version,
-                             extra_meta={"checkpoint": args.checkpoint})
+                             extra_meta={"checkpoint": args.checkpoint},
+                             weight_dtype=args.weight_dtype,
+                             warmup_batches=args.warmup_batches)
     print("exported", out)
 
 
