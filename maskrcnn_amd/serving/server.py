@@ -58,7 +58,8 @@ class ModelState:
     def __init__(self, name: str, base_path: str, device=None,
                  devices=None, max_batch_size: int = 32,
                  batch_timeout_micros: int = 2000, config=None,
-                 version_policy: str = "latest", version_labels=None):
+                 version_policy: str = "latest", version_labels=None,
+                 enable_warmup: bool = True):
         """version_policy: 'latest' (TF-Serving default — new version
         replaces old) or 'all' (every discovered version stays
         servable; requests pin via model_spec.version).
@@ -69,6 +70,7 @@ class ModelState:
         assert version_policy in ("latest", "all"), version_policy
         self.name = name
         self.version_labels = dict(version_labels or {})
+        self.enable_warmup = enable_warmup
         self.base_path = base_path
         self.devices = list(devices) if devices else [device]
         self.max_batch_size = max_batch_size
@@ -119,7 +121,8 @@ class ModelState:
         try:
             from .warmup import read_warmup_requests
             from .wireproto import tensor_to_numpy
-            reqs = read_warmup_requests(export_dir)
+            reqs = read_warmup_requests(export_dir) \
+                if self.enable_warmup else []
             for i, wreq in enumerate(reqs):
                 images = tensor_to_numpy(wreq.inputs["input_image"])
                 if images.dtype != np.uint8:
@@ -232,7 +235,8 @@ class ModelServer:
                  batch_timeout_micros: int = 2000,
                  poll_interval_s: float = 5.0, config=None,
                  model_configs=None, version_policy: str = "latest",
-                 grpc_ingress: str = "python"):
+                 grpc_ingress: str = "python",
+                 enable_model_warmup: bool = True):
         """Single model (model_name + model_base_path, TF-Serving's
         --model_name/--model_base_path mode) or several (model_configs =
         parse_model_config(models.config), the --model_config_file
@@ -247,7 +251,8 @@ class ModelServer:
             device=device, devices=devices,
             max_batch_size=max_batch_size,
             batch_timeout_micros=batch_timeout_micros, config=config,
-            version_policy=version_policy)
+            version_policy=version_policy,
+            enable_warmup=enable_model_warmup)
         self.models: Dict[str, ModelState] = {
             mc["name"]: ModelState(
                 mc["name"], mc["base_path"],
@@ -259,6 +264,11 @@ class ModelServer:
         self.rest_port = rest_port
         self.metrics_port = metrics_port
         self.poll_interval_s = poll_interval_s
+        # --model_config_file_poll_wait_seconds support (set by main()
+        # or callers; 0 = only reload via the ModelService RPC).
+        self.model_config_path: Optional[str] = None
+        self.config_poll_interval_s: float = 0.0
+        self._last_config_poll = 0.0
         if grpc_ingress not in ("python", "cpp"):
             raise ValueError(f"grpc_ingress must be python|cpp, "
                              f"got {grpc_ingress!r}")
@@ -292,7 +302,11 @@ class ModelServer:
                 new[mc["name"]] = cur
             else:
                 st = ModelState(mc["name"], mc["base_path"],
-                                **self._model_kwargs)
+                                **{**self._model_kwargs,
+                                   **{k: mc[k]
+                                      for k in ("version_policy",
+                                                "version_labels")
+                                      if k in mc}})
                 st.maybe_reload()
                 if st.version is None:
                     raise FileNotFoundError(
@@ -723,6 +737,20 @@ class ModelServer:
 
     def _poll_loop(self):
         while not self._stop.wait(self.poll_interval_s):
+            # TF-Serving --model_config_file_poll_wait_seconds: re-read
+            # the models.config file itself and apply adds/removes.
+            if self.model_config_path and \
+                    self.config_poll_interval_s > 0:
+                now = time.monotonic()
+                if now - self._last_config_poll >= \
+                        self.config_poll_interval_s:
+                    self._last_config_poll = now
+                    try:
+                        from .config_file import parse_model_config
+                        self.reload_config(
+                            parse_model_config(self.model_config_path))
+                    except Exception:  # noqa: BLE001
+                        log.exception("model_config_file poll failed")
             for state in self.models.values():
                 try:
                     state.maybe_reload()
@@ -773,6 +801,18 @@ def main():  # pragma: no cover
                     default="python",
                     help="cpp = native HTTP/2 front-end "
                          "(csrc/ingress.cpp, no GIL on the wire path)")
+    ap.add_argument("--file_system_poll_wait_seconds", type=float,
+                    default=5.0,
+                    help="version-directory poll interval "
+                         "(TF-Serving flag; 0 disables polling)")
+    ap.add_argument("--model_config_file_poll_wait_seconds", type=float,
+                    default=0.0,
+                    help="re-read --model_config_file at this interval "
+                         "and apply adds/removes (0 = RPC reload only)")
+    ap.add_argument("--enable_model_warmup",
+                    action=argparse.BooleanOptionalAction, default=True,
+                    help="replay assets.extra/tf_serving_warmup_requests"
+                         " at version load (TF-Serving default on)")
     args = ap.parse_args()
     logging.basicConfig(level=logging.INFO)
     devices = None
@@ -797,7 +837,13 @@ def main():  # pragma: no cover
         metrics_port=args.metrics_port,
         devices=devices, model_configs=model_configs,
         max_batch_size=max_batch, batch_timeout_micros=timeout_us,
-        grpc_ingress=args.grpc_ingress)
+        grpc_ingress=args.grpc_ingress,
+        poll_interval_s=args.file_system_poll_wait_seconds or 3600.0,
+        enable_model_warmup=args.enable_model_warmup)
+    if args.model_config_file:
+        server.model_config_path = args.model_config_file
+        server.config_poll_interval_s = \
+            args.model_config_file_poll_wait_seconds
     server.start()
     server.wait()
 

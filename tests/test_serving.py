@@ -1315,3 +1315,55 @@ def test_model_warmup_replayed_at_load(tmp_path):
         for slot in st.versions.values():
             for b in slot["batchers"]:
                 b.close()
+
+
+def test_model_config_file_polling(tmp_path):
+    """--model_config_file_poll_wait_seconds: editing models.config on
+    disk adds/removes served models without an RPC."""
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+    from maskrcnn_amd.serving.server import ModelServer
+
+    class TinyConfig(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 100
+        POST_NMS_ROIS_INFERENCE = 10
+
+    cfg = TinyConfig()
+    m = MaskRCNN("inference", cfg, model_dir=str(tmp_path))
+    base_a = str(tmp_path / "models" / "a")
+    base_b = str(tmp_path / "models" / "b")
+    export_saved_model(m, cfg, base_a, version=1)
+    export_saved_model(m, cfg, base_b, version=1)
+
+    mc = tmp_path / "models.config"
+
+    def write_cfg(names_paths):
+        body = "".join(
+            "config { name: '%s' base_path: '%s' }\n" % np
+            for np_ in [None] for np in names_paths)
+        mc.write_text("model_config_list { %s }\n" % body)
+
+    write_cfg([("a", base_a)])
+    server = ModelServer(
+        model_configs=[{"name": "a", "base_path": base_a}],
+        grpc_port=0, rest_port=None, max_batch_size=1,
+        batch_timeout_micros=100, config=cfg, poll_interval_s=0.2)
+    server.model_config_path = str(mc)
+    server.config_poll_interval_s = 0.2
+    server.start()
+    try:
+        assert set(server.models) == {"a"}
+        write_cfg([("a", base_a), ("b", base_b)])
+        deadline = time.time() + 30
+        while time.time() < deadline and set(server.models) != {"a", "b"}:
+            time.sleep(0.2)
+        assert set(server.models) == {"a", "b"}
+        write_cfg([("b", base_b)])
+        deadline = time.time() + 30
+        while time.time() < deadline and set(server.models) != {"b"}:
+            time.sleep(0.2)
+        assert set(server.models) == {"b"}
+    finally:
+        server.stop()
