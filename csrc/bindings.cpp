@@ -44,6 +44,9 @@ void conv3x3_v5_launch(const void*, const void*, const float*, const void*,
                        void*, int, int, int, int, int, bool, hipStream_t);
 void conv3x3_wrw_launch(const void*, const void*, float*, int, int, int,
                         int, int, hipStream_t);
+void conv3x3_splitk_launch(const void*, const void*, const float*, void*,
+                           void*, int, int, int, int, int, int, bool,
+                           hipStream_t);
 void conv3x3_fp8_launch(const void*, const void*, const float*,
                         const float*, const float*, void*, int, int, int,
                         int, int, bool, hipStream_t);
@@ -295,6 +298,20 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w_packed,
     // the 256-row tiles leave that grid at ~1 WG/CU). v1 below 32768.
     ver = (M >= 131072 && Cout >= 128) ? 5
         : (M >= 32768 && Cout >= 128) ? 3 : 1;
+  if (ver == 6) {
+    // Split-K for fill-bound shapes (M tiles x N tiles < #CUs): the
+    // 144 (tap, k-chunk) steps spread over grid.z into per-z fp32
+    // slabs, reduced in fixed order (deterministic, no atomics).
+    // workspace = split*M*N f32 (16 MB at the target shape).
+    const int split = 8;  // (2,512,32²,512) -> 32x8x8 = 2048 blocks
+    auto ws = torch::empty({(long)split * M * Cout},
+                           x.options().dtype(torch::kFloat));
+    conv3x3_splitk_launch(x.data_ptr(), w_packed.data_ptr(),
+                          bias_f.data_ptr<float>(), ws.data_ptr(),
+                          out.data_ptr(), B, H, W, Cin, Cout, split,
+                          relu, cur_stream());
+    return out;
+  }
   if (ver >= 2) {
     // Persistent 64-B zero page per device (glds cannot predicate
     // out-of-range lanes; they DMA from here instead).
