@@ -65,17 +65,24 @@ def main():
         v3 = timeit(lambda: ext.conv3x3_fwd(x, wp, bf, True, 3))
         v4 = timeit(lambda: ext.conv3x3_fwd(x, wp, bf, True, 4))
         v5 = timeit(lambda: ext.conv3x3_fwd(x, wp, bf, True, 5))
+        # v6 split-K targets fill-bound small-M grids only.
+        v6 = timeit(lambda: ext.conv3x3_fwd(x, wp, bf, True, 6)) \
+            if B * H * W <= 8192 else float("inf")
         ref = timeit(lambda: F.relu(F.conv2d(x, w, b.bfloat16(),
                                              padding=1)))
         flops = 2.0 * B * H * W * Cout * Cin * 9
-        tfs = [flops / (t * 1e-3) / 1e12 for t in (v1, v2, v3, v4, v5)]
+        tfs = [flops / (t * 1e-3) / 1e12
+               for t in (v1, v2, v3, v4, v5, v6)]
+        v6s = (f"v6={v6:7.3f}ms({tfs[5]:5.0f}TF) "
+               if v6 != float("inf") else "")
         print(f"{str((B, Cin, H, W, Cout)):>28} "
               f"v1={v1:7.3f}ms({tfs[0]:5.0f}TF) "
               f"v2={v2:7.3f}ms({tfs[1]:5.0f}TF) "
               f"v3={v3:7.3f}ms({tfs[2]:5.0f}TF) "
               f"v4={v4:7.3f}ms({tfs[3]:5.0f}TF) "
-              f"v5={v5:7.3f}ms({tfs[4]:5.0f}TF) miopen={ref:7.3f}ms "
-              f"best_ratio={ref / min(v1, v2, v3, v4, v5):5.2f}x")
+              f"v5={v5:7.3f}ms({tfs[4]:5.0f}TF) {v6s}"
+              f"miopen={ref:7.3f}ms "
+              f"best_ratio={ref / min(v1, v2, v3, v4, v5, v6):5.2f}x")
 
     # fpn_fuse vs unfused (conv1x1 + interpolate + add).
     print("\nfpn_fuse (1x1 + up2 + add):")
