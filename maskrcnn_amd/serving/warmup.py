@@ -22,7 +22,7 @@ import os
 import struct
 from typing import List
 
-from .wireproto import PredictRequest, _r_varint, _w_len_field
+from .wireproto import PredictRequest, _w_len_field
 
 _CRC_TABLE = []
 
