@@ -414,3 +414,28 @@ def test_ingress_request_trailers_half_close():
         s.close()
     finally:
         srv.stop()
+
+
+def test_ingress_randomized_sizes_stress():
+    """50 echo calls of random sizes 0..512 KB on one channel — framing
+    and flow control must be exact at every size (incl. 0 bytes)."""
+    import random
+
+    def handler(path, payload):
+        return (0, "", bytes(payload))
+
+    srv = ingress.IngressServer(0, handler, 4)
+    srv.start()
+    try:
+        ch, stub = _raw_stub(srv.port)
+        rng = random.Random(42)
+        for i in range(50):
+            n = rng.choice([0, 1, 5, rng.randint(0, 512 * 1024)])
+            msg = bytes(rng.getrandbits(8) for _ in range(min(n, 256))) \
+                * (n // min(n, 256) if n else 0) + b"x" * (
+                    n % min(n, 256) if n else 0)
+            msg = msg[:n]
+            assert stub(msg, timeout=15) == msg, f"size {n} (call {i})"
+        ch.close()
+    finally:
+        srv.stop()
