@@ -112,14 +112,30 @@ inline void frame_header(uint8_t* p, uint32_t len, uint8_t type,
   wr_u32(p + 5, sid & 0x7fffffffu);
 }
 
-// HPACK literal-without-indexing, new name, no Huffman (RFC 7541 §6.2.2
-// with the 7-bit string-length prefix; lengths here are all < 127).
+// HPACK 7-bit-prefix integer (RFC 7541 §5.1) — string lengths can
+// exceed 126 (long grpc-message error details), so the multi-byte
+// form is required, not optional.
+void hpack_len7(std::vector<uint8_t>& out, size_t v) {
+  if (v < 127) {
+    out.push_back(uint8_t(v));
+    return;
+  }
+  out.push_back(0x7f);
+  v -= 127;
+  while (v >= 128) {
+    out.push_back(uint8_t((v & 0x7f) | 0x80));
+    v >>= 7;
+  }
+  out.push_back(uint8_t(v));
+}
+
+// HPACK literal-without-indexing, new name, no Huffman (RFC 7541 §6.2.2).
 void hpack_literal(std::vector<uint8_t>& out, const std::string& name,
                    const std::string& value) {
   out.push_back(0x00);
-  out.push_back(uint8_t(name.size()));
+  hpack_len7(out, name.size());
   out.insert(out.end(), name.begin(), name.end());
-  out.push_back(uint8_t(value.size()));
+  hpack_len7(out, value.size());
   out.insert(out.end(), value.begin(), value.end());
 }
 

@@ -439,3 +439,27 @@ def test_ingress_randomized_sizes_stress():
         ch.close()
     finally:
         srv.stop()
+
+
+def test_ingress_long_error_message():
+    """grpc-message values over 127 bytes require HPACK's multi-byte
+    7-bit-prefix length (a one-byte length here corrupted the header
+    block)."""
+    import grpc
+    long_msg = "x" * 300 + " END"
+
+    def handler(path, payload):
+        return (13, long_msg, b"")
+
+    srv = ingress.IngressServer(0, handler, 2)
+    srv.start()
+    try:
+        ch, stub = _raw_stub(srv.port)
+        with pytest.raises(grpc.RpcError) as e:
+            stub(b"q", timeout=10)
+        assert e.value.code() == grpc.StatusCode.INTERNAL
+        assert e.value.details().endswith(" END")
+        assert len(e.value.details()) >= 300
+        ch.close()
+    finally:
+        srv.stop()
