@@ -28,6 +28,7 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <atomic>
 #include <chrono>
 #include <condition_variable>
@@ -38,6 +39,7 @@
 #include <string>
 #include <thread>
 #include <unordered_map>
+#include <unordered_set>
 #include <vector>
 
 namespace py = pybind11;
@@ -285,16 +287,45 @@ class IngressServer {
       if (t.joinable()) t.join();
     workers_.clear();
     {
-      std::lock_guard<std::mutex> lk(cmu_);
-      for (auto& t : conn_threads_)
+      // Joining MUST happen without cmu_ held: exiting conn threads
+      // take cmu_ to mark themselves finished.
+      std::vector<std::thread> ts;
+      {
+        std::lock_guard<std::mutex> lk(cmu_);
+        ts.swap(conn_threads_);
+        finished_.clear();
+      }
+      for (auto& t : ts)
         if (t.joinable()) t.join();
-      conn_threads_.clear();
     }
     ::close(listen_fd_);
     listen_fd_ = -1;
   }
 
  private:
+  // Join connection threads that have finished (their conn_loop
+  // signalled done). Without this a long-running daemon accumulates a
+  // zombie std::thread handle + expired weak_ptr per client that ever
+  // connected. Called with cmu_ HELD.
+  void reap_locked() {
+    if (finished_.empty()) return;
+    for (auto it = conn_threads_.begin(); it != conn_threads_.end();) {
+      auto fin = finished_.find(it->get_id());
+      if (fin != finished_.end()) {
+        it->join();
+        finished_.erase(fin);
+        it = conn_threads_.erase(it);
+      } else {
+        ++it;
+      }
+    }
+    conns_.erase(std::remove_if(conns_.begin(), conns_.end(),
+                                [](const std::weak_ptr<Conn>& w) {
+                                  return w.expired();
+                                }),
+                 conns_.end());
+  }
+
   // ------------------------------------------------------------ accept
   void accept_loop() {
     while (running_) {
@@ -306,9 +337,13 @@ class IngressServer {
       conn->fd = fd;
       nghttp2_hd_inflate_new(&conn->inflater);
       std::lock_guard<std::mutex> lk(cmu_);
+      reap_locked();
       conns_.push_back(conn);
-      conn_threads_.emplace_back(
-          [this, conn] { conn_loop(conn); });
+      conn_threads_.emplace_back([this, conn] {
+        conn_loop(conn);
+        std::lock_guard<std::mutex> lk2(cmu_);
+        finished_.insert(std::this_thread::get_id());
+      });
     }
   }
 
@@ -737,6 +772,7 @@ class IngressServer {
   std::mutex cmu_;
   std::vector<std::weak_ptr<Conn>> conns_;
   std::vector<std::thread> conn_threads_;
+  std::unordered_set<std::thread::id> finished_;
   std::mutex qmu_;
   std::condition_variable qcv_;
   std::deque<Request> queue_;
