@@ -272,10 +272,17 @@ class IngressServer {
         std::this_thread::sleep_for(std::chrono::milliseconds(10));
       }
     }
-    {  // kick every live connection so reader threads return
+    {  // GOAWAY(NO_ERROR) then kick every live connection so reader
+       // threads return; grpc clients surface UNAVAILABLE instead of a
+       // raw socket reset.
       std::lock_guard<std::mutex> lk(cmu_);
       for (auto& w : conns_)
         if (auto c = w.lock()) {
+          uint8_t ga[17];
+          frame_header(ga, 8, F_GOAWAY, 0, 0);
+          wr_u32(ga + 9, 0x7fffffffu);  // last-stream-id
+          wr_u32(ga + 13, 0);           // NO_ERROR
+          c->write_all(ga, 17);
           c->dead = true;
           ::shutdown(c->fd, SHUT_RDWR);
           c->cv.notify_all();
