@@ -1367,3 +1367,45 @@ def test_model_config_file_polling(tmp_path):
         assert set(server.models) == {"b"}
     finally:
         server.stop()
+
+
+def test_batcher_uint8_mask_top_pad_combination(shapes_config):
+    """The production hot path in one unit: uint8 wire images + the
+    mrcnn_mask_top filter + pad-to-max batching, through DynamicBatcher
+    -> engine (engine-side GPU-analog gather + pinned-collate path),
+    with per-request slices returned intact."""
+    from maskrcnn_amd.serving.batcher import DynamicBatcher
+    from maskrcnn_amd.serving.engine import InferenceEngine
+
+    cfg = shapes_config
+    eng = InferenceEngine(cfg, device="cpu", use_bf16=False)
+    b = DynamicBatcher(eng, max_batch_size=4, batch_timeout_micros=50000,
+                       pad_to_max=True)
+    H = cfg.IMAGE_SHAPE[0]
+    rng = np.random.RandomState(0)
+    meta = np.zeros((1, cfg.IMAGE_META_SIZE), np.float32)
+    meta[:, 4:7] = (H, H, 3)
+    meta[:, 7:11] = (0, 0, H, H)
+    futs = []
+    imgs = []
+    for i in range(3):  # 3 requests -> one batch padded to 4
+        img = (rng.rand(1, H, H, 3) * 255).astype(np.uint8)
+        imgs.append(img)
+        futs.append(b.submit(img, meta.copy(),
+                             outputs=["mrcnn_detection",
+                                      "mrcnn_mask_top"]))
+    outs = [f.result(timeout=120) for f in futs]
+    D = cfg.DETECTION_MAX_INSTANCES
+    for out in outs:
+        assert set(out) >= {"mrcnn_detection", "mrcnn_mask_top"}
+        assert out["mrcnn_detection"].shape == (1, D, 6)
+        assert out["mrcnn_mask_top"].shape == (1, D, 28, 28)
+    # Each request's slice must equal running it alone (same molding,
+    # no cross-contamination from batching/padding).
+    solo = eng.predict(imgs[1], meta.copy(),
+                       outputs=["mrcnn_detection", "mrcnn_mask_top"])
+    np.testing.assert_allclose(outs[1]["mrcnn_detection"],
+                               solo["mrcnn_detection"], atol=1e-4)
+    np.testing.assert_allclose(outs[1]["mrcnn_mask_top"],
+                               solo["mrcnn_mask_top"], atol=1e-4)
+    b.close()
