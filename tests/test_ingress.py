@@ -508,3 +508,37 @@ def test_ingress_grpc_length_overflow():
         ch.close()
     finally:
         srv.stop()
+
+
+def test_cpp_ingress_version_hot_reload(cpp_served_model, tmp_path):
+    """A new version dropped on disk is picked up by the poller and
+    served through the cpp ingress without restarting it."""
+    import time
+    server, cfg, m = cpp_served_model
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+    from maskrcnn_amd.serving.client import PREDICT_METHOD
+    import grpc
+
+    start_version = server.state.version
+    export_saved_model(m, cfg, server.state.base_path,
+                       version=start_version + 1)
+    deadline = time.time() + 60
+    while time.time() < deadline and \
+            server.state.version != start_version + 1:
+        time.sleep(0.25)
+    assert server.state.version == start_version + 1
+
+    ch = grpc.insecure_channel(f"localhost:{server.grpc_port}")
+    stub = ch.unary_unary(
+        PREDICT_METHOD, request_serializer=lambda r: r.encode(),
+        response_deserializer=PredictResponse.decode)
+    H = cfg.IMAGE_SHAPE[0]
+    req = PredictRequest(model_spec=ModelSpec(name="mask_rcnn"))
+    req.inputs["input_image"] = numpy_to_tensor(
+        np.zeros((1, H, H, 3), np.float32))
+    req.inputs["input_image_meta"] = numpy_to_tensor(
+        np.zeros((1, cfg.IMAGE_META_SIZE), np.float32))
+    req.output_filter = ["mrcnn_detection"]
+    resp = stub(req, timeout=60)
+    assert resp.model_spec.version == start_version + 1
+    ch.close()
