@@ -35,6 +35,9 @@ try:
     REQS = Counter("predict_requests_total", "Predict requests",
                    ["proto"])
     ERRS = Counter("predict_errors_total", "Predict errors", ["proto"])
+    REJECTS = Counter("predict_rejected_total",
+                      "Predicts shed by the full batching queue "
+                      "(RESOURCE_EXHAUSTED/429)", ["proto"])
     LAT = Histogram("predict_latency_seconds", "Predict latency",
                     buckets=(.005, .01, .025, .05, .1, .25, .5, 1, 2.5, 5))
     BATCH = Histogram("predict_batch_size", "executed batch size",
@@ -445,7 +448,7 @@ class ModelServer:
                     context.abort(grpc.StatusCode.NOT_FOUND, str(e))
                 except BatcherQueueFull as e:
                     if _PROM:
-                        ERRS.labels("grpc").inc()
+                        REJECTS.labels("grpc").inc()
                     context.abort(grpc.StatusCode.RESOURCE_EXHAUSTED,
                                   str(e))
                 except Exception as e:  # noqa: BLE001
@@ -531,7 +534,7 @@ class ModelServer:
                 return (5, str(e), b"")  # NOT_FOUND
             except BatcherQueueFull as e:
                 if _PROM:
-                    ERRS.labels("grpc").inc()
+                    REJECTS.labels("grpc").inc()
                 return (8, str(e), b"")  # RESOURCE_EXHAUSTED
             except Exception as e:  # noqa: BLE001
                 if _PROM:
@@ -689,6 +692,8 @@ class ModelServer:
                     ERRS.labels("rest").inc()
                 from .batcher import BatcherQueueFull
                 if isinstance(e, BatcherQueueFull):
+                    if _PROM:
+                        REJECTS.labels("rest").inc()
                     return JSONResponse({"error": str(e)},
                                         status_code=429)
                 log.exception("REST predict failed")
