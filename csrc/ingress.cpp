@@ -148,7 +148,36 @@ struct Stream {
   bool headers_done = false;
   bool end_stream = false;
   int64_t send_window = 65535;  // reset from peer INITIAL_WINDOW_SIZE
+  // Absolute deadline from the client's grpc-timeout header (steady
+  // clock); time_point::max() = none.
+  std::chrono::steady_clock::time_point deadline =
+      std::chrono::steady_clock::time_point::max();
 };
+
+// grpc-timeout: "<digits><unit>", unit in H M S m u n (RFC: gRPC
+// HTTP/2 transport spec). Returns max() on parse failure.
+inline std::chrono::steady_clock::time_point
+parse_grpc_timeout(const uint8_t* v, size_t n) {
+  if (n < 2 || n > 9) return std::chrono::steady_clock::time_point::max();
+  uint64_t val = 0;
+  for (size_t i = 0; i + 1 < n; ++i) {
+    if (v[i] < '0' || v[i] > '9')
+      return std::chrono::steady_clock::time_point::max();
+    val = val * 10 + (v[i] - '0');
+  }
+  uint64_t ns;
+  switch (v[n - 1]) {
+    case 'H': ns = val * 3600ull * 1000000000ull; break;
+    case 'M': ns = val * 60ull * 1000000000ull; break;
+    case 'S': ns = val * 1000000000ull; break;
+    case 'm': ns = val * 1000000ull; break;
+    case 'u': ns = val * 1000ull; break;
+    case 'n': ns = val; break;
+    default: return std::chrono::steady_clock::time_point::max();
+  }
+  return std::chrono::steady_clock::now() +
+         std::chrono::nanoseconds(ns);
+}
 
 struct Conn {
   int fd = -1;
@@ -595,6 +624,9 @@ class IngressServer {
       if (iflags & HD_INFLATE_EMIT) {
         if (nv.namelen == 5 && memcmp(nv.name, ":path", 5) == 0)
           st->path.assign((const char*)nv.value, nv.valuelen);
+        else if (nv.namelen == 12 &&
+                 memcmp(nv.name, "grpc-timeout", 12) == 0)
+          st->deadline = parse_grpc_timeout(nv.value, nv.valuelen);
       }
       if (iflags & HD_INFLATE_FINAL) {
         nghttp2_hd_inflate_end_headers(conn->inflater);
@@ -641,6 +673,18 @@ class IngressServer {
     int status = 0;
     std::string message;
     std::string resp;
+    // A request whose client deadline already passed while queued must
+    // not waste a model batch: answer DEADLINE_EXCEEDED immediately
+    // (the client has usually RSTed the stream anyway — the write then
+    // just fails quietly).
+    if (req.stream &&
+        std::chrono::steady_clock::now() > req.stream->deadline) {
+      send_response(req.conn, req.sid, req.stream, 4,
+                    "Deadline Exceeded (expired while queued)", "");
+      std::lock_guard<std::mutex> lk(req.conn->mu);
+      req.conn->streams.erase(req.sid);
+      return;
+    }
     // gRPC framing: 1-byte compressed flag + u32 length + message.
     if (req.body.size() < 5) {
       status = 13;

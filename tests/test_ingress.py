@@ -542,3 +542,42 @@ def test_cpp_ingress_version_hot_reload(cpp_served_model, tmp_path):
     resp = stub(req, timeout=60)
     assert resp.model_spec.version == start_version + 1
     ch.close()
+
+
+def test_ingress_expired_deadline_skips_handler():
+    """grpc-timeout honored: a request whose client deadline passed
+    while queued gets DEADLINE_EXCEEDED WITHOUT entering the handler
+    (no wasted model batch). One worker; first request occupies it past
+    the second's 200 ms deadline."""
+    import time
+    calls = []
+
+    def handler(path, payload):
+        calls.append(bytes(payload))
+        time.sleep(1.0)
+        return (0, "", b"done")
+
+    srv = ingress.IngressServer(0, handler, 1)  # single worker
+    srv.start()
+    try:
+        ch1, stub1 = _raw_stub(srv.port)
+        ch2, stub2 = _raw_stub(srv.port)
+        res = {}
+
+        def slow():
+            res["slow"] = stub1(b"slow", timeout=30)
+
+        t = threading.Thread(target=slow)
+        t.start()
+        time.sleep(0.15)  # ensure 'slow' holds the worker
+        import grpc
+        with pytest.raises(grpc.RpcError):
+            stub2(b"fast", timeout=0.2)  # expires while queued
+        t.join(timeout=30)
+        assert res.get("slow") == b"done"
+        time.sleep(0.5)  # give the worker time to (not) run 'fast'
+        assert calls == [b"slow"], calls
+        ch1.close()
+        ch2.close()
+    finally:
+        srv.stop()
