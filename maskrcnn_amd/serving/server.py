@@ -680,9 +680,26 @@ class ModelServer:
                     return JSONResponse(
                         {"error": "missing inputs/instances"},
                         status_code=400)
+                # Extension (mirrors the gRPC output_filter): restrict
+                # which outputs are computed AND serialized — a full
+                # 7-output JSON response tolist()s the 33 MB mask
+                # tensor. Supports mrcnn_mask_top like gRPC.
+                want = body.get("output_filter")
+                if want is not None and not isinstance(want, list):
+                    return JSONResponse(
+                        {"error": "output_filter must be a list"},
+                        status_code=400)
+                eng_want = want
+                if want and "mrcnn_mask_top" in want and \
+                        "mrcnn_detection" not in want:
+                    eng_want = want + ["mrcnn_detection"]
                 t0 = time.monotonic()
                 outputs, _ = state.predict(images, metas, anchors,
-                                           version=want_version)
+                                           version=want_version,
+                                           outputs=eng_want)
+                if want:
+                    outputs = {k: v for k, v in outputs.items()
+                               if k in want}
                 if _PROM:
                     LAT.observe(time.monotonic() - t0)
                 return {"outputs": {k: v.tolist()

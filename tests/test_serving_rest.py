@@ -277,3 +277,32 @@ def test_rest_version_labels_route(tmp_path):
         for slot in st.versions.values():
             for b in slot["batchers"]:
                 b.close()
+
+
+def test_rest_output_filter_extension(rest_server):
+    """REST body `output_filter` (wire extension mirroring gRPC):
+    restricts computed+serialized outputs, supports mrcnn_mask_top."""
+    import requests
+    server, cfg, port = rest_server
+    H = cfg.IMAGE_SHAPE[0]
+    meta = np.zeros((1, cfg.IMAGE_META_SIZE), np.float32)
+    meta[:, 4:7] = (H, H, 3)
+    meta[:, 7:11] = (0, 0, H, H)
+    body = {"inputs": {
+        "input_image": np.zeros((1, H, H, 3), np.float32).tolist(),
+        "input_image_meta": meta.tolist()},
+        "output_filter": ["mrcnn_detection", "mrcnn_mask_top"]}
+    r = requests.post(
+        f"http://127.0.0.1:{port}/v1/models/mask_rcnn:predict",
+        json=body, timeout=120)
+    assert r.status_code == 200, r.text
+    out = r.json()["outputs"]
+    assert set(out) == {"mrcnn_detection", "mrcnn_mask_top"}
+    D = cfg.DETECTION_MAX_INSTANCES
+    assert np.asarray(out["mrcnn_mask_top"]).shape == (1, D, 28, 28)
+    # bad filter type -> 400
+    body["output_filter"] = "mrcnn_detection"
+    r = requests.post(
+        f"http://127.0.0.1:{port}/v1/models/mask_rcnn:predict",
+        json=body, timeout=60)
+    assert r.status_code == 400
