@@ -702,6 +702,14 @@ class ModelServer:
                                if k in want}
                 if _PROM:
                     LAT.observe(time.monotonic() - t0)
+                if "instances" in body:
+                    # TF REST row schema: "instances" requests answer
+                    # with "predictions" — one dict per input row.
+                    B = images.shape[0]
+                    preds = [{k: v[i].tolist()
+                              for k, v in outputs.items()}
+                             for i in range(B)]
+                    return {"predictions": preds}
                 return {"outputs": {k: v.tolist()
                                     for k, v in outputs.items()}}
             except Exception as e:  # noqa: BLE001

@@ -112,8 +112,13 @@ def test_rest_predict_instances_format(rest_server):
         f"http://127.0.0.1:{port}/v1/models/mask_rcnn:predict",
         json={"instances": [inst]}, timeout=120)
     assert r.status_code == 200, r.text[:300]
-    out = r.json()["outputs"]
-    assert "mrcnn_detection" in out
+    # TF REST row schema: instances -> "predictions", one per row,
+    # WITHOUT the batch dim (columnar "inputs" -> "outputs" keeps it).
+    preds = r.json()["predictions"]
+    assert len(preds) == 1
+    assert "mrcnn_detection" in preds[0]
+    D = cfg.DETECTION_MAX_INSTANCES
+    assert np.asarray(preds[0]["mrcnn_detection"]).shape == (D, 6)
 
 
 def test_rest_signature_name_validation(rest_server):
