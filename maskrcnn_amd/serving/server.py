@@ -554,11 +554,16 @@ class ModelServer:
         outer = self
 
         @app.get("/v1/models/{model_name}/metadata")
-        def model_metadata(model_name: str):
+        @app.get("/v1/models/{model_name}/versions/{version}/metadata")
+        def model_metadata(model_name: str, version: int = None):
             state = outer.models.get(model_name)
             if state is None or state.engine is None:
                 return JSONResponse({"error": "model not found"},
                                     status_code=404)
+            if version is not None and version not in state.versions:
+                return JSONResponse(
+                    {"error": f"version {version} not loaded"},
+                    status_code=404)
             sig = build_signature(state.engine.config)
             def ti(t):
                 return {"dtype": "DT_FLOAT",
@@ -566,7 +571,9 @@ class ModelServer:
                                                  for d in t.shape]},
                         "name": t.name}
             return {"model_spec": {"name": state.name,
-                                   "version": str(state.version)},
+                                   "version": str(
+                                       version if version is not None
+                                       else state.version)},
                     "metadata": {"signature_def": {"signature_def": {
                         "serving_default": {
                             "inputs": {k: ti(v)

@@ -311,3 +311,17 @@ def test_rest_output_filter_extension(rest_server):
         f"http://127.0.0.1:{port}/v1/models/mask_rcnn:predict",
         json=body, timeout=60)
     assert r.status_code == 400
+
+
+def test_rest_per_version_metadata(rest_server):
+    import requests
+    server, cfg, port = rest_server
+    r = requests.get(
+        f"http://127.0.0.1:{port}/v1/models/mask_rcnn/versions/1/"
+        f"metadata", timeout=30)
+    assert r.status_code == 200
+    assert r.json()["model_spec"]["version"] == "1"
+    r = requests.get(
+        f"http://127.0.0.1:{port}/v1/models/mask_rcnn/versions/9/"
+        f"metadata", timeout=30)
+    assert r.status_code == 404
