@@ -412,8 +412,18 @@ class ModelServer:
         from .wireproto import (GetModelStatusResponse, ModelVersionStatus,
                                 STATE_AVAILABLE)
         state = self._get_state(request.model_spec.name)
+        versions = sorted(state.versions)
+        want = request.model_spec.version
+        if want is not None:
+            # TF-Serving: a pinned model_spec.version reports only that
+            # version (unknown -> NOT_FOUND).
+            if want not in state.versions:
+                raise KeyError(
+                    f"model {state.name!r} version {want} not found "
+                    f"(serving {versions})")
+            versions = [want]
         statuses = [ModelVersionStatus(version=v, state=STATE_AVAILABLE)
-                    for v in sorted(state.versions)] or \
+                    for v in versions] or \
             [ModelVersionStatus(version=0, state=0)]
         return GetModelStatusResponse(model_version_status=statuses)
 

@@ -1409,3 +1409,27 @@ def test_batcher_uint8_mask_top_pad_combination(shapes_config):
     np.testing.assert_allclose(outs[1]["mrcnn_mask_top"],
                                solo["mrcnn_mask_top"], atol=1e-4)
     b.close()
+
+
+def test_grpc_status_version_pin(served_model):
+    """GetModelStatus with model_spec.version reports only that
+    version; unknown version -> NOT_FOUND."""
+    server, cfg, _ = served_model
+    import grpc
+    from maskrcnn_amd.serving.wireproto import (GetModelStatusRequest,
+                                                GetModelStatusResponse)
+    ch = grpc.insecure_channel(f"localhost:{server.grpc_port}")
+    stub = ch.unary_unary(
+        "/tensorflow.serving.ModelService/GetModelStatus",
+        request_serializer=lambda r: r.encode(),
+        response_deserializer=GetModelStatusResponse.decode)
+    cur = server.state.version
+    resp = stub(GetModelStatusRequest(
+        model_spec=ModelSpec(name="mask_rcnn", version=cur)), timeout=30)
+    assert [s.version for s in resp.model_version_status] == [cur]
+    with pytest.raises(grpc.RpcError) as e:
+        stub(GetModelStatusRequest(
+            model_spec=ModelSpec(name="mask_rcnn", version=987)),
+            timeout=30)
+    assert e.value.code() == grpc.StatusCode.NOT_FOUND
+    ch.close()
