@@ -1433,3 +1433,36 @@ def test_grpc_status_version_pin(served_model):
             timeout=30)
     assert e.value.code() == grpc.StatusCode.NOT_FOUND
     ch.close()
+
+
+def test_fp8_export_with_warmup_loads(tmp_path):
+    """fp8-weight SavedModel + warmup records: the quantized engine
+    loads, replays warmup, and serves (config #5 interaction)."""
+    from maskrcnn_amd.data.synthetic import ShapesConfig
+    from maskrcnn_amd.model import MaskRCNN
+    from maskrcnn_amd.serving.savedmodel import export_saved_model
+    from maskrcnn_amd.serving.server import ModelState
+
+    class TinyConfig(ShapesConfig):
+        IMAGES_PER_GPU = 1
+        PRE_NMS_LIMIT = 100
+        POST_NMS_ROIS_INFERENCE = 10
+
+    cfg = TinyConfig()
+    m = MaskRCNN("inference", cfg, model_dir=str(tmp_path))
+    base = str(tmp_path / "models" / "fp8wu")
+    export_saved_model(m, cfg, base, version=1, weight_dtype="fp8",
+                       warmup_batches=[1])
+    st = ModelState("fp8wu", base, max_batch_size=1,
+                    batch_timeout_micros=100, config=cfg)
+    st.maybe_reload()
+    assert st.version == 1
+    H = cfg.IMAGE_SHAPE[0]
+    meta = np.zeros((1, cfg.IMAGE_META_SIZE), np.float32)
+    meta[:, 4:7] = (H, H, 3)
+    meta[:, 7:11] = (0, 0, H, H)
+    out, v = st.predict(np.zeros((1, H, H, 3), np.float32), meta)
+    assert v == 1 and "mrcnn_detection" in out
+    for slot in st.versions.values():
+        for b in slot["batchers"]:
+            b.close()
