@@ -1466,3 +1466,19 @@ def test_fp8_export_with_warmup_loads(tmp_path):
     for slot in st.versions.values():
         for b in slot["batchers"]:
             b.close()
+
+
+def test_warmup_tfrecord_golden_bytes(tmp_path):
+    """Pin the TFRecord wire format byte-for-byte: u64le length,
+    u32le masked-crc32c(length), payload, u32le masked-crc32c(payload).
+    crc32c('abc') = 0x364B3FB7 is the published Castagnoli check value;
+    TF's mask is ((c>>15)|(c<<17)) + 0xA282EAD8."""
+    import struct
+    from maskrcnn_amd.serving.warmup import tfrecord_write
+    p = str(tmp_path / "g.tfrecord")
+    tfrecord_write(p, [b"abc"])
+    blob = open(p, "rb").read()
+    assert blob == (struct.pack("<Q", 3) +
+                    struct.pack("<I", 0x0E4999B0) +
+                    b"abc" +
+                    struct.pack("<I", 0x21F1576E)), blob.hex()
