@@ -581,3 +581,27 @@ def test_ingress_expired_deadline_skips_handler():
         ch2.close()
     finally:
         srv.stop()
+
+
+def test_serving_client_worker_against_cpp_ingress(cpp_served_model):
+    """The real measurement client (bench/serving_client_worker.py)
+    round-trips against the cpp ingress: uint8 + mask_top wire path,
+    zero errors."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    server, cfg, _ = cpp_served_model
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "bench/serving_client_worker.py",
+         "--target", f"127.0.0.1:{server.grpc_port}",
+         "--requests", "6", "--threads", "2",
+         "--image-size", str(cfg.IMAGE_SHAPE[0]),
+         "--num-classes", str(cfg.NUM_CLASSES)],
+        capture_output=True, text=True, timeout=300, cwd=root)
+    assert r.returncode == 0, r.stderr[-2000:]
+    rec = json.loads(r.stdout.strip().splitlines()[-1])
+    assert rec["n_errors"] == 0, rec
+    assert len(rec["lats"]) == 6
