@@ -450,17 +450,18 @@ class ModelServer:
 
         def _guard(fn):
             def call(request, context):
+                # Don't start model work for an already-expired RPC
+                # (same shedding the cpp ingress does for queued
+                # requests past their grpc-timeout). OUTSIDE the try:
+                # context.abort raises to unwind, and the broad handler
+                # below must not catch-and-abort a second time.
+                tr = context.time_remaining()
+                if tr is not None and tr <= 0:
+                    if _PROM:
+                        REJECTS.labels("grpc").inc()
+                    context.abort(grpc.StatusCode.DEADLINE_EXCEEDED,
+                                  "deadline expired before dispatch")
                 try:
-                    # Don't start model work for an already-expired RPC
-                    # (same shedding the cpp ingress does for queued
-                    # requests past their grpc-timeout).
-                    tr = context.time_remaining()
-                    if tr is not None and tr <= 0:
-                        if _PROM:
-                            REJECTS.labels("grpc").inc()
-                        context.abort(
-                            grpc.StatusCode.DEADLINE_EXCEEDED,
-                            "deadline expired before dispatch")
                     return fn(request)
                 except KeyError as e:
                     if _PROM:
