@@ -15,6 +15,10 @@ class HipBuildExt(_build_ext):
     def run(self):
         import build_ext as bx
         bx.build(verbose=True)
+        try:  # optional native gRPC front-end (g++ + libnghttp2)
+            bx.build_ingress(verbose=True)
+        except Exception as e:  # noqa: BLE001
+            print(f"WARNING: _ingress build skipped: {e}")
 
 
 setup(
@@ -23,7 +27,7 @@ setup(
     description=("MI355X-native Mask R-CNN training & serving "
                  "(PyTorch-ROCm + CDNA4 HIP kernels + RCCL)"),
     packages=find_packages(include=["maskrcnn_amd", "maskrcnn_amd.*"]),
-    package_data={"maskrcnn_amd": ["_C.so"]},
+    package_data={"maskrcnn_amd": ["_C.so", "_ingress.so"]},
     cmdclass={"build_ext": HipBuildExt},
     python_requires=">=3.10",
 )
