@@ -569,7 +569,12 @@ def test_ingress_expired_deadline_skips_handler():
 
         t = threading.Thread(target=slow)
         t.start()
-        time.sleep(0.15)  # ensure 'slow' holds the worker
+        # Wait until 'slow' has actually ENTERED the handler (a fixed
+        # sleep raced thread scheduling on loaded boxes).
+        deadline = time.time() + 20
+        while time.time() < deadline and not calls:
+            time.sleep(0.01)
+        assert calls == [b"slow"]
         import grpc
         with pytest.raises(grpc.RpcError):
             stub2(b"fast", timeout=0.2)  # expires while queued
