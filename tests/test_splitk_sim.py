@@ -97,3 +97,115 @@ def test_splitk_sim_split_invariance():
     # fp32 summation-order noise only (different slab order)
     np.testing.assert_allclose(r1, r4, rtol=1e-4, atol=1e-5)
     np.testing.assert_allclose(r1, r8, rtol=1e-4, atol=1e-5)
+
+
+def _sim_splitk_lanes(x, w, bias, split=SPLIT, relu=True):
+    """LANE-LEVEL transcription of conv3x3_splitk.hip: per-thread LDS
+    staging (lrow = tid>>2, lcol = (tid&3)*8, two short4s = 16
+    channels), per-lane MFMA fragment reads (arow/bcol/kfrag) with the
+    v_mfma_f32_16x16x32_bf16 semantics (acc lane l row r covers output
+    row 16*?+..., see fragment map), and the exact epilogue index map
+    (m = m0 + wave_m*32 + mi*16 + (lane>>4)*4 + r). Numerically fp32
+    (the HW multiplies bf16 inputs; tolerance covers that)."""
+    B, H, W, Cin = x.shape
+    Cout = w.shape[1]
+    M = B * H * W
+    BM = BN = 64
+    BKc = 32
+    KC = Cin // BKc
+    CHUNKS = 9 * KC
+    grid_x = (M + BM - 1) // BM
+    grid_y = (Cout + BN - 1) // BN
+    slabs = np.zeros((split, M, Cout), np.float32)
+
+    for bz in range(split):
+        for bx in range(grid_x):
+            for by in range(grid_y):
+                m0 = bx * BM
+                n0 = by * BN
+                # acc[wave][mi][ni][lane][r]
+                acc = np.zeros((4, 2, 2, 64, 4), np.float32)
+                for c in range(bz, CHUNKS, split):
+                    tap, k0 = c // KC, (c % KC) * BKc
+                    dy, dx = tap // 3 - 1, tap % 3 - 1
+                    a_tile = np.zeros((BM, BKc), np.float32)
+                    b_tile = np.zeros((BN, BKc), np.float32)
+                    for tid in range(256):
+                        lrow = tid >> 2
+                        lcol = (tid & 3) * 8
+                        m_load = m0 + lrow
+                        if m_load < M:
+                            px = m_load % W
+                            py = (m_load // W) % H
+                            pb = m_load // (W * H)
+                            sy, sx = py + dy, px + dx
+                            if 0 <= sy < H and 0 <= sx < W:
+                                a_tile[lrow, lcol:lcol + 16] = \
+                                    x[pb, sy, sx, k0 + lcol:k0 + lcol + 16]
+                        n = n0 + lrow
+                        if n < Cout:
+                            b_tile[lrow, lcol:lcol + 16] = \
+                                w[tap, n, k0 + lcol:k0 + lcol + 16]
+                    # Standard mfma_16x16x32 per-fragment math: for the
+                    # 16x16 output block (mi, ni) of wave (wave_m,
+                    # wave_n): OUT = Afrag @ Bfrag^T where Afrag is the
+                    # 16x32 block rows [wave_m*32+mi*16, +16) cols
+                    # [0, 32) of a_tile, Bfrag likewise of b_tile.
+                    for wave in range(4):
+                        wave_m, wave_n = wave >> 1, wave & 1
+                        for mi in range(2):
+                            r0 = wave_m * 32 + mi * 16
+                            A = a_tile[r0:r0 + 16, :]
+                            for ni in range(2):
+                                c0 = wave_n * 32 + ni * 16
+                                Bf = b_tile[c0:c0 + 16, :]
+                                out16 = A @ Bf.T
+                                # distribute into the acc lane map:
+                                # out16[row, col]: col = lane & 15,
+                                # row = (lane>>4)*4 + r
+                                for lane in range(64):
+                                    col = lane & 15
+                                    base = (lane >> 4) * 4
+                                    for r in range(4):
+                                        acc[wave, mi, ni, lane, r] += \
+                                            out16[base + r, col]
+                # epilogue: the kernel's exact index map
+                slab = slabs[bz]
+                for wave in range(4):
+                    wave_m, wave_n = wave >> 1, wave & 1
+                    for mi in range(2):
+                        for ni in range(2):
+                            for lane in range(64):
+                                ncol = n0 + wave_n * 32 + ni * 16 \
+                                    + (lane & 15)
+                                if ncol >= Cout:
+                                    continue
+                                for r in range(4):
+                                    m = m0 + wave_m * 32 + mi * 16 \
+                                        + (lane >> 4) * 4 + r
+                                    if m >= M:
+                                        continue
+                                    slab[m, ncol] = \
+                                        acc[wave, mi, ni, lane, r]
+    out = slabs.sum(axis=0) + bias[None, :]
+    if relu:
+        out = np.maximum(out, 0.0)
+    return out.reshape(B, H, W, Cout)
+
+
+def test_splitk_lane_level_sim_matches_conv2d():
+    """Full lane-level transcription (staging maps, fragment blocks,
+    acc lane map, epilogue index map) vs torch conv2d — the same
+    methodology that HW-validated conv3x3_wrw first try."""
+    rng = np.random.default_rng(5)
+    B, H, W, Cin, Cout = 1, 4, 5, 32, 48  # ragged N (48 < 64 tile)
+    x = rng.standard_normal((B, H, W, Cin), dtype=np.float32)
+    wk = rng.standard_normal((3, 3, Cin, Cout), dtype=np.float32) * 0.1
+    bias = rng.standard_normal(Cout).astype(np.float32)
+    w_packed = wk.transpose(0, 1, 3, 2).reshape(9, Cout, Cin)
+    got = _sim_splitk_lanes(x, w_packed, bias, split=4)
+    xt = torch.from_numpy(x).permute(0, 3, 1, 2)
+    wt = torch.from_numpy(wk).permute(3, 2, 0, 1)
+    ref = F.conv2d(xt, wt, torch.from_numpy(bias), padding=1)
+    ref = torch.relu(ref).permute(0, 2, 3, 1).numpy()
+    np.testing.assert_allclose(got, ref, rtol=1e-4, atol=1e-4)
