@@ -209,3 +209,19 @@ def test_splitk_lane_level_sim_matches_conv2d():
     ref = F.conv2d(xt, wt, torch.from_numpy(bias), padding=1)
     ref = torch.relu(ref).permute(0, 2, 3, 1).numpy()
     np.testing.assert_allclose(got, ref, rtol=1e-4, atol=1e-4)
+
+
+def test_splitk_lane_level_multi_tile():
+    """M > 64 (two M-tiles) + 64-channel K (two k-chunks per tap)."""
+    rng = np.random.default_rng(8)
+    B, H, W, Cin, Cout = 2, 6, 6, 64, 32   # M = 72 -> grid_x = 2
+    x = rng.standard_normal((B, H, W, Cin), dtype=np.float32)
+    wk = rng.standard_normal((3, 3, Cin, Cout), dtype=np.float32) * 0.1
+    bias = np.zeros(Cout, np.float32)
+    w_packed = wk.transpose(0, 1, 3, 2).reshape(9, Cout, Cin)
+    got = _sim_splitk_lanes(x, w_packed, bias, split=8, relu=False)
+    xt = torch.from_numpy(x).permute(0, 3, 1, 2)
+    wt = torch.from_numpy(wk).permute(3, 2, 0, 1)
+    ref = F.conv2d(xt, wt, torch.from_numpy(bias), padding=1)
+    ref = ref.permute(0, 2, 3, 1).numpy()
+    np.testing.assert_allclose(got, ref, rtol=1e-4, atol=1e-4)
