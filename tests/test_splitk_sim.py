@@ -140,12 +140,12 @@ def _sim_splitk_lanes(x, w, bias, split=SPLIT, relu=True):
                             pb = m_load // (W * H)
                             sy, sx = py + dy, px + dx
                             if 0 <= sy < H and 0 <= sx < W:
-                                a_tile[lrow, lcol:lcol + 16] = \
-                                    x[pb, sy, sx, k0 + lcol:k0 + lcol + 16]
+                                a_tile[lrow, lcol:lcol + 8] = \
+                                    x[pb, sy, sx, k0 + lcol:k0 + lcol + 8]
                         n = n0 + lrow
                         if n < Cout:
-                            b_tile[lrow, lcol:lcol + 16] = \
-                                w[tap, n, k0 + lcol:k0 + lcol + 16]
+                            b_tile[lrow, lcol:lcol + 8] = \
+                                w[tap, n, k0 + lcol:k0 + lcol + 8]
                     # Standard mfma_16x16x32 per-fragment math: for the
                     # 16x16 output block (mi, ni) of wave (wave_m,
                     # wave_n): OUT = Afrag @ Bfrag^T where Afrag is the
