@@ -92,6 +92,7 @@ constexpr uint32_t OUR_MAX_FRAME = 4u << 20;       // we accept 4 MB frames
 constexpr uint32_t OUR_CONN_WINDOW = 0x7fffffffu;  // ~2 GB recv window
 constexpr uint32_t OUR_STREAM_WINDOW = 0x40000000u;
 constexpr size_t MAX_BODY = 256u << 20;  // grpc.max_receive_message_length
+constexpr uint32_t MAX_STREAMS = 128;    // advertised + enforced per conn
 const char PREFACE[] = "PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n";
 
 inline uint32_t rd_u32(const uint8_t* p) {
@@ -147,6 +148,7 @@ struct Stream {
   std::vector<uint8_t> header_block;  // fragments until END_HEADERS
   bool headers_done = false;
   bool end_stream = false;
+  bool refused = false;  // over MAX_STREAMS: RST after HPACK decode
   int64_t send_window = 65535;  // reset from peer INITIAL_WINDOW_SIZE
   // Absolute deadline from the client's grpc-timeout header (steady
   // clock); time_point::max() = none.
@@ -416,17 +418,21 @@ class IngressServer {
       return;
     // our SETTINGS + a big connection window
     {
-      uint8_t s[9 + 12 + 13];
-      // MAX_FRAME_SIZE(0x5), INITIAL_WINDOW_SIZE(0x4)
-      frame_header(s, 12, F_SETTINGS, 0, 0);
+      uint8_t s[9 + 18 + 13];
+      // MAX_FRAME_SIZE(0x5), INITIAL_WINDOW_SIZE(0x4),
+      // MAX_CONCURRENT_STREAMS(0x3) — bounded partial-body state.
+      frame_header(s, 18, F_SETTINGS, 0, 0);
       s[9] = 0;
       s[10] = 0x5;
       wr_u32(s + 11, OUR_MAX_FRAME);
       s[15] = 0;
       s[16] = 0x4;
       wr_u32(s + 17, OUR_STREAM_WINDOW);
-      frame_header(s + 21, 4, F_WINDOW_UPDATE, 0, 0);
-      wr_u32(s + 30, OUR_CONN_WINDOW - 65535);
+      s[21] = 0;
+      s[22] = 0x3;
+      wr_u32(s + 23, MAX_STREAMS);
+      frame_header(s + 27, 4, F_WINDOW_UPDATE, 0, 0);
+      wr_u32(s + 36, OUR_CONN_WINDOW - 65535);
       if (!conn->write_all(s, sizeof(s))) return;
     }
     uint32_t continuation_sid = 0;  // nonzero while expecting CONTINUATION
@@ -510,6 +516,13 @@ class IngressServer {
           st->header_block.assign(p + off, p + end);
           {
             std::lock_guard<std::mutex> lk(conn->mu);
+            // Enforce (with slack for streams in response flight)
+            // what our SETTINGS advertised: a protocol-violating
+            // client does not get unbounded partial-body state. The
+            // refused stream's header block must STILL run through
+            // the inflater (connection-wide HPACK table), so the
+            // refusal is applied after decode in finish_headers.
+            st->refused = conn->streams.size() >= 2 * MAX_STREAMS;
             conn->streams[sid] = st;
           }
           if (flags & FL_END_HEADERS) {
@@ -638,6 +651,12 @@ class IngressServer {
     st->header_block.clear();
     st->header_block.shrink_to_fit();
     st->headers_done = true;
+    if (st->refused) {
+      rst_stream(conn, sid, 0x7 /*REFUSED_STREAM*/);
+      std::lock_guard<std::mutex> lk(conn->mu);
+      conn->streams.erase(sid);
+      return true;  // connection stays healthy
+    }
     if (st->end_stream) dispatch(conn, sid, st);  // bodyless request
     return true;
   }
