@@ -249,8 +249,10 @@ class IngressServer {
       a.sin6_family = AF_INET6;
       a.sin6_addr = in6addr_any;
       a.sin6_port = htons(uint16_t(port));
-      if (::bind(listen_fd_, (sockaddr*)&a, sizeof(a)) != 0)
+      if (::bind(listen_fd_, (sockaddr*)&a, sizeof(a)) != 0) {
+        ::close(listen_fd_);
         throw std::runtime_error("bind() failed");
+      }
       socklen_t len = sizeof(a);
       getsockname(listen_fd_, (sockaddr*)&a, &len);
       port_ = ntohs(a.sin6_port);
@@ -259,14 +261,18 @@ class IngressServer {
       a.sin_family = AF_INET;
       a.sin_addr.s_addr = INADDR_ANY;
       a.sin_port = htons(uint16_t(port));
-      if (::bind(listen_fd_, (sockaddr*)&a, sizeof(a)) != 0)
+      if (::bind(listen_fd_, (sockaddr*)&a, sizeof(a)) != 0) {
+        ::close(listen_fd_);
         throw std::runtime_error("bind() failed");
+      }
       socklen_t len = sizeof(a);
       getsockname(listen_fd_, (sockaddr*)&a, &len);
       port_ = ntohs(a.sin_port);
     }
-    if (::listen(listen_fd_, 128) != 0)
+    if (::listen(listen_fd_, 128) != 0) {
+      ::close(listen_fd_);
       throw std::runtime_error("listen() failed");
+    }
   }
 
   ~IngressServer() { stop(); }
